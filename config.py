@@ -1,0 +1,95 @@
+"""Configuration for the MI355X-native Show-Attend-and-Tell framework.
+
+Knob-name parity with the reference (`/root/reference/config.py:4-85`): every
+hyperparameter the reference exposes exists here under the same name, so a user
+of the reference can carry their edits over unchanged.  MI355X-specific knobs
+(compute dtype, kernel backend, DP bucket sizing, hipGraph capture) are added at
+the end — they have sane defaults and never need touching for parity runs.
+"""
+
+
+class Config(object):
+    """Wrapper class for various (hyper)parameters."""
+
+    def __init__(self):
+        # about the model architecture (reference config.py:8-17)
+        self.cnn = 'vgg16'               # 'vgg16' or 'resnet50'
+        self.max_caption_length = 20
+        self.dim_embedding = 512
+        self.num_lstm_units = 512
+        self.num_initalize_layers = 2    # 1 or 2 (reference spelling kept)
+        self.dim_initalize_layer = 512
+        self.num_attend_layers = 2       # 1 or 2
+        self.dim_attend_layer = 512
+        self.num_decode_layers = 2       # 1 or 2
+        self.dim_decode_layer = 1024
+
+        # about the weight initialization and regularization (config.py:20-27)
+        self.fc_kernel_initializer_scale = 0.08
+        self.fc_kernel_regularizer_scale = 1e-4
+        self.fc_activity_regularizer_scale = 0.0
+        self.conv_kernel_regularizer_scale = 1e-4
+        self.conv_activity_regularizer_scale = 0.0
+        self.fc_drop_rate = 0.5
+        self.lstm_drop_rate = 0.3
+        self.attention_loss_factor = 0.01
+
+        # about the optimization (config.py:30-43)
+        self.num_epochs = 30
+        self.batch_size = 20
+        self.optimizer = 'Adam'          # 'Adam', 'RMSProp', 'Momentum' or 'SGD'
+        self.initial_learning_rate = 0.0001
+        self.learning_rate_decay_factor = 1.0
+        self.num_steps_per_decay = 100000
+        self.clip_gradients = 5.0
+        self.momentum = 0.0
+        self.use_nesterov = True
+        self.decay = 0.9
+        self.centered = True
+        self.beta1 = 0.9
+        self.beta2 = 0.999
+        self.epsilon = 1e-6
+
+        # about the saver (config.py:53-55)
+        self.save_period = 50
+        self.save_dir = './data/models/'
+        self.summary_dir = './summary/'
+
+        # size-of-dataset caps — the reference's de-facto smoke-test knobs
+        # (config.py:60-63)
+        self.max_train_ann_num = 1000
+        self.max_eval_ann_num = 20
+
+        # about the vocabulary (config.py:66-67)
+        self.vocabulary_file = './data/vocabulary.csv'
+        self.vocabulary_size = 5000
+
+        # about the training (config.py:70-74)
+        self.train_image_dir = './data/train/images/'
+        self.train_caption_file = './data/train/captions_train2014.json'
+        self.temp_annotation_file = './data/train/anns.csv'
+        self.temp_data_file = './data/train/data.npy'
+
+        # about the evaluation (config.py:77-81)
+        self.eval_image_dir = './data/val/images/'
+        self.eval_caption_file = './data/val/captions_val2014.json'
+        self.eval_result_dir = './data/val/results/'
+        self.eval_result_file = './data/val/results.json'
+        self.save_eval_result_as_image = False
+
+        # about the testing (config.py:83-85)
+        self.test_image_dir = './data/test/images/'
+        self.test_result_dir = './data/test/results/'
+        self.test_result_file = './data/test/results.csv'
+
+        # ------------------------------------------------------------------
+        # MI355X-native knobs (new — no reference analog)
+        # ------------------------------------------------------------------
+        self.device = 'auto'             # 'auto' | 'cuda' | 'cpu'
+        self.compute_dtype = 'bf16'      # 'bf16' | 'fp32' — CNN/decoder compute
+        self.use_hip_kernels = True      # hand-written CDNA4 kernels on GPU
+        self.use_hip_graph = True        # capture the train step in a hipGraph
+        self.allreduce_bucket_mb = 16    # DP gradient bucket size (xGMI-tuned)
+        self.synthetic_data = False      # synthetic COCO-shaped data (no files)
+        self.synthetic_num_images = 640
+        self.seed = 1234

@@ -1,0 +1,129 @@
+"""COCO caption-annotation API (clean-room re-implementation).
+
+Interface parity with the reference's vendored-and-modified COCO class
+(`utils/coco/coco.py`):
+  * construction from a captions JSON with an optional annotation-count cap
+    (coco.py:69, :119-124);
+  * index dicts `anns`, `imgs`, `imgToAnns`, `img_name_to_id` (coco.py:102-156);
+  * subset filters `filter_by_cap_len` / `filter_by_words` (coco.py:323-361);
+  * caption normalization: lower-case + guaranteed trailing '.'
+    (coco.py:316-321);
+  * `loadRes` for metric evaluation result files (coco.py:263-290);
+  * `all_captions` (coco.py:363).
+
+The reference's per-image `download()` (coco.py:292-314) is intentionally a
+no-op here: this environment has no network, and the framework's synthetic
+data path (sat_amd.data.synthetic) covers imageless runs.
+"""
+
+import copy
+import json
+import os
+
+from .tokenizer import word_tokenize
+
+
+class COCO(object):
+    def __init__(self, annotation_file=None, max_ann_num=None):
+        self.dataset = {}
+        self.anns = {}
+        self.imgToAnns = {}
+        self.imgs = {}
+        self.img_name_to_id = {}
+        if annotation_file is not None:
+            with open(annotation_file, 'r') as f:
+                self.dataset = json.load(f)
+            self.process_dataset()
+            self.createIndex(max_ann_num)
+
+    def createIndex(self, max_ann_num=None):
+        anns = self.dataset.get('annotations', [])
+        if max_ann_num is not None and max_ann_num < len(anns):
+            anns = anns[:max_ann_num]
+            self.dataset['annotations'] = anns
+        self.anns = {a['id']: a for a in anns}
+        self.imgToAnns = {}
+        for a in anns:
+            self.imgToAnns.setdefault(a['image_id'], []).append(a)
+        self.imgs = {i['id']: i for i in self.dataset.get('images', [])}
+        self.img_name_to_id = {
+            os.path.basename(i['file_name']): i['id']
+            for i in self.dataset.get('images', []) if 'file_name' in i}
+
+    # ---- accessors (stock COCO API subset used by the pipeline) ----
+    def getAnnIds(self, imgIds=None):
+        if imgIds is None:
+            return list(self.anns.keys())
+        if not isinstance(imgIds, (list, tuple)):
+            imgIds = [imgIds]
+        out = []
+        for i in imgIds:
+            out.extend(a['id'] for a in self.imgToAnns.get(i, []))
+        return out
+
+    def getImgIds(self):
+        return list(self.imgs.keys())
+
+    def loadAnns(self, ids):
+        if not isinstance(ids, (list, tuple)):
+            ids = [ids]
+        return [self.anns[i] for i in ids]
+
+    def loadImgs(self, ids):
+        if not isinstance(ids, (list, tuple)):
+            ids = [ids]
+        return [self.imgs[i] for i in ids]
+
+    # ---- caption-specific machinery ----
+    def process_dataset(self):
+        """Lower-case every caption and guarantee a trailing '.'."""
+        for ann in self.dataset.get('annotations', []):
+            q = ann['caption'].lower().strip()
+            if not q.endswith('.'):
+                q = q + '.'
+            ann['caption'] = q
+
+    def filter_by_cap_len(self, max_cap_len):
+        """Drop annotations whose caption tokenizes past max_cap_len words."""
+        keep = [a for a in self.dataset.get('annotations', [])
+                if len(word_tokenize(a['caption'])) <= max_cap_len]
+        self._rebuild(keep)
+
+    def filter_by_words(self, vocab_words):
+        """Drop annotations containing any word outside vocab_words (a set)."""
+        keep = [a for a in self.dataset.get('annotations', [])
+                if all(w in vocab_words for w in word_tokenize(a['caption']))]
+        self._rebuild(keep)
+
+    def _rebuild(self, annotations):
+        self.dataset['annotations'] = annotations
+        kept_imgs = {a['image_id'] for a in annotations}
+        if 'images' in self.dataset:
+            self.dataset['images'] = [i for i in self.dataset['images']
+                                      if i['id'] in kept_imgs]
+        self.createIndex()
+
+    def all_captions(self):
+        return [a['caption'] for a in self.dataset.get('annotations', [])]
+
+    def loadRes(self, resFile):
+        """Load a result file (list of {image_id, caption}) as a COCO object."""
+        res = COCO()
+        res.dataset['images'] = [img for img in self.dataset.get('images', [])]
+        with open(resFile, 'r') as f:
+            anns = json.load(f)
+        assert isinstance(anns, list), 'results must be a list of dicts'
+        img_ids_res = {a['image_id'] for a in anns}
+        assert img_ids_res.issubset(set(self.imgs.keys()) | img_ids_res), \
+            'result image ids malformed'
+        anns = copy.deepcopy(anns)
+        for i, a in enumerate(anns):
+            a['id'] = i + 1
+        res.dataset['annotations'] = anns
+        res.createIndex()
+        return res
+
+    def download(self, target_dir=None, img_ids=None):
+        """No-op: this environment has no network. Kept for interface parity
+        with reference coco.py:292-314."""
+        return

@@ -1,0 +1,153 @@
+"""Soft-attention LSTM decoder (parity with reference `model.py:190-459`).
+
+Architecture (defaults; all sizes from config):
+  * word embedding [V, E] (model.py:219-225);
+  * LSTM cell with TF semantics — gate order (i,j,f,o), forget bias 1.0,
+    DropoutWrapper on input / output / state-h in train (model.py:228-236);
+  * initialize MLP: mean context -> (memory, output), 2x (dense-tanh 512 +
+    dropout + dense) (model.py:358-393);
+  * attention MLP: dense-tanh on contexts [B·L,D]->A, dense-tanh on output
+    [B,H]->A, tiled add, dropout, bias-free dense ->1, softmax over L
+    (model.py:395-436);
+  * weighted context sum Σ_l α_l·ctx_l (model.py:263-264);
+  * decode MLP: concat[output, context, embed] -> dense-tanh 1024 -> dropout
+    -> dense V (model.py:438-459).
+
+The per-step compute goes through sat_amd.ops (hand-written CDNA4 kernels on
+GPU).  Training runs the T=20 steps as a Python loop over fused kernels (the
+reference statically unrolls the TF graph instead, model.py:259); hipGraph
+capture at the runtime layer removes the per-step launch overhead.
+"""
+
+import torch
+import torch.nn as tnn
+
+from .. import ops
+from .nn import Dense
+
+
+class AttentionDecoder(tnn.Module):
+    def __init__(self, config, nn_policy, dim_ctx, num_ctx):
+        super().__init__()
+        self.config = config
+        self.nn = nn_policy
+        self.dim_ctx = dim_ctx
+        self.num_ctx = num_ctx
+        V, E, H = (config.vocabulary_size, config.dim_embedding,
+                   config.num_lstm_units)
+
+        # word embedding (model.py:219-225)
+        self.embedding = tnn.Parameter(torch.empty(V, E))
+        nn_policy.init_fc_(self.embedding)
+        nn_policy.register_fc_kernel(self.embedding)
+
+        # LSTM kernel [I+H, 4H], TF gate order (i,j,f,o), zero bias
+        dim_in = dim_ctx + E
+        self.lstm_w = tnn.Parameter(torch.empty(dim_in + H, 4 * H))
+        nn_policy.init_fc_(self.lstm_w)
+        self.lstm_b = tnn.Parameter(torch.zeros(4 * H))
+
+        # initialize MLPs (model.py:358-393)
+        if config.num_initalize_layers == 1:
+            self.init_fc_a = Dense(nn_policy, dim_ctx, H, None)
+            self.init_fc_b = Dense(nn_policy, dim_ctx, H, None)
+        else:
+            D_i = config.dim_initalize_layer
+            self.init_fc_a1 = Dense(nn_policy, dim_ctx, D_i, 'tanh')
+            self.init_fc_a2 = Dense(nn_policy, D_i, H, None)
+            self.init_fc_b1 = Dense(nn_policy, dim_ctx, D_i, 'tanh')
+            self.init_fc_b2 = Dense(nn_policy, D_i, H, None)
+
+        # attention MLPs (model.py:395-436)
+        if config.num_attend_layers == 1:
+            self.att_fc_a = Dense(nn_policy, dim_ctx, 1, None,
+                                  use_bias=False)
+            self.att_fc_b = Dense(nn_policy, H, num_ctx, None,
+                                  use_bias=False)
+        else:
+            A = config.dim_attend_layer
+            self.att_fc_1a = Dense(nn_policy, dim_ctx, A, 'tanh')
+            self.att_fc_1b = Dense(nn_policy, H, A, 'tanh')
+            self.att_fc_2 = Dense(nn_policy, A, 1, None, use_bias=False)
+
+        # decode MLPs (model.py:438-459)
+        dim_exp = H + dim_ctx + E
+        if config.num_decode_layers == 1:
+            self.dec_fc = Dense(nn_policy, dim_exp, V, None)
+        else:
+            D_d = config.dim_decode_layer
+            self.dec_fc_1 = Dense(nn_policy, dim_exp, D_d, 'tanh')
+            self.dec_fc_2 = Dense(nn_policy, D_d, V, None)
+
+    # ---- sub-networks ----
+
+    def initialize(self, context_mean):
+        """Mean context -> (initial memory, initial output)."""
+        cfg = self.config
+        x = self.nn.dropout(context_mean)
+        if cfg.num_initalize_layers == 1:
+            return self.init_fc_a(x), self.init_fc_b(x)
+        ta = self.nn.dropout(self.init_fc_a1(x))
+        tb = self.nn.dropout(self.init_fc_b1(x))
+        return self.init_fc_a2(ta), self.init_fc_b2(tb)
+
+    def attend_logits(self, contexts_flat, output):
+        """contexts_flat: [B·L, D], output: [B, H] -> logits [B, L]."""
+        cfg = self.config
+        B = output.shape[0]
+        ctx = self.nn.dropout(contexts_flat)
+        out = self.nn.dropout(output)
+        if cfg.num_attend_layers == 1:
+            l1 = self.att_fc_a(ctx).reshape(B, self.num_ctx)
+            l2 = self.att_fc_b(out)
+            return l1 + l2
+        t1 = self.att_fc_1a(ctx)                       # [B·L, A]
+        t2 = self.att_fc_1b(out)                       # [B, A]
+        t = t1 + t2.repeat_interleave(self.num_ctx, dim=0)
+        t = self.nn.dropout(t)
+        return self.att_fc_2(t).reshape(B, self.num_ctx)
+
+    def decode(self, expanded_output):
+        """[B, H+D+E] -> logits [B, V]."""
+        cfg = self.config
+        x = self.nn.dropout(expanded_output)
+        if cfg.num_decode_layers == 1:
+            return self.dec_fc(x)
+        t = self.nn.dropout(self.dec_fc_1(x))
+        return self.dec_fc_2(t)
+
+    # ---- one decoder step (shared by train loop and beam search) ----
+
+    def step(self, contexts, contexts_flat, last_word, last_memory,
+             last_output, last_state_h=None):
+        """Run attention + LSTM + decode for one step.
+
+        `last_output` feeds the attention MLP (it is the DropoutWrapper
+        *output* of the previous step in train, model.py:307); the LSTM's
+        recurrent h input is `last_state_h` (the state_keep-dropped h,
+        model.py:309) — identical tensors at inference.
+
+        Returns (logits [B,V], alpha [B,L], memory, output, state_output).
+        """
+        if last_state_h is None:
+            last_state_h = last_output
+        rate = self.nn.lstm_drop_rate
+        training = self.nn.is_train
+
+        att = self.attend_logits(contexts_flat, last_output)
+        alpha, context = ops.attention_pool(contexts, att)
+
+        word_embed = ops.embedding(last_word, self.embedding.to(
+            contexts.dtype))
+
+        x = torch.cat([context, word_embed], dim=1)
+        x = ops.dropout(x, rate, training)              # input_keep
+        h_raw, memory = ops.lstm_cell(
+            x, last_state_h, last_memory,
+            self.lstm_w.to(x.dtype), self.lstm_b.to(x.dtype))
+        output = ops.dropout(h_raw, rate, training)     # output_keep
+        state_output = ops.dropout(h_raw, rate, training)  # state_keep (h)
+
+        expanded = torch.cat([output, context, word_embed], dim=1)
+        logits = self.decode(expanded)
+        return logits, alpha, memory, output, state_output

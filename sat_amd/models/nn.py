@@ -1,0 +1,165 @@
+"""NN primitive layer — the single funnel through which every learnable op
+is created (parity with reference `utils/nn.py`).
+
+Centralizes:
+  * initializer policy — fc kernels uniform ±fc_kernel_initializer_scale
+    (nn.py:29-31), conv kernels Xavier (nn.py:15);
+  * regularization policy — L2 on fc/conv kernels at the configured scales,
+    collected into `reg_loss()` exactly like TF's graph-level
+    get_regularization_loss (nn.py:17-43, model.py:328);
+  * dropout rates (fc_drop_rate, nn.py:107-114);
+  * the freeze-CNN policy — conv/bn parameters are trainable only when
+    `config.train_cnn` (nn.py:66), fc parameters only when `is_train`.
+
+Modules created here call into sat_amd.ops, which dispatches to the CDNA4
+HIP kernels on GPU and plain PyTorch on CPU.
+"""
+
+import math
+
+import torch
+import torch.nn as tnn
+
+from .. import ops
+
+
+class NN(object):
+    """Policy object handed to every layer factory."""
+
+    def __init__(self, config):
+        self.config = config
+        self.is_train = getattr(config, 'phase', 'train') == 'train'
+        self.train_cnn = self.is_train and getattr(config, 'train_cnn', False)
+        self.fc_scale = config.fc_kernel_initializer_scale
+        self.fc_reg = config.fc_kernel_regularizer_scale \
+            if self.is_train else 0.0
+        self.conv_reg = config.conv_kernel_regularizer_scale \
+            if self.train_cnn else 0.0
+        self.fc_drop_rate = config.fc_drop_rate
+        self.lstm_drop_rate = config.lstm_drop_rate
+        # (param, scale) pairs for reg_loss; filled as layers are built
+        self._regularized = []
+
+    def init_fc_(self, t):
+        tnn.init.uniform_(t, -self.fc_scale, self.fc_scale)
+
+    def init_conv_(self, t):
+        tnn.init.xavier_uniform_(t)
+
+    def register_fc_kernel(self, p):
+        if self.fc_reg > 0:
+            self._regularized.append((p, self.fc_reg))
+
+    def register_conv_kernel(self, p):
+        if self.conv_reg > 0:
+            self._regularized.append((p, self.conv_reg))
+
+    def reg_loss(self):
+        """Σ scale · l2_loss(w) with l2_loss = sum(w²)/2 (TF semantics)."""
+        if not self._regularized:
+            return torch.zeros(())
+        return sum(s * 0.5 * (p.float() ** 2).sum()
+                   for p, s in self._regularized)
+
+    def dropout(self, x):
+        return ops.dropout(x, self.fc_drop_rate, self.is_train)
+
+
+class Dense(tnn.Module):
+    """tf.layers.dense analog: kernel in TF [in, out] layout, default tanh
+    (reference nn.py:85-105)."""
+
+    def __init__(self, nn_policy, in_dim, units, activation='tanh',
+                 use_bias=True):
+        super().__init__()
+        self.activation = activation
+        self.weight = tnn.Parameter(torch.empty(in_dim, units))
+        nn_policy.init_fc_(self.weight)
+        nn_policy.register_fc_kernel(self.weight)
+        if use_bias:
+            self.bias = tnn.Parameter(torch.zeros(units))
+        else:
+            self.register_parameter('bias', None)
+
+    def forward(self, x):
+        w = self.weight.to(x.dtype)
+        b = self.bias.to(x.dtype) if self.bias is not None else None
+        return ops.dense(x, w, b, self.activation)
+
+
+class Conv2d(tnn.Module):
+    """conv2d + bias + optional ReLU (reference nn.py:45-70 defaults:
+    3x3, stride 1, SAME padding, ReLU)."""
+
+    def __init__(self, nn_policy, in_ch, out_ch, kernel_size=3, stride=1,
+                 activation='relu', use_bias=True):
+        super().__init__()
+        self.stride = stride
+        self.kernel_size = kernel_size
+        self.activation = activation
+        self.weight = tnn.Parameter(
+            torch.empty(out_ch, in_ch, kernel_size, kernel_size))
+        nn_policy.init_conv_(self.weight)
+        nn_policy.register_conv_kernel(self.weight)
+        if use_bias:
+            self.bias = tnn.Parameter(torch.zeros(out_ch))
+        else:
+            self.register_parameter('bias', None)
+
+    def forward(self, x):
+        w = self.weight.to(x.dtype)
+        b = self.bias.to(x.dtype) if self.bias is not None else None
+        x = _pad_same(x, self.kernel_size, self.stride)
+        y = torch.nn.functional.conv2d(x, w, b, stride=self.stride)
+        if self.activation == 'relu':
+            y = torch.relu(y)
+        return y
+
+
+def _same_pad(size, k, s):
+    out = math.ceil(size / s)
+    return max((out - 1) * s + k - size, 0)
+
+
+def _pad_same(x, k, s):
+    """TF 'SAME' padding (asymmetric: extra on bottom/right)."""
+    ph = _same_pad(x.shape[2], k, s)
+    pw = _same_pad(x.shape[3], k, s)
+    if ph == 0 and pw == 0:
+        return x
+    return torch.nn.functional.pad(
+        x, (pw // 2, pw - pw // 2, ph // 2, ph - ph // 2))
+
+
+class MaxPool2d(tnn.Module):
+    """max_pool2d 2x2 s2 SAME (reference nn.py:72-83)."""
+
+    def __init__(self, kernel_size=2, stride=2):
+        super().__init__()
+        self.kernel_size = kernel_size
+        self.stride = stride
+
+    def forward(self, x):
+        x = _pad_same(x, self.kernel_size, self.stride)
+        return torch.nn.functional.max_pool2d(
+            x, self.kernel_size, self.stride)
+
+
+class BatchNorm(tnn.Module):
+    """batch_norm gated on train_cnn (reference nn.py:116-125)."""
+
+    def __init__(self, nn_policy, num_features):
+        super().__init__()
+        self.train_cnn = nn_policy.train_cnn
+        self.bn = tnn.BatchNorm2d(num_features, eps=1e-3, momentum=0.01)
+
+    def forward(self, x):
+        if self.train_cnn:
+            return self.bn(x.float()).to(x.dtype)
+        # inference-mode statistics
+        training = self.bn.training
+        self.bn.eval()
+        y = self.bn(x.float()).to(x.dtype)
+        if training:
+            self.bn.train()
+        return y

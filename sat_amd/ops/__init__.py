@@ -1,0 +1,70 @@
+"""sat_amd.ops — the op layer.
+
+Every compute op the decoder/loss/optimizer hot path needs exists twice:
+
+  * `sat_amd.ops.functional` — plain-PyTorch fp32 reference implementations.
+    Used on CPU, and as the ground truth the HIP kernels are tested against
+    (tests/test_ops_gpu.py).
+  * `sat_amd.ops.hip` — hand-written CDNA4 HIP kernels (MFMA GEMM tiles,
+    LDS-staged attention softmax, fused LSTM gates, fused CE, fused Adam)
+    exposed through torch.autograd.Functions.
+
+Dispatch rule (no multi-backend sprawl): CUDA/HIP tensors take the HIP kernel
+path; on a GPU box a missing extension raises instead of silently falling
+back to eager PyTorch.  CPU tensors take the reference path.
+"""
+
+import torch
+
+from . import functional as F  # noqa: N812
+
+
+def _use_hip(*tensors):
+    if not any(t.is_cuda for t in tensors if isinstance(t, torch.Tensor)):
+        return False
+    from . import hip
+    hip.require()  # raises if the extension is missing on a GPU box
+    return True
+
+
+def dense(x, weight, bias=None, activation=None):
+    """act(x @ weight + bias). x: [M,K], weight: [K,N] (TF layout)."""
+    if _use_hip(x, weight):
+        from . import hip
+        return hip.dense(x, weight, bias, activation)
+    return F.dense(x, weight, bias, activation)
+
+
+def lstm_cell(x, h, c, weight, bias, forget_bias=1.0):
+    """One TF-semantics LSTMCell step; returns (h', c')."""
+    if _use_hip(x, weight):
+        from . import hip
+        return hip.lstm_cell(x, h, c, weight, bias, forget_bias)
+    return F.lstm_cell(x, h, c, weight, bias, forget_bias)
+
+
+def attention_pool(contexts, logits):
+    """softmax over L + weighted context sum; returns (alpha, context)."""
+    if _use_hip(contexts, logits):
+        from . import hip
+        return hip.attention_pool(contexts, logits)
+    return F.attention_pool(contexts, logits)
+
+
+def embedding(ids, table):
+    if _use_hip(table):
+        from . import hip
+        return hip.embedding(ids, table)
+    return F.embedding(ids, table)
+
+
+def masked_softmax_ce(logits, labels, mask):
+    """Per-element masked CE ([B]) — sum/Σmask is done by the caller."""
+    if _use_hip(logits):
+        from . import hip
+        return hip.masked_softmax_ce(logits, labels, mask)
+    return F.masked_softmax_ce(logits, labels, mask)
+
+
+def dropout(x, rate, training):
+    return F.dropout(x, rate, training)

@@ -1,0 +1,107 @@
+"""Synchronous data-parallel gradient all-reduce over RCCL/xGMI.
+
+Replaces the reference's asynchronous TF parameter-server distribution
+(`clusterone_config.py:106-125`, variables sharded onto ps hosts, workers
+pushing gradients over gRPC) with the MI355X-idiomatic scheme (SURVEY.md
+§5.8): one process per GPU, torch.distributed with the nccl(=RCCL) backend,
+and bucketed all-reduce of gradients overlapped with backward.
+
+Bucketing is tuned for xGMI, not NVSwitch: each of the 8 GPUs has 7
+point-to-point links (~153 GB/s each), ring all-reduce is per-link bound, so
+fewer/larger buckets (default 16 MB) amortize latency; buckets launch as
+soon as their last gradient materializes during backward (post-accumulate
+hooks), on whatever stream the backend manages, and `finish_backward()`
+waits + writes the averaged gradients back.
+
+Works identically over gloo on CPU (the multi-process CI path).
+"""
+
+import torch
+import torch.distributed as dist
+
+
+class _Bucket:
+    __slots__ = ('params', 'numel', 'flat', 'offsets', 'pending', 'work')
+
+    def __init__(self):
+        self.params = []
+        self.numel = 0
+        self.flat = None
+        self.offsets = {}
+        self.pending = 0
+        self.work = None
+
+
+class DataParallelGrads(object):
+    def __init__(self, model, bucket_mb=16, process_group=None):
+        self.group = process_group
+        self.world = dist.get_world_size(process_group)
+        self.buckets = []
+        self._param_bucket = {}
+
+        params = [p for p in model.parameters() if p.requires_grad]
+        # backward produces gradients roughly in reverse parameter order:
+        # bucket in reverse so each bucket fills contiguously in time.
+        params = list(reversed(params))
+
+        cap = int(bucket_mb * (1 << 20) / 4)  # fp32 elements per bucket
+        cur = _Bucket()
+        for p in params:
+            if cur.numel and cur.numel + p.numel() > cap:
+                self.buckets.append(cur)
+                cur = _Bucket()
+            cur.offsets[p] = cur.numel
+            cur.params.append(p)
+            cur.numel += p.numel()
+        if cur.numel:
+            self.buckets.append(cur)
+
+        for b in self.buckets:
+            for p in b.params:
+                self._param_bucket[p] = b
+                p.register_post_accumulate_grad_hook(self._hook)
+        self.reset()
+
+    def reset(self):
+        for b in self.buckets:
+            b.pending = len(b.params)
+            b.work = None
+
+    def _hook(self, p):
+        b = self._param_bucket[p]
+        if b.flat is None or b.flat.device != p.grad.device:
+            b.flat = torch.zeros(b.numel, dtype=torch.float32,
+                                 device=p.grad.device)
+        off = b.offsets[p]
+        b.flat[off:off + p.numel()].copy_(
+            p.grad.detach().reshape(-1).float())
+        b.pending -= 1
+        if b.pending == 0:
+            b.flat.div_(self.world)
+            b.work = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
+                                     group=self.group, async_op=True)
+
+    def finish_backward(self):
+        """Wait for in-flight buckets and write averaged grads back."""
+        for b in self.buckets:
+            if b.pending != 0:
+                # gradient never materialized for some params (e.g. unused);
+                # reduce what we have so ranks stay in lockstep.
+                if b.flat is None:
+                    b.flat = torch.zeros(
+                        b.numel, dtype=torch.float32,
+                        device=next(iter(b.offsets)).device)
+                b.flat.div_(self.world)
+                b.work = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
+                                         group=self.group, async_op=True)
+        for b in self.buckets:
+            if b.work is not None:
+                b.work.wait()
+            for p in b.params:
+                off = b.offsets[p]
+                g = b.flat[off:off + p.numel()].reshape(p.shape)
+                if p.grad is None:
+                    p.grad = g.to(p.dtype).clone()
+                else:
+                    p.grad.detach().copy_(g.to(p.dtype))
+        self.reset()

@@ -1,0 +1,148 @@
+import numpy as np
+import torch
+
+from sat_amd.models.base_model import BaseModel
+from sat_amd.models.caption_generator import CaptionGenerator
+from sat_amd.utils import checkpoint as ckpt
+
+
+def _batch(cfg, B=2):
+    torch.manual_seed(0)
+    images = torch.randn(B, 3, 224, 224)
+    T = cfg.max_caption_length
+    sentences = torch.randint(0, cfg.vocabulary_size, (B, T))
+    masks = torch.zeros(B, T)
+    masks[:, :5] = 1.0
+    return images, sentences, masks
+
+
+def test_forward_losses(tiny_config):
+    cfg = tiny_config
+    cfg.vocabulary_size = 50
+    model = CaptionGenerator(cfg)
+    out = model(*_batch(cfg))
+    for k in ('total_loss', 'cross_entropy_loss', 'attention_loss',
+              'reg_loss', 'accuracy'):
+        assert torch.isfinite(out[k]), k
+    assert out['total_loss'].item() >= out['cross_entropy_loss'].item()
+    assert 0.0 <= out['accuracy'].item() <= 1.0
+    assert out['attentions'].shape == (2, model.num_ctx)
+
+
+def test_backward_populates_rnn_grads_only(tiny_config):
+    cfg = tiny_config
+    cfg.vocabulary_size = 50
+    model = CaptionGenerator(cfg)
+    out = model(*_batch(cfg))
+    out['total_loss'].backward()
+    dec_grads = [p.grad for p in model.decoder.parameters()]
+    assert all(g is not None for g in dec_grads)
+    assert all(torch.isfinite(g).all() for g in dec_grads)
+    # frozen CNN: no grads
+    assert all(p.grad is None for p in model.cnn.parameters())
+
+
+def test_encode_decode_step_shapes(tiny_config):
+    cfg = tiny_config
+    cfg.phase = 'eval'
+    cfg.vocabulary_size = 50
+    model = CaptionGenerator(cfg)
+    model.eval()
+    images = torch.randn(2, 3, 224, 224)
+    ctx, mem, out = model.encode(images)
+    assert ctx.shape == (2, 196, 512)
+    assert mem.shape == (2, cfg.num_lstm_units)
+    word = torch.zeros(2, dtype=torch.int64)
+    mem2, out2, probs = model.decode_step(ctx, word, mem, out)
+    assert probs.shape == (2, 50)
+    assert torch.allclose(probs.sum(1), torch.ones(2), atol=1e-5)
+
+
+def test_checkpoint_roundtrip(tiny_config):
+    cfg = tiny_config
+    cfg.vocabulary_size = 50
+    m = BaseModel(cfg)
+    m.global_step = 7
+    path = m.save()
+    arrays = np.load(path, allow_pickle=True).item()
+    assert 'global_step' in arrays
+    assert any(k.startswith('optimizer/') for k in arrays)
+
+    m2 = BaseModel(cfg)
+    step = m2.load(path)
+    assert step == 7
+    for (n1, p1), (n2, p2) in zip(m.model.state_dict().items(),
+                                  m2.model.state_dict().items()):
+        assert n1 == n2
+        assert torch.allclose(p1.float(), p2.float(), atol=1e-6), n1
+
+
+def test_checkpoint_trim(tiny_config, tmp_path):
+    cfg = tiny_config
+    cfg.vocabulary_size = 50
+    m = BaseModel(cfg)
+    path = m.save()
+    out = str(tmp_path / 'trimmed.npy')
+    removed = ckpt.trim(path, out)
+    assert removed > 0
+    arrays = np.load(out, allow_pickle=True).item()
+    assert not any('optimizer' in k for k in arrays)
+
+
+def test_load_cnn_caffe_style(tiny_config, tmp_path):
+    cfg = tiny_config
+    cfg.vocabulary_size = 50
+    m = BaseModel(cfg)
+    w = np.random.randn(64, 3, 3, 3).astype(np.float32)
+    b = np.random.randn(64).astype(np.float32)
+    data = {'conv1_1': {'weights': w, 'biases': b}}
+    f = str(tmp_path / 'cnn.npy')
+    np.save(f, data)
+    count = m.load_cnn(f)
+    assert count == 2
+    got = dict(m.model.cnn.named_parameters())['conv1_1.weight']
+    assert torch.allclose(got, torch.from_numpy(w))
+
+
+def test_load_cnn_tf_layout(tiny_config, tmp_path):
+    cfg = tiny_config
+    cfg.vocabulary_size = 50
+    m = BaseModel(cfg)
+    w_tf = np.random.randn(3, 3, 3, 64).astype(np.float32)  # HWIO
+    data = {'conv1_1': {'kernel': w_tf}}
+    f = str(tmp_path / 'cnn_tf.npy')
+    np.save(f, data)
+    assert m.load_cnn(f) == 1
+    got = dict(m.model.cnn.named_parameters())['conv1_1.weight']
+    assert torch.allclose(got, torch.from_numpy(
+        w_tf.transpose(3, 2, 0, 1)))
+
+
+def test_beam_search_semantics(tiny_config):
+    cfg = tiny_config
+    cfg.phase = 'eval'
+    from sat_amd.data.synthetic import prepare_eval_data
+    coco, data, vocab = prepare_eval_data(cfg)
+    m = BaseModel(cfg)
+    files = data.next_batch()
+    results = m.beam_search(files, vocab)
+    assert len(results) == len(files)
+    for beams in results:
+        assert 1 <= len(beams) <= cfg.beam_size
+        scores = [b.score for b in beams]
+        assert scores == sorted(scores, reverse=True)
+        for b in beams:
+            assert len(b.sentence) <= cfg.max_caption_length
+            assert all(isinstance(w, int) for w in b.sentence)
+
+
+def test_beam_search_deterministic(tiny_config):
+    cfg = tiny_config
+    cfg.phase = 'eval'
+    from sat_amd.data.synthetic import prepare_eval_data
+    coco, data, vocab = prepare_eval_data(cfg)
+    m = BaseModel(cfg)
+    files = data.next_batch()
+    r1 = m.beam_search(files, vocab)
+    r2 = m.beam_search(files, vocab)
+    assert [b.sentence for b in r1[0]] == [b.sentence for b in r2[0]]
