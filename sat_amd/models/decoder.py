@@ -43,7 +43,7 @@ class AttentionDecoder(tnn.Module):
 
         # LSTM kernel [I+H, 4H], TF gate order (i,j,f,o), zero bias
         dim_in = dim_ctx + E
-        self.lstm_w = tnn.Parameter(torch.empty(dim_in + H, 4 * H))
+        self.lstm_w = tnn.Parameter(torch.empty(4 * H, dim_in + H))
         nn_policy.init_fc_(self.lstm_w)
         self.lstm_b = tnn.Parameter(torch.zeros(4 * H))
 
@@ -91,8 +91,10 @@ class AttentionDecoder(tnn.Module):
         tb = self.nn.dropout(self.init_fc_b1(x))
         return self.init_fc_a2(ta), self.init_fc_b2(tb)
 
-    def attend_logits(self, contexts_flat, output):
-        """contexts_flat: [B·L, D], output: [B, H] -> logits [B, L]."""
+    def attend(self, contexts, contexts_flat, output):
+        """Attention: MLP scores + softmax over L + weighted context sum
+        (model.py:395-436, :263-264).  Returns (alpha [B,L], context [B,D]).
+        """
         cfg = self.config
         B = output.shape[0]
         ctx = self.nn.dropout(contexts_flat)
@@ -100,12 +102,14 @@ class AttentionDecoder(tnn.Module):
         if cfg.num_attend_layers == 1:
             l1 = self.att_fc_a(ctx).reshape(B, self.num_ctx)
             l2 = self.att_fc_b(out)
-            return l1 + l2
+            return ops.attention_pool(contexts, l1 + l2)
         t1 = self.att_fc_1a(ctx)                       # [B·L, A]
         t2 = self.att_fc_1b(out)                       # [B, A]
         t = t1 + t2.repeat_interleave(self.num_ctx, dim=0)
         t = self.nn.dropout(t)
-        return self.att_fc_2(t).reshape(B, self.num_ctx)
+        # fused tail: scores GEMV (fc_2, bias-free) + softmax + pool
+        v = self.att_fc_2.weight.reshape(-1).to(t.dtype)
+        return ops.attention_score_pool(t, v, contexts)
 
     def decode(self, expanded_output):
         """[B, H+D+E] -> logits [B, V]."""
@@ -134,8 +138,7 @@ class AttentionDecoder(tnn.Module):
         rate = self.nn.lstm_drop_rate
         training = self.nn.is_train
 
-        att = self.attend_logits(contexts_flat, last_output)
-        alpha, context = ops.attention_pool(contexts, att)
+        alpha, context = self.attend(contexts, contexts_flat, last_output)
 
         word_embed = ops.embedding(last_word, self.embedding.to(
             contexts.dtype))

@@ -66,14 +66,14 @@ class NN(object):
 
 
 class Dense(tnn.Module):
-    """tf.layers.dense analog: kernel in TF [in, out] layout, default tanh
-    (reference nn.py:85-105)."""
+    """tf.layers.dense analog (reference nn.py:85-105); kernel stored
+    [out, in] (torch layout) for transpose-free MFMA consumption."""
 
     def __init__(self, nn_policy, in_dim, units, activation='tanh',
                  use_bias=True):
         super().__init__()
         self.activation = activation
-        self.weight = tnn.Parameter(torch.empty(in_dim, units))
+        self.weight = tnn.Parameter(torch.empty(units, in_dim))
         nn_policy.init_fc_(self.weight)
         nn_policy.register_fc_kernel(self.weight)
         if use_bias:

@@ -20,7 +20,13 @@ from . import functional as F  # noqa: N812
 
 
 def _use_hip(*tensors):
-    if not any(t.is_cuda for t in tensors if isinstance(t, torch.Tensor)):
+    """GPU bf16 tensors -> HIP kernels (hard error if the extension is
+    missing); CPU -> reference path.  fp32 on GPU is the explicit
+    `compute_dtype='fp32'` debug knob and runs eager PyTorch-ROCm."""
+    primary = tensors[0]
+    if not primary.is_cuda:
+        return False
+    if primary.dtype != torch.bfloat16:
         return False
     from . import hip
     hip.require()  # raises if the extension is missing on a GPU box
@@ -28,7 +34,7 @@ def _use_hip(*tensors):
 
 
 def dense(x, weight, bias=None, activation=None):
-    """act(x @ weight + bias). x: [M,K], weight: [K,N] (TF layout)."""
+    """act(x @ weight.T + bias). x: [M,K], weight: [N,K] (torch layout)."""
     if _use_hip(x, weight):
         from . import hip
         return hip.dense(x, weight, bias, activation)
@@ -49,6 +55,16 @@ def attention_pool(contexts, logits):
         from . import hip
         return hip.attention_pool(contexts, logits)
     return F.attention_pool(contexts, logits)
+
+
+def attention_score_pool(temp_flat, v, contexts):
+    """Fused attention tail: scores GEMV + softmax over L + context sum.
+    Replaces the reference's N=1 GEMM (attend fc_2, model.py:429-434) +
+    softmax + weighted-sum trio with two fused kernels on GPU."""
+    if _use_hip(contexts, temp_flat):
+        from . import hip
+        return hip.attention_score_pool(temp_flat, v, contexts)
+    return F.attention_score_pool(temp_flat, v, contexts)
 
 
 def embedding(ids, table):

@@ -1,0 +1,37 @@
+// Python bindings for the sat_amd CDNA4 kernel layer (sat_amd._C).
+#include <torch/extension.h>
+#include <vector>
+
+at::Tensor dense_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, int64_t act);
+std::vector<at::Tensor> lstm_pointwise_fwd(at::Tensor gates, at::Tensor c,
+                                           double fb);
+std::vector<at::Tensor> lstm_pointwise_bwd(at::Tensor gates, at::Tensor c,
+                                           at::Tensor dh, at::Tensor dc,
+                                           double fb);
+at::Tensor attn_score_fwd(at::Tensor temp, at::Tensor v);
+std::vector<at::Tensor> attn_pool_fwd(at::Tensor ctx, at::Tensor logits);
+at::Tensor embedding_fwd(at::Tensor ids, at::Tensor table);
+at::Tensor embedding_bwd(at::Tensor ids, at::Tensor dy, int64_t rows);
+std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor labels,
+                               at::Tensor mask);
+at::Tensor ce_bwd(at::Tensor logits, at::Tensor labels, at::Tensor mask,
+                  at::Tensor lse, at::Tensor dloss);
+at::Tensor grad_sq_norm(std::vector<at::Tensor> grads);
+void adam_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
+               std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
+               double lr, double b1, double b2, double eps, int64_t step,
+               double clip, at::Tensor gsq);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("dense_fwd", &dense_fwd, "MFMA GEMM + bias/act (bf16)");
+    m.def("lstm_pointwise_fwd", &lstm_pointwise_fwd);
+    m.def("lstm_pointwise_bwd", &lstm_pointwise_bwd);
+    m.def("attn_score_fwd", &attn_score_fwd);
+    m.def("attn_pool_fwd", &attn_pool_fwd);
+    m.def("embedding_fwd", &embedding_fwd);
+    m.def("embedding_bwd", &embedding_bwd);
+    m.def("ce_fwd", &ce_fwd);
+    m.def("ce_bwd", &ce_bwd);
+    m.def("grad_sq_norm", &grad_sq_norm);
+    m.def("adam_step", &adam_step);
+}

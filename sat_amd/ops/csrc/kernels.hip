@@ -1,0 +1,439 @@
+// Fused pointwise / reduction kernels of the decoder hot path (gfx950).
+//
+// Covers the non-GEMM rows of the reference's implicit kernel surface
+// (SURVEY.md §2.3): LSTM gate fusion, the LDS-staged 196-location attention
+// softmax + weighted context sum, embedding gather/scatter-grad, fused
+// masked softmax cross-entropy, and the fused global-norm-clipped Adam.
+
+#include "common.h"
+
+// ======================================================================
+// LSTM gate fusion (TF LSTMCell semantics: gate order i,j,f,o; forget
+// bias added pre-sigmoid; no peepholes).  gates: [B,4H] from the GEMM.
+// ======================================================================
+
+__global__ void lstm_pw_fwd_kernel(const bf16* __restrict__ gates,
+                                   const bf16* __restrict__ c,
+                                   bf16* __restrict__ h_out,
+                                   bf16* __restrict__ c_out,
+                                   int B, int H, float fb) {
+    int idx = blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= B * H) return;
+    int b = idx / H, hh = idx % H;
+    const bf16* g = gates + (int64_t)b * 4 * H;
+    float gi = sigmoidf(bf2f(g[hh]));
+    float gj = tanhf(bf2f(g[H + hh]));
+    float gf = sigmoidf(bf2f(g[2 * H + hh]) + fb);
+    float go = sigmoidf(bf2f(g[3 * H + hh]));
+    float cn = bf2f(c[idx]) * gf + gi * gj;
+    h_out[idx] = f2bf(tanhf(cn) * go);
+    c_out[idx] = f2bf(cn);
+}
+
+__global__ void lstm_pw_bwd_kernel(const bf16* __restrict__ gates,
+                                   const bf16* __restrict__ c,
+                                   const bf16* __restrict__ dh,
+                                   const bf16* __restrict__ dc,
+                                   bf16* __restrict__ dgates,
+                                   bf16* __restrict__ dc_prev,
+                                   int B, int H, float fb) {
+    int idx = blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= B * H) return;
+    int b = idx / H, hh = idx % H;
+    const bf16* g = gates + (int64_t)b * 4 * H;
+    float gi = sigmoidf(bf2f(g[hh]));
+    float gj = tanhf(bf2f(g[H + hh]));
+    float gf = sigmoidf(bf2f(g[2 * H + hh]) + fb);
+    float go = sigmoidf(bf2f(g[3 * H + hh]));
+    float cp = bf2f(c[idx]);
+    float cn = cp * gf + gi * gj;
+    float tc = tanhf(cn);
+    float dhv = bf2f(dh[idx]);
+    float dcv = (dc != nullptr) ? bf2f(dc[idx]) : 0.f;
+    float dct = dcv + dhv * go * (1.f - tc * tc);
+    bf16* dg = dgates + (int64_t)b * 4 * H;
+    dg[hh]         = f2bf(dct * gj * gi * (1.f - gi));       // di
+    dg[H + hh]     = f2bf(dct * gi * (1.f - gj * gj));       // dj
+    dg[2 * H + hh] = f2bf(dct * cp * gf * (1.f - gf));       // df
+    dg[3 * H + hh] = f2bf(dhv * tc * go * (1.f - go));       // do
+    dc_prev[idx]   = f2bf(dct * gf);
+}
+
+std::vector<at::Tensor> lstm_pointwise_fwd(at::Tensor gates, at::Tensor c,
+                                           double fb) {
+    CHECK_GPU(gates); CHECK_CONTIG(gates); CHECK_BF16(gates);
+    CHECK_GPU(c); CHECK_CONTIG(c); CHECK_BF16(c);
+    int B = c.size(0), H = c.size(1);
+    TORCH_CHECK(gates.size(0) == B && gates.size(1) == 4 * H);
+    auto h_out = at::empty_like(c);
+    auto c_out = at::empty_like(c);
+    int n = B * H;
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(lstm_pw_fwd_kernel, dim3(cdiv(n, 256)), dim3(256), 0, s,
+                       (const bf16*)gates.data_ptr(), (const bf16*)c.data_ptr(),
+                       (bf16*)h_out.data_ptr(), (bf16*)c_out.data_ptr(),
+                       B, H, (float)fb);
+    HIP_OK(hipGetLastError());
+    return {h_out, c_out};
+}
+
+std::vector<at::Tensor> lstm_pointwise_bwd(at::Tensor gates, at::Tensor c,
+                                           at::Tensor dh, at::Tensor dc,
+                                           double fb) {
+    CHECK_GPU(gates); CHECK_CONTIG(gates); CHECK_BF16(gates);
+    int B = c.size(0), H = c.size(1);
+    auto dgates = at::empty_like(gates);
+    auto dc_prev = at::empty_like(c);
+    const bf16* dc_ptr = nullptr;
+    if (dc.defined() && dc.numel() > 0) dc_ptr = (const bf16*)dc.data_ptr();
+    int n = B * H;
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(lstm_pw_bwd_kernel, dim3(cdiv(n, 256)), dim3(256), 0, s,
+                       (const bf16*)gates.data_ptr(), (const bf16*)c.data_ptr(),
+                       (const bf16*)dh.data_ptr(), dc_ptr,
+                       (bf16*)dgates.data_ptr(), (bf16*)dc_prev.data_ptr(),
+                       B, H, (float)fb);
+    HIP_OK(hipGetLastError());
+    return {dgates, dc_prev};
+}
+
+// ======================================================================
+// Attention: scores GEMV (temp[M,A] @ v[A] -> logits[M]) and the fused
+// LDS-staged softmax-over-L + weighted context sum (model.py:435,263-264).
+// ======================================================================
+
+__global__ void attn_score_kernel(const bf16* __restrict__ temp,
+                                  const bf16* __restrict__ v,
+                                  float* __restrict__ logits,
+                                  int M, int A) {
+    // 4 waves per block, one row per wave
+    int row = blockIdx.x * 4 + (threadIdx.x >> 6);
+    if (row >= M) return;
+    int lane = threadIdx.x & 63;
+    const bf16* t = temp + (int64_t)row * A;
+    float acc = 0.f;
+    // A % 8 == 0: 8-element chunks tile A exactly; chunk c = lane + 64k
+    for (int a0 = lane * 8; a0 + 8 <= A; a0 += 64 * 8) {
+        bf16x8 tv = *(const bf16x8*)(t + a0);
+        bf16x8 vv = *(const bf16x8*)(v + a0);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) acc += bf2f(tv[e]) * bf2f(vv[e]);
+    }
+    acc = wave_sum(acc);
+    if (lane == 0) logits[row] = acc;
+}
+
+at::Tensor attn_score_fwd(at::Tensor temp, at::Tensor v) {
+    CHECK_GPU(temp); CHECK_CONTIG(temp); CHECK_BF16(temp);
+    CHECK_GPU(v); CHECK_CONTIG(v); CHECK_BF16(v);
+    int M = temp.size(0), A = temp.size(1);
+    TORCH_CHECK(A % 8 == 0, "A must be a multiple of 8 (got ", A, ")");
+    auto logits = at::empty({M}, temp.options().dtype(at::kFloat));
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(attn_score_kernel, dim3(cdiv(M, 4)), dim3(256), 0, s,
+                       (const bf16*)temp.data_ptr(), (const bf16*)v.data_ptr(),
+                       (float*)logits.data_ptr(), M, A);
+    HIP_OK(hipGetLastError());
+    return logits;
+}
+
+#define MAX_L 1024
+
+__global__ void attn_pool_kernel(const bf16* __restrict__ ctx,   // [B,L,D]
+                                 const float* __restrict__ logits, // [B,L]
+                                 float* __restrict__ alpha,       // [B,L]
+                                 bf16* __restrict__ pooled,       // [B,D]
+                                 int L, int D) {
+    __shared__ float sa[MAX_L];
+    __shared__ float red[8];
+    int b = blockIdx.x;
+    int tid = threadIdx.x;
+
+    // softmax over L (L <= MAX_L), staged in LDS
+    float lmax = -1e30f;
+    for (int l = tid; l < L; l += blockDim.x) {
+        float x = logits[(int64_t)b * L + l];
+        sa[l] = x;
+        lmax = fmaxf(lmax, x);
+    }
+    lmax = wave_max(lmax);
+    if ((tid & 63) == 0) red[tid >> 6] = lmax;
+    __syncthreads();
+    float m = -1e30f;
+    for (int w = 0; w < (int)(blockDim.x >> 6); ++w) m = fmaxf(m, red[w]);
+    __syncthreads();
+
+    float lsum = 0.f;
+    for (int l = tid; l < L; l += blockDim.x) {
+        float e = __expf(sa[l] - m);
+        sa[l] = e;
+        lsum += e;
+    }
+    lsum = wave_sum(lsum);
+    if ((tid & 63) == 0) red[tid >> 6] = lsum;
+    __syncthreads();
+    float z = 0.f;
+    for (int w = 0; w < (int)(blockDim.x >> 6); ++w) z += red[w];
+    float inv = 1.0f / z;
+    __syncthreads();
+    for (int l = tid; l < L; l += blockDim.x) {
+        sa[l] *= inv;
+        alpha[(int64_t)b * L + l] = sa[l];
+    }
+    __syncthreads();
+
+    // pooled[d] = sum_l alpha[l] * ctx[l,d]; threads own columns
+    const bf16* cb = ctx + (int64_t)b * L * D;
+    for (int d = tid; d < D; d += blockDim.x) {
+        float acc = 0.f;
+        for (int l = 0; l < L; ++l)
+            acc += sa[l] * bf2f(cb[(int64_t)l * D + d]);
+        pooled[(int64_t)b * D + d] = f2bf(acc);
+    }
+}
+
+std::vector<at::Tensor> attn_pool_fwd(at::Tensor ctx, at::Tensor logits) {
+    CHECK_GPU(ctx); CHECK_CONTIG(ctx); CHECK_BF16(ctx);
+    CHECK_GPU(logits); CHECK_CONTIG(logits); CHECK_F32(logits);
+    int B = ctx.size(0), L = ctx.size(1), D = ctx.size(2);
+    TORCH_CHECK(L <= MAX_L, "attention locations exceed LDS stage");
+    auto alpha = at::empty({B, L}, ctx.options().dtype(at::kFloat));
+    auto pooled = at::empty({B, D}, ctx.options());
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(attn_pool_kernel, dim3(B), dim3(256), 0, s,
+                       (const bf16*)ctx.data_ptr(),
+                       (const float*)logits.data_ptr(),
+                       (float*)alpha.data_ptr(), (bf16*)pooled.data_ptr(),
+                       L, D);
+    HIP_OK(hipGetLastError());
+    return {alpha, pooled};
+}
+
+// ======================================================================
+// Embedding gather / scatter-add grad (model.py:273; table [V,E] bf16).
+// ======================================================================
+
+__global__ void embedding_fwd_kernel(const int64_t* __restrict__ ids,
+                                     const bf16* __restrict__ table,
+                                     bf16* __restrict__ out,
+                                     int B, int E) {
+    int idx = blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= B * E) return;
+    int b = idx / E, e = idx % E;
+    out[idx] = table[ids[b] * E + e];
+}
+
+__global__ void embedding_bwd_kernel(const int64_t* __restrict__ ids,
+                                     const bf16* __restrict__ dy,
+                                     float* __restrict__ dtable,
+                                     int B, int E) {
+    int idx = blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= B * E) return;
+    int b = idx / E, e = idx % E;
+    atomicAdd(dtable + ids[b] * E + e, bf2f(dy[idx]));
+}
+
+at::Tensor embedding_fwd(at::Tensor ids, at::Tensor table) {
+    CHECK_GPU(ids); CHECK_CONTIG(ids);
+    CHECK_GPU(table); CHECK_CONTIG(table); CHECK_BF16(table);
+    TORCH_CHECK(ids.scalar_type() == at::kLong, "ids must be int64");
+    int B = ids.numel(), E = table.size(1);
+    auto out = at::empty({B, E}, table.options());
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(embedding_fwd_kernel, dim3(cdiv(B * E, 256)),
+                       dim3(256), 0, s,
+                       (const int64_t*)ids.data_ptr(),
+                       (const bf16*)table.data_ptr(),
+                       (bf16*)out.data_ptr(), B, E);
+    HIP_OK(hipGetLastError());
+    return out;
+}
+
+at::Tensor embedding_bwd(at::Tensor ids, at::Tensor dy, int64_t rows) {
+    CHECK_GPU(ids); CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_BF16(dy);
+    int B = ids.numel(), E = dy.size(1);
+    auto dtable = at::zeros({rows, E}, dy.options().dtype(at::kFloat));
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(embedding_bwd_kernel, dim3(cdiv(B * E, 256)),
+                       dim3(256), 0, s,
+                       (const int64_t*)ids.data_ptr(),
+                       (const bf16*)dy.data_ptr(),
+                       (float*)dtable.data_ptr(), B, E);
+    HIP_OK(hipGetLastError());
+    return dtable;
+}
+
+// ======================================================================
+// Fused masked softmax cross-entropy over the vocabulary (model.py:294-297)
+// ======================================================================
+
+__global__ void ce_fwd_kernel(const bf16* __restrict__ logits, // [B,V]
+                              const int64_t* __restrict__ labels,
+                              const float* __restrict__ mask,
+                              float* __restrict__ losses,
+                              float* __restrict__ lse,
+                              int V) {
+    __shared__ float red[8];
+    int b = blockIdx.x;
+    int tid = threadIdx.x;
+    const bf16* row = logits + (int64_t)b * V;
+
+    float lmax = -1e30f;
+    for (int v = tid; v < V; v += blockDim.x)
+        lmax = fmaxf(lmax, bf2f(row[v]));
+    lmax = wave_max(lmax);
+    if ((tid & 63) == 0) red[tid >> 6] = lmax;
+    __syncthreads();
+    float m = -1e30f;
+    for (int w = 0; w < (int)(blockDim.x >> 6); ++w) m = fmaxf(m, red[w]);
+    __syncthreads();
+
+    float lsum = 0.f;
+    for (int v = tid; v < V; v += blockDim.x)
+        lsum += __expf(bf2f(row[v]) - m);
+    lsum = wave_sum(lsum);
+    if ((tid & 63) == 0) red[tid >> 6] = lsum;
+    __syncthreads();
+    if (tid == 0) {
+        float z = 0.f;
+        for (int w = 0; w < (int)(blockDim.x >> 6); ++w) z += red[w];
+        float l = m + __logf(z);
+        lse[b] = l;
+        losses[b] = (l - bf2f(row[labels[b]])) * mask[b];
+    }
+}
+
+__global__ void ce_bwd_kernel(const bf16* __restrict__ logits,
+                              const int64_t* __restrict__ labels,
+                              const float* __restrict__ mask,
+                              const float* __restrict__ lse,
+                              const float* __restrict__ dloss,
+                              bf16* __restrict__ dlogits,
+                              int B, int V) {
+    int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= (int64_t)B * V) return;
+    int b = idx / V, v = idx % V;
+    float p = __expf(bf2f(logits[idx]) - lse[b]);
+    float g = (p - (labels[b] == v ? 1.f : 0.f)) * mask[b] * dloss[b];
+    dlogits[idx] = f2bf(g);
+}
+
+std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor labels,
+                               at::Tensor mask) {
+    CHECK_GPU(logits); CHECK_CONTIG(logits); CHECK_BF16(logits);
+    CHECK_GPU(labels); CHECK_GPU(mask); CHECK_F32(mask);
+    int B = logits.size(0), V = logits.size(1);
+    auto losses = at::empty({B}, logits.options().dtype(at::kFloat));
+    auto lse = at::empty({B}, logits.options().dtype(at::kFloat));
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(ce_fwd_kernel, dim3(B), dim3(256), 0, s,
+                       (const bf16*)logits.data_ptr(),
+                       (const int64_t*)labels.data_ptr(),
+                       (const float*)mask.data_ptr(),
+                       (float*)losses.data_ptr(), (float*)lse.data_ptr(), V);
+    HIP_OK(hipGetLastError());
+    return {losses, lse};
+}
+
+at::Tensor ce_bwd(at::Tensor logits, at::Tensor labels, at::Tensor mask,
+                  at::Tensor lse, at::Tensor dloss) {
+    CHECK_GPU(logits); CHECK_CONTIG(logits); CHECK_BF16(logits);
+    int B = logits.size(0), V = logits.size(1);
+    auto dlogits = at::empty_like(logits);
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(ce_bwd_kernel,
+                       dim3(cdiv((int64_t)B * V, 256)), dim3(256), 0, s,
+                       (const bf16*)logits.data_ptr(),
+                       (const int64_t*)labels.data_ptr(),
+                       (const float*)mask.data_ptr(),
+                       (const float*)lse.data_ptr(),
+                       (const float*)dloss.data_ptr(),
+                       (bf16*)dlogits.data_ptr(), B, V);
+    HIP_OK(hipGetLastError());
+    return dlogits;
+}
+
+// ======================================================================
+// Fused global-norm-clipped Adam (model.py:461-513 + optimize_loss clip).
+// Two phases: grad_sq_norm accumulates Σ‖g‖² across all tensors into one
+// device scalar; adam_step applies clip scale + bias-corrected Adam.
+// ======================================================================
+
+__global__ void sq_norm_kernel(const float* __restrict__ g, int64_t n,
+                               float* __restrict__ out) {
+    int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    float acc = 0.f;
+    for (int64_t i = idx; i < n; i += (int64_t)gridDim.x * blockDim.x)
+        acc += g[i] * g[i];
+    acc = wave_sum(acc);
+    __shared__ float red[8];
+    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = acc;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        float t = 0.f;
+        for (int w = 0; w < (int)(blockDim.x >> 6); ++w) t += red[w];
+        atomicAdd(out, t);
+    }
+}
+
+__global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
+                            float* __restrict__ m, float* __restrict__ v,
+                            const float* __restrict__ gsq,
+                            int64_t n, float lr, float b1, float b2,
+                            float eps, float bc1, float bc2, float clip) {
+    float scale = 1.f;
+    if (clip > 0.f) {
+        float norm = sqrtf(*gsq);
+        if (norm > clip) scale = clip / norm;
+    }
+    int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    for (int64_t i = idx; i < n; i += (int64_t)gridDim.x * blockDim.x) {
+        float gi = g[i] * scale;
+        float mi = b1 * m[i] + (1.f - b1) * gi;
+        float vi = b2 * v[i] + (1.f - b2) * gi * gi;
+        m[i] = mi;
+        v[i] = vi;
+        float mhat = mi / bc1;
+        float vhat = vi / bc2;
+        p[i] -= lr * mhat / (sqrtf(vhat) + eps);
+    }
+}
+
+at::Tensor grad_sq_norm(std::vector<at::Tensor> grads) {
+    TORCH_CHECK(!grads.empty());
+    auto out = at::zeros({1}, grads[0].options().dtype(at::kFloat));
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    for (auto& g : grads) {
+        CHECK_GPU(g); CHECK_CONTIG(g); CHECK_F32(g);
+        int64_t n = g.numel();
+        int blocks = (int)std::min<int64_t>(cdiv(n, 256), 1024);
+        hipLaunchKernelGGL(sq_norm_kernel, dim3(blocks), dim3(256), 0, s,
+                           (const float*)g.data_ptr(), n,
+                           (float*)out.data_ptr());
+    }
+    HIP_OK(hipGetLastError());
+    return out;
+}
+
+void adam_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
+               std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
+               double lr, double b1, double b2, double eps, int64_t step,
+               double clip, at::Tensor gsq) {
+    float bc1 = 1.f - powf((float)b1, (float)step);
+    float bc2 = 1.f - powf((float)b2, (float)step);
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    for (size_t i = 0; i < params.size(); ++i) {
+        auto& p = params[i];
+        CHECK_GPU(p); CHECK_CONTIG(p); CHECK_F32(p);
+        int64_t n = p.numel();
+        int blocks = (int)std::min<int64_t>(cdiv(n, 256), 2048);
+        hipLaunchKernelGGL(adam_kernel, dim3(blocks), dim3(256), 0, s,
+                           (float*)p.data_ptr(),
+                           (const float*)grads[i].data_ptr(),
+                           (float*)ms[i].data_ptr(), (float*)vs[i].data_ptr(),
+                           (const float*)gsq.data_ptr(), n,
+                           (float)lr, (float)b1, (float)b2, (float)eps,
+                           bc1, bc2, (float)clip);
+    }
+    HIP_OK(hipGetLastError());
+}

@@ -3,7 +3,9 @@
 These are the CPU path and the numerics ground truth for the HIP kernels.
 Semantics mirror the TF ops the reference launches:
 
-  * dense           — tf.layers.dense (x @ W + b, TF [in,out] kernel layout)
+  * dense           — tf.layers.dense semantics; kernels are stored [out,in]
+                      (torch Linear layout) so the GPU MFMA kernel reads a
+                      [N,K]-contiguous B operand with no transpose
   * lstm_cell       — tf.nn.rnn_cell.LSTMCell (gate order i,j,f,o; forget
                       bias added pre-sigmoid; no peepholes)
   * attention_pool  — softmax over locations + Σ_l α_l·ctx_l
@@ -18,7 +20,8 @@ import torch.nn.functional as tF
 
 
 def dense(x, weight, bias=None, activation=None):
-    y = x.matmul(weight)
+    """act(x @ weight.T + bias); weight: [out, in]."""
+    y = x.matmul(weight.t())
     if bias is not None:
         y = y + bias
     if activation == 'tanh':
@@ -33,10 +36,10 @@ def dense(x, weight, bias=None, activation=None):
 def lstm_cell(x, h, c, weight, bias, forget_bias=1.0):
     """TF LSTMCell step.
 
-    x: [B, I], h/c: [B, H], weight: [I+H, 4H] with gate order (i, j, f, o),
+    x: [B, I], h/c: [B, H], weight: [4H, I+H] with gate order (i, j, f, o),
     bias: [4H].  Returns (new_h, new_c).
     """
-    gates = torch.cat([x, h], dim=1).matmul(weight) + bias
+    gates = torch.cat([x, h], dim=1).matmul(weight.t()) + bias
     i, j, f, o = gates.chunk(4, dim=1)
     new_c = c * torch.sigmoid(f + forget_bias) + \
         torch.sigmoid(i) * torch.tanh(j)
@@ -49,6 +52,15 @@ def attention_pool(contexts, logits):
     alpha = torch.softmax(logits, dim=1)
     context = (contexts * alpha.unsqueeze(2)).sum(dim=1)
     return alpha, context
+
+
+def attention_score_pool(temp_flat, v, contexts):
+    """Fused attention tail (reference model.py:429-435 + :263-264):
+    logits = temp_flat[B·L,A] @ v[A], alpha = softmax over L, context =
+    Σ_l α_l·ctx_l.  Returns (alpha [B,L], context [B,D])."""
+    B, L = contexts.shape[0], contexts.shape[1]
+    logits = temp_flat.matmul(v).reshape(B, L)
+    return attention_pool(contexts, logits)
 
 
 def embedding(ids, table):

@@ -1,17 +1,18 @@
 """HIP kernel path: autograd wrappers over the sat_amd._C extension.
 
-The extension (built by setup.py / __graft_entry__.build() from
+The extension (built in-tree by setup.py / __graft_entry__.build() from
 sat_amd/ops/csrc/*.hip, gfx950-only) provides:
 
-  dense_fwd(x, w, bias, act)            MFMA GEMM + fused bias/activation
-  lstm_pointwise_fwd / _bwd             fused LSTM gate math (i,j,f,o)
-  attn_softmax_ctx_fwd / _bwd           LDS-staged softmax over L + ctx-sum
-  embedding_fwd / embedding_bwd         gather / scatter-add
-  ce_fwd / ce_bwd                       fused masked softmax cross-entropy
-  grad_sq_norm / adam_step              fused global-norm clip + Adam
+  dense_fwd                       MFMA GEMM + fused bias/activation (bf16)
+  lstm_pointwise_fwd / _bwd       fused LSTM gate math (i,j,f,o; recompute bwd)
+  attn_score_fwd                  scores GEMV temp[M,A]·v[A] (no N=1 GEMM)
+  attn_pool_fwd                   LDS-staged softmax over L + ctx-weighted sum
+  embedding_fwd / embedding_bwd   gather / scatter-add
+  ce_fwd / ce_bwd                 fused masked softmax cross-entropy
+  grad_sq_norm / adam_step        fused global-norm clip + Adam
 
-Backward GEMMs (dX = dY Wᵀ, dW = Xᵀ dY) go through torch.matmul, i.e.
-hipBLASLt — plain library GEMMs, per the MI355X design rules; everything
+Backward GEMMs (dX = dY·W, dW = dYᵀ·X) go through torch.matmul, i.e.
+hipBLASLt — plain library GEMMs per the MI355X design rules; everything
 fused is hand-written CDNA4.
 
 On a GPU box a missing extension is a hard error (`require()`): the HIP path
@@ -42,12 +43,21 @@ def require():
 
 
 _ACT = {None: 0, 'none': 0, 'tanh': 1, 'relu': 2}
+_EMPTY = {}
+
+
+def _empty(device):
+    key = str(device)
+    if key not in _EMPTY:
+        _EMPTY[key] = torch.empty(0, device=device, dtype=torch.bfloat16)
+    return _EMPTY[key]
 
 
 class _Dense(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, b, act):
-        y = _C.dense_fwd(x, w, b if b is not None else torch.Tensor(),
+        y = _C.dense_fwd(x.contiguous(), w.contiguous(),
+                         b.contiguous() if b is not None else _empty(x.device),
                          _ACT[act])
         ctx.save_for_backward(x, w, y)
         ctx.act = act
@@ -58,36 +68,42 @@ class _Dense(torch.autograd.Function):
     def backward(ctx, dy):
         x, w, y = ctx.saved_tensors
         if ctx.act == 'tanh':
-            dpre = dy * (1 - y.float() * y.float()).to(dy.dtype)
+            yf = y.float()
+            dpre = (dy.float() * (1 - yf * yf)).to(dy.dtype)
         elif ctx.act == 'relu':
             dpre = dy * (y > 0).to(dy.dtype)
         else:
             dpre = dy
-        dx = dpre.matmul(w.t()) if ctx.needs_input_grad[0] else None
-        dw = x.t().matmul(dpre) if ctx.needs_input_grad[1] else None
+        dx = dpre.matmul(w) if ctx.needs_input_grad[0] else None
+        dw = dpre.t().matmul(x) if ctx.needs_input_grad[1] else None
         db = dpre.sum(0) if (ctx.has_bias and ctx.needs_input_grad[2]) \
             else None
         return dx, dw, db, None
 
 
 def dense(x, weight, bias=None, activation=None):
+    """act(x @ weightᵀ + bias); weight stored [out, in]."""
     return _Dense.apply(x, weight, bias, activation)
 
 
 class _LSTMPointwise(torch.autograd.Function):
-    """gates [B,4H] + c [B,H] -> (h', c'); saves sigmoid/tanh activations."""
+    """gates [B,4H] + c [B,H] -> (h', c'); backward recomputes activations
+    from the saved gates (cheap elementwise; avoids 4 extra saved tensors)."""
 
     @staticmethod
     def forward(ctx, gates, c, forget_bias):
-        h_new, c_new, saved = _C.lstm_pointwise_fwd(gates, c, forget_bias)
-        ctx.save_for_backward(c, c_new, saved)
+        h_new, c_new = _C.lstm_pointwise_fwd(gates, c, forget_bias)
+        ctx.save_for_backward(gates, c)
+        ctx.fb = forget_bias
         return h_new, c_new
 
     @staticmethod
     def backward(ctx, dh, dc):
-        c, c_new, saved = ctx.saved_tensors
+        gates, c = ctx.saved_tensors
+        dc_arg = dc.contiguous() if dc is not None else \
+            torch.empty(0, device=gates.device, dtype=gates.dtype)
         dgates, dc_prev = _C.lstm_pointwise_bwd(
-            dh.contiguous(), dc.contiguous(), c, c_new, saved)
+            gates, c, dh.contiguous(), dc_arg, ctx.fb)
         return dgates, dc_prev, None
 
 
@@ -97,23 +113,66 @@ def lstm_cell(x, h, c, weight, bias, forget_bias=1.0):
     return _LSTMPointwise.apply(gates, c, forget_bias)
 
 
-class _AttnSoftmaxCtx(torch.autograd.Function):
+class _AttnPool(torch.autograd.Function):
+    """(contexts [B,L,D] bf16, logits [B,L] fp32) -> (alpha fp32, pooled
+    bf16).  Forward is the fused LDS-staged kernel; backward composes
+    hipBLASLt bmm + eager softmax-grad (library GEMMs, not hot)."""
+
     @staticmethod
     def forward(ctx, contexts, logits):
-        alpha, pooled = _C.attn_softmax_ctx_fwd(contexts, logits)
+        alpha, pooled = _C.attn_pool_fwd(contexts, logits.contiguous())
         ctx.save_for_backward(contexts, alpha)
         return alpha, pooled
 
     @staticmethod
     def backward(ctx, dalpha, dpooled):
         contexts, alpha = ctx.saved_tensors
-        dctx, dlogits = _C.attn_softmax_ctx_bwd(
-            contexts, alpha, dalpha.contiguous(), dpooled.contiguous())
+        # d(alpha)/d(logits) softmax backward + pooled path
+        da = torch.bmm(contexts.float(),
+                       dpooled.float().unsqueeze(2)).squeeze(2)  # [B,L]
+        if dalpha is not None:
+            da = da + dalpha
+        dlogits = alpha * (da - (alpha * da).sum(dim=1, keepdim=True))
+        dctx = None
+        if ctx.needs_input_grad[0]:
+            dctx = (alpha.unsqueeze(2) * dpooled.float().unsqueeze(1)) \
+                .to(contexts.dtype)
         return dctx, dlogits
 
 
+class _AttnScore(torch.autograd.Function):
+    """temp [M,A] bf16 @ v [A] bf16 -> logits fp32 [M] (GEMV kernel)."""
+
+    @staticmethod
+    def forward(ctx, temp, v):
+        logits = _C.attn_score_fwd(temp.contiguous(), v.contiguous())
+        ctx.save_for_backward(temp, v)
+        return logits
+
+    @staticmethod
+    def backward(ctx, dlogits):
+        temp, v = ctx.saved_tensors
+        dl = dlogits.contiguous()
+        dtemp = None
+        dv = None
+        if ctx.needs_input_grad[0]:
+            dtemp = (dl.unsqueeze(1) * v.float().unsqueeze(0)) \
+                .to(temp.dtype)
+        if ctx.needs_input_grad[1]:
+            dv = temp.float().t().matmul(dl).to(v.dtype)
+        return dtemp, dv
+
+
+def attention_score_pool(temp_flat, v, contexts):
+    """Fused attention tail: logits = temp·v, alpha = softmax_L, context =
+    Σ_l α_l ctx_l.  Returns (alpha [B,L] fp32, pooled [B,D])."""
+    logits = _AttnScore.apply(temp_flat, v).reshape(contexts.shape[0],
+                                                    contexts.shape[1])
+    return _AttnPool.apply(contexts, logits)
+
+
 def attention_pool(contexts, logits):
-    return _AttnSoftmaxCtx.apply(contexts, logits)
+    return _AttnPool.apply(contexts, logits.float())
 
 
 class _Embedding(torch.autograd.Function):
@@ -138,31 +197,33 @@ def embedding(ids, table):
 class _MaskedCE(torch.autograd.Function):
     @staticmethod
     def forward(ctx, logits, labels, mask):
-        losses, lse = _C.ce_fwd(logits, labels, mask)
+        losses, lse = _C.ce_fwd(logits, labels, mask.contiguous())
         ctx.save_for_backward(logits, labels, mask, lse)
         return losses
 
     @staticmethod
     def backward(ctx, dloss):
         logits, labels, mask, lse = ctx.saved_tensors
-        dlogits = _C.ce_bwd(logits, labels, mask, lse, dloss.contiguous())
+        dlogits = _C.ce_bwd(logits, labels, mask, lse,
+                            dloss.contiguous().float())
         return dlogits, None, None
 
 
 def masked_softmax_ce(logits, labels, mask):
-    return _MaskedCE.apply(logits, labels, mask)
+    return _MaskedCE.apply(logits.contiguous(), labels, mask)
 
 
 # ---- fused optimizer (no autograd; called by sat_amd.optim) ----
 
 def grad_sq_norm(grads):
-    return _C.grad_sq_norm(list(grads))
+    return _C.grad_sq_norm([g.contiguous() for g in grads])
 
 
 def adam_step(params, grads, ms, vs, lr, beta1, beta2, eps, step,
               clip, grad_sq):
     """Fused multi-tensor Adam. `grad_sq` is the on-device Σ‖g‖² scalar from
     grad_sq_norm; the kernel derives scale = clip/max(clip, √grad_sq)."""
-    _C.adam_step(list(params), list(grads), list(ms), list(vs),
+    _C.adam_step(list(params), [g.contiguous() for g in grads],
+                 list(ms), list(vs),
                  float(lr), float(beta1), float(beta2), float(eps),
                  int(step), float(clip), grad_sq)

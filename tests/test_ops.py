@@ -13,17 +13,17 @@ from sat_amd.ops import functional as F
 def test_dense_matches_manual():
     torch.manual_seed(0)
     x = torch.randn(4, 5)
-    w = torch.randn(5, 3)
+    w = torch.randn(3, 5)
     b = torch.randn(3)
     y = F.dense(x, w, b, 'tanh')
-    ref = torch.tanh(x @ w + b)
+    ref = torch.tanh(x @ w.t() + b)
     assert torch.allclose(y, ref)
 
 
 def test_dense_no_activation_no_bias():
     x = torch.randn(4, 5)
-    w = torch.randn(5, 3)
-    assert torch.allclose(F.dense(x, w), x @ w)
+    w = torch.randn(3, 5)
+    assert torch.allclose(F.dense(x, w), x @ w.t())
 
 
 def test_lstm_cell_matches_tf_semantics():
@@ -33,12 +33,12 @@ def test_lstm_cell_matches_tf_semantics():
     x = torch.randn(B, I)
     h = torch.randn(B, H)
     c = torch.randn(B, H)
-    w = torch.randn(I + H, 4 * H)
+    w = torch.randn(4 * H, I + H)
     b = torch.randn(4 * H)
 
     new_h, new_c = F.lstm_cell(x, h, c, w, b, forget_bias=1.0)
 
-    g = torch.cat([x, h], 1) @ w + b
+    g = torch.cat([x, h], 1) @ w.t() + b
     i, j, f, o = g[:, :H], g[:, H:2*H], g[:, 2*H:3*H], g[:, 3*H:]
     exp_c = c * torch.sigmoid(f + 1.0) + torch.sigmoid(i) * torch.tanh(j)
     exp_h = torch.tanh(exp_c) * torch.sigmoid(o)
@@ -63,7 +63,7 @@ def test_lstm_cell_matches_torch_lstmcell():
     def tf_order(m):
         i, f, g, o = m.chunk(4, dim=0)
         return torch.cat([i, g, f, o], dim=0)
-    w = torch.cat([tf_order(wi), tf_order(wh)], dim=1).t().contiguous()
+    w = torch.cat([tf_order(wi), tf_order(wh)], dim=1).contiguous()
     b = tf_order(bias.unsqueeze(1)).squeeze(1)
     new_h, new_c = F.lstm_cell(x, h, c, w, b, forget_bias=0.0)
     assert torch.allclose(new_h, ht, atol=1e-5)

@@ -1,0 +1,216 @@
+"""GPU numerics tests: every HIP kernel vs the plain-PyTorch fp32 reference
+(sat_amd.ops.functional).  bf16 kernels are compared against the fp32
+reference computed on bf16-rounded inputs, with tolerances sized for one
+bf16 rounding on inputs + fp32 accumulation in the kernel."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from sat_amd.ops import functional as F
+    from sat_amd.ops import hip
+
+DEV = 'cuda'
+
+
+def _bf(x):
+    return x.to(DEV, torch.bfloat16)
+
+
+def _rel_err(a, b):
+    denom = b.float().abs().max().clamp_min(1e-6)
+    return ((a.float() - b.float()).abs().max() / denom).item()
+
+
+@pytest.mark.parametrize('M,N,K', [
+    (128, 128, 64),      # single tile
+    (6272, 512, 512),    # attention projection (B=32, L=196)
+    (32, 2048, 1536),    # LSTM gates
+    (32, 5000, 1024),    # decode logits (N edge: 5000 % 128 != 0)
+    (100, 130, 72),      # everything-edge
+    (1, 16, 8),          # degenerate
+])
+@pytest.mark.parametrize('act', [None, 'tanh', 'relu'])
+def test_dense_fwd(M, N, K, act):
+    torch.manual_seed(0)
+    x = _bf(torch.randn(M, K))
+    w = _bf(torch.randn(N, K) * 0.05)
+    b = _bf(torch.randn(N))
+    y = hip.dense(x, w, b, act)
+    ref = F.dense(x.float(), w.float(), b.float(), act)
+    assert y.shape == (M, N)
+    assert _rel_err(y, ref) < 2e-2
+
+
+def test_dense_asymmetric_layout():
+    """Catches transposed C-write / swapped operands (asymmetric B)."""
+    M, N, K = 64, 48, 32
+    x = torch.zeros(M, K)
+    x[3, 5] = 1.0
+    w = torch.zeros(N, K)
+    w[7, 5] = 2.0
+    w[7, 6] = 99.0  # must not contribute
+    y = hip.dense(_bf(x), _bf(w), None, None).float().cpu()
+    assert abs(y[3, 7].item() - 2.0) < 1e-3
+    assert y.abs().sum().item() == pytest.approx(2.0, abs=1e-2)
+
+
+def test_dense_backward():
+    torch.manual_seed(1)
+    M, N, K = 96, 64, 40
+    x = _bf(torch.randn(M, K)).requires_grad_(True)
+    w = _bf(torch.randn(N, K) * 0.1).requires_grad_(True)
+    b = _bf(torch.randn(N)).requires_grad_(True)
+    y = hip.dense(x, w, b, 'tanh')
+    g = torch.randn_like(y)
+    y.backward(g)
+
+    xr = x.detach().float().requires_grad_(True)
+    wr = w.detach().float().requires_grad_(True)
+    br = b.detach().float().requires_grad_(True)
+    F.dense(xr, wr, br, 'tanh').backward(g.float())
+    assert _rel_err(x.grad, xr.grad) < 5e-2
+    assert _rel_err(w.grad, wr.grad) < 5e-2
+    assert _rel_err(b.grad, br.grad) < 5e-2
+
+
+def test_lstm_cell_fwd_bwd():
+    torch.manual_seed(2)
+    B, I, H = 32, 1024, 512
+    x = _bf(torch.randn(B, I)).requires_grad_(True)
+    h = _bf(torch.randn(B, H)).requires_grad_(True)
+    c = _bf(torch.randn(B, H)).requires_grad_(True)
+    w = _bf(torch.randn(4 * H, I + H) * 0.02).requires_grad_(True)
+    b = _bf(torch.zeros(4 * H)).requires_grad_(True)
+
+    nh, nc = hip.lstm_cell(x, h, c, w, b)
+    refs = [t.detach().float().requires_grad_(True)
+            for t in (x, h, c, w, b)]
+    rh, rc = F.lstm_cell(*refs)
+    assert _rel_err(nh, rh) < 2e-2
+    assert _rel_err(nc, rc) < 2e-2
+
+    gh = torch.randn_like(nh)
+    gc = torch.randn_like(nc)
+    (nh.float() * gh.float()).sum().backward(retain_graph=True)
+    (nc.float() * gc.float()).sum().backward()
+    (rh * gh.float()).sum().backward(retain_graph=True)
+    (rc * gc.float()).sum().backward()
+    for t, r in zip((x, h, c, w), refs[:4]):
+        assert _rel_err(t.grad, r.grad) < 5e-2
+
+
+def test_attention_score_pool_fwd_bwd():
+    torch.manual_seed(3)
+    B, L, A, D = 32, 196, 512, 512
+    temp = _bf(torch.randn(B * L, A)).requires_grad_(True)
+    v = _bf(torch.randn(A) * 0.05).requires_grad_(True)
+    ctx = _bf(torch.randn(B, L, D)).requires_grad_(True)
+
+    alpha, pooled = hip.attention_score_pool(temp, v, ctx)
+    tr, vr, cr = [t.detach().float().requires_grad_(True)
+                  for t in (temp, v, ctx)]
+    ra, rp = F.attention_score_pool(tr, vr, cr)
+    assert alpha.shape == (B, L)
+    assert torch.allclose(alpha.sum(1),
+                          torch.ones(B, device=DEV), atol=1e-4)
+    assert _rel_err(alpha, ra) < 2e-2
+    assert _rel_err(pooled, rp) < 2e-2
+
+    ga = torch.randn_like(alpha)
+    gp = torch.randn_like(pooled)
+    (alpha * ga).sum().backward(retain_graph=True)
+    (pooled.float() * gp.float()).sum().backward()
+    (ra * ga).sum().backward(retain_graph=True)
+    (rp * gp.float()).sum().backward()
+    assert _rel_err(temp.grad, tr.grad) < 5e-2
+    assert _rel_err(v.grad, vr.grad) < 5e-2
+    assert _rel_err(ctx.grad, cr.grad) < 5e-2
+
+
+def test_attention_pool_small_l():
+    """ResNet50 grid: L=49."""
+    B, L, D = 4, 49, 2048
+    ctx = _bf(torch.randn(B, L, D))
+    logits = torch.randn(B, L, device=DEV)
+    alpha, pooled = hip.attention_pool(ctx, logits)
+    ra, rp = F.attention_pool(ctx.float(), logits)
+    assert _rel_err(alpha, ra) < 1e-2
+    assert _rel_err(pooled, rp) < 2e-2
+
+
+def test_embedding_fwd_bwd():
+    torch.manual_seed(4)
+    V, E, B = 5000, 512, 64
+    table = _bf(torch.randn(V, E)).requires_grad_(True)
+    ids = torch.randint(0, V, (B,), device=DEV)
+    ids[0] = ids[1]  # duplicate: scatter-add must accumulate
+    out = hip.embedding(ids, table)
+    assert torch.equal(out[0], out[1])
+    g = torch.randn_like(out)
+    out.backward(g)
+    ref = torch.zeros(V, E, device=DEV)
+    ref.index_add_(0, ids, g.float())
+    assert _rel_err(table.grad, ref) < 2e-2
+
+
+def test_masked_ce_fwd_bwd():
+    torch.manual_seed(5)
+    B, V = 640, 5000
+    logits = _bf(torch.randn(B, V)).requires_grad_(True)
+    labels = torch.randint(0, V, (B,), device=DEV)
+    mask = (torch.rand(B, device=DEV) > 0.3).float()
+    ce = hip.masked_softmax_ce(logits, labels, mask)
+    lr = logits.detach().float().requires_grad_(True)
+    ref = F.masked_softmax_ce(lr, labels, mask)
+    assert _rel_err(ce, ref) < 1e-2
+    assert (ce[mask == 0].abs() < 1e-6).all()
+
+    ce.sum().backward()
+    ref.sum().backward()
+    assert _rel_err(logits.grad, lr.grad) < 5e-2
+
+
+def test_adam_fused_matches_eager():
+    torch.manual_seed(6)
+    shapes = [(1000,), (64, 32), (5000, 16)]
+    params = [torch.randn(s, device=DEV) for s in shapes]
+    grads = [torch.randn(s, device=DEV) * 3 for s in shapes]
+    ms = [torch.zeros(s, device=DEV) for s in shapes]
+    vs = [torch.zeros(s, device=DEV) for s in shapes]
+    p_ref = [p.clone() for p in params]
+
+    lr, b1, b2, eps, clip = 1e-3, 0.9, 0.999, 1e-6, 5.0
+    gsq = hip.grad_sq_norm(grads)
+    norm_ref = torch.sqrt(sum((g ** 2).sum() for g in grads))
+    assert _rel_err(gsq.sqrt(), norm_ref) < 1e-4
+
+    hip.adam_step(params, grads, ms, vs, lr, b1, b2, eps, 1, clip, gsq)
+
+    scale = min(1.0, clip / norm_ref.item())
+    for p, g in zip(p_ref, grads):
+        gc = g * scale
+        m = (1 - b1) * gc
+        v = (1 - b2) * gc * gc
+        mh = m / (1 - b1)
+        vh = v / (1 - b2)
+        p -= lr * mh / (vh.sqrt() + eps)
+    for p, r in zip(params, p_ref):
+        assert _rel_err(p, r) < 1e-4
+
+
+def test_fused_optimizer_in_training(tiny_config):
+    """Full Optimizer.step on GPU uses the fused kernels end-to-end."""
+    from sat_amd.optim import Optimizer
+    from config import Config
+    cfg = Config()
+    p = torch.nn.Parameter(torch.randn(100, device=DEV))
+    opt = Optimizer(cfg, [p])
+    p.grad = torch.randn(100, device=DEV)
+    before = p.detach().clone()
+    opt.step()
+    torch.cuda.synchronize()
+    assert not torch.equal(before, p.detach())
+    assert torch.isfinite(p).all()
