@@ -178,10 +178,11 @@ def attention_pool(contexts, logits):
 class _Embedding(torch.autograd.Function):
     @staticmethod
     def forward(ctx, ids, table):
+        ids = ids.contiguous()
         ctx.save_for_backward(ids)
         ctx.rows = table.shape[0]
         ctx.table_dtype = table.dtype
-        return _C.embedding_fwd(ids, table)
+        return _C.embedding_fwd(ids, table.contiguous())
 
     @staticmethod
     def backward(ctx, dy):
@@ -197,7 +198,9 @@ def embedding(ids, table):
 class _MaskedCE(torch.autograd.Function):
     @staticmethod
     def forward(ctx, logits, labels, mask):
-        losses, lse = _C.ce_fwd(logits, labels, mask.contiguous())
+        labels = labels.contiguous()
+        mask = mask.contiguous()
+        losses, lse = _C.ce_fwd(logits, labels, mask)
         ctx.save_for_backward(logits, labels, mask, lse)
         return losses
 
