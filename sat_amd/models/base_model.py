@@ -60,6 +60,7 @@ class BaseModel(object):
 
         self.model = CaptionGenerator(config).to(self.device)
         self.global_step = 0
+        self._engine = None
 
         self.is_train = getattr(config, 'phase', 'train') == 'train'
         self.optimizer = Optimizer(
@@ -88,14 +89,25 @@ class BaseModel(object):
         return t.to(self.device, non_blocking=True)
 
     def train_step(self, images, sentences, masks):
-        """One fwd+bwd+optimizer step on device tensors. Returns loss dict."""
+        """One fwd+bwd+optimizer step on device tensors. Returns loss dict.
+
+        On GPU with config.use_hip_graph the whole step runs as one hipGraph
+        replay (sat_amd.engine.GraphedTrainStep)."""
         self.model.train()
-        out = self.model(images, sentences, masks)
-        self.optimizer.zero_grad()
-        out['total_loss'].backward()
-        if self.ddp is not None:
-            self.ddp.finish_backward()
-        self.optimizer.step()
+        if self._engine is None and self.device.type == 'cuda' \
+                and getattr(self.config, 'use_hip_graph', True):
+            from ..engine import GraphedTrainStep
+            self._engine = GraphedTrainStep(self.model, self.optimizer,
+                                            self.ddp)
+        if self._engine is not None:
+            out = self._engine.step(images, sentences, masks)
+        else:
+            out = self.model(images, sentences, masks)
+            self.optimizer.zero_grad()
+            out['total_loss'].backward()
+            if self.ddp is not None:
+                self.ddp.finish_backward()
+            self.optimizer.step()
         self.global_step += 1
         return out
 

@@ -39,6 +39,7 @@ class CaptionGenerator(tnn.Module):
         self.decoder = AttentionDecoder(config, self.nn,
                                         self.dim_ctx, self.num_ctx)
         self.global_step = 0
+        self._cnn_channels_last = False
 
         # freeze policy: CNN params trainable only with --train_cnn
         # (reference nn.py:66); everything non-trainable at eval/test.
@@ -49,14 +50,26 @@ class CaptionGenerator(tnn.Module):
 
     # ---- encoder ----
 
+    def _compute_dtype(self, images):
+        return (torch.bfloat16
+                if (images.is_cuda
+                    and getattr(self.config, 'compute_dtype', 'bf16')
+                    == 'bf16')
+                else torch.float32)
+
     def compute_contexts(self, images):
-        """images: [B,3,224,224] float -> contexts [B,L,D]."""
-        dtype = (torch.bfloat16
-                 if (images.is_cuda
-                     and getattr(self.config, 'compute_dtype', 'bf16')
-                     == 'bf16')
-                 else torch.float32)
+        """images: [B,3,224,224] float -> contexts [B,L,D].
+
+        On GPU the conv stack runs NHWC (channels_last): MIOpen's bf16
+        igemm kernels are NHWC-native; NCHW falls back to a naive conv
+        (measured 70% of step time, profiles/r01_baseline.md)."""
+        dtype = self._compute_dtype(images)
         images = images.to(dtype)
+        if images.is_cuda:
+            if not self._cnn_channels_last:
+                self.cnn.to(memory_format=torch.channels_last)
+                self._cnn_channels_last = True
+            images = images.contiguous(memory_format=torch.channels_last)
         if self.train_cnn:
             return self.cnn(images)
         with torch.no_grad():
@@ -76,6 +89,7 @@ class CaptionGenerator(tnn.Module):
 
         contexts = self.compute_contexts(images)
         contexts_flat = contexts.reshape(-1, self.dim_ctx)
+        self.decoder.precast(contexts.dtype)
 
         context_mean = contexts.float().mean(dim=1).to(contexts.dtype)
         initial_memory, initial_output = self.decoder.initialize(
@@ -116,6 +130,7 @@ class CaptionGenerator(tnn.Module):
         if isinstance(reg_loss, torch.Tensor):
             reg_loss = reg_loss.to(cross_entropy_loss.device)
 
+        self.decoder.clear_cast()
         total_loss = cross_entropy_loss + attention_loss + reg_loss
         accuracy = torch.stack(num_correct).sum() / mask_sum
 
@@ -142,7 +157,9 @@ class CaptionGenerator(tnn.Module):
     def decode_step(self, contexts, last_word, last_memory, last_output):
         """One inference decoder step -> (memory, output, probs [B,V])."""
         contexts_flat = contexts.reshape(-1, self.dim_ctx)
+        self.decoder.precast(contexts.dtype)
         logits, _alpha, memory, output, _sh = self.decoder.step(
             contexts, contexts_flat, last_word, last_memory, last_output)
+        self.decoder.clear_cast()
         probs = torch.softmax(logits.float(), dim=1)
         return memory, output, probs

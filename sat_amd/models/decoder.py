@@ -79,6 +79,39 @@ class AttentionDecoder(tnn.Module):
             self.dec_fc_1 = Dense(nn_policy, dim_exp, D_d, 'tanh')
             self.dec_fc_2 = Dense(nn_policy, D_d, V, None)
 
+        # per-forward cast cache (precast()/clear_cast())
+        self._emb_c = None
+        self._lstm_wc = None
+        self._lstm_bc = None
+        self._att_vc = None
+
+    # ---- per-forward weight cast cache ----
+
+    def precast(self, dtype):
+        """Cast every decoder weight to the compute dtype ONCE per model
+        forward (instead of once per dense call x T steps): removes ~450
+        bf16-cast kernel launches per training step while keeping the casts
+        inside the autograd graph."""
+        from .nn import Dense
+        for m in self.modules():
+            if isinstance(m, Dense):
+                m.precast(dtype)
+        self._emb_c = self.embedding.to(dtype)
+        self._lstm_wc = self.lstm_w.to(dtype)
+        self._lstm_bc = self.lstm_b.to(dtype)
+        if self.config.num_attend_layers != 1:
+            self._att_vc = self.att_fc_2._wc.reshape(-1)
+
+    def clear_cast(self):
+        from .nn import Dense
+        for m in self.modules():
+            if isinstance(m, Dense):
+                m.clear_cast()
+        self._emb_c = None
+        self._lstm_wc = None
+        self._lstm_bc = None
+        self._att_vc = None
+
     # ---- sub-networks ----
 
     def initialize(self, context_mean):
@@ -108,7 +141,8 @@ class AttentionDecoder(tnn.Module):
         t = t1 + t2.repeat_interleave(self.num_ctx, dim=0)
         t = self.nn.dropout(t)
         # fused tail: scores GEMV (fc_2, bias-free) + softmax + pool
-        v = self.att_fc_2.weight.reshape(-1).to(t.dtype)
+        v = self._att_vc if self._att_vc is not None \
+            else self.att_fc_2.weight.reshape(-1).to(t.dtype)
         return ops.attention_score_pool(t, v, contexts)
 
     def decode(self, expanded_output):
@@ -140,14 +174,17 @@ class AttentionDecoder(tnn.Module):
 
         alpha, context = self.attend(contexts, contexts_flat, last_output)
 
-        word_embed = ops.embedding(last_word, self.embedding.to(
-            contexts.dtype))
+        emb = self._emb_c if self._emb_c is not None \
+            else self.embedding.to(contexts.dtype)
+        word_embed = ops.embedding(last_word, emb)
 
         x = torch.cat([context, word_embed], dim=1)
         x = ops.dropout(x, rate, training)              # input_keep
-        h_raw, memory = ops.lstm_cell(
-            x, last_state_h, last_memory,
-            self.lstm_w.to(x.dtype), self.lstm_b.to(x.dtype))
+        lw = self._lstm_wc if self._lstm_wc is not None \
+            else self.lstm_w.to(x.dtype)
+        lb = self._lstm_bc if self._lstm_bc is not None \
+            else self.lstm_b.to(x.dtype)
+        h_raw, memory = ops.lstm_cell(x, last_state_h, last_memory, lw, lb)
         output = ops.dropout(h_raw, rate, training)     # output_keep
         state_output = ops.dropout(h_raw, rate, training)  # state_keep (h)
 

@@ -54,8 +54,10 @@ class VGG16(tnn.Module):
         x = images
         for layer in self._layers:
             x = layer(x)
-        # [B,512,14,14] -> [B,196,512]
-        return x.flatten(2).transpose(1, 2).contiguous()
+        # [B,512,14,14] -> [B,196,512]; for channels_last (GPU) the NHWC
+        # physical layout makes this permute+reshape a zero-copy view
+        return x.permute(0, 2, 3, 1).reshape(
+            x.shape[0], -1, x.shape[1]).contiguous()
 
 
 class _Bottleneck(tnn.Module):
@@ -116,7 +118,8 @@ class ResNet50(tnn.Module):
         x = self.pool1(x)
         for blk in self._blocks:
             x = blk(x)
-        return x.flatten(2).transpose(1, 2).contiguous()
+        return x.permute(0, 2, 3, 1).reshape(
+            x.shape[0], -1, x.shape[1]).contiguous()
 
 
 def build_encoder(name, nn_policy):

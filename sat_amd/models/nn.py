@@ -80,8 +80,23 @@ class Dense(tnn.Module):
             self.bias = tnn.Parameter(torch.zeros(units))
         else:
             self.register_parameter('bias', None)
+        self._wc = None  # per-forward cast cache (set by precast())
+        self._bc = None
+
+    def precast(self, dtype):
+        """Cast weights once per model forward; the cast stays in the
+        autograd graph, so T uses of the layer accumulate into one fp32
+        parameter grad (like TF's reuse_variables, model.py:312)."""
+        self._wc = self.weight.to(dtype)
+        self._bc = self.bias.to(dtype) if self.bias is not None else None
+
+    def clear_cast(self):
+        self._wc = None
+        self._bc = None
 
     def forward(self, x):
+        if self._wc is not None and self._wc.dtype == x.dtype:
+            return ops.dense(x, self._wc, self._bc, self.activation)
         w = self.weight.to(x.dtype)
         b = self.bias.to(x.dtype) if self.bias is not None else None
         return ops.dense(x, w, b, self.activation)

@@ -19,7 +19,8 @@ at::Tensor ce_bwd(at::Tensor logits, at::Tensor labels, at::Tensor mask,
 at::Tensor grad_sq_norm(std::vector<at::Tensor> grads);
 void adam_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
                std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
-               double lr, double b1, double b2, double eps, int64_t step,
+               at::Tensor step_dev, double lr0, double decay_factor,
+               double steps_per_decay, double b1, double b2, double eps,
                double clip, at::Tensor gsq);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {

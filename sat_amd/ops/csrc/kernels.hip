@@ -376,11 +376,22 @@ __global__ void sq_norm_kernel(const float* __restrict__ g, int64_t n,
     }
 }
 
+// step (and with it bias correction + staircase LR decay) comes from a
+// DEVICE scalar so the whole optimizer is hipGraph-capturable: replays see
+// the advancing counter, no host scalar is baked in at capture time.
 __global__ void adam_kernel(float* __restrict__ p, const float* __restrict__ g,
                             float* __restrict__ m, float* __restrict__ v,
                             const float* __restrict__ gsq,
-                            int64_t n, float lr, float b1, float b2,
-                            float eps, float bc1, float bc2, float clip) {
+                            const float* __restrict__ step_dev,
+                            int64_t n, float lr0, float decay_factor,
+                            float steps_per_decay, float b1, float b2,
+                            float eps, float clip) {
+    float step = *step_dev;
+    float bc1 = 1.f - powf(b1, step);
+    float bc2 = 1.f - powf(b2, step);
+    float lr = lr0;
+    if (decay_factor < 1.f)
+        lr = lr0 * powf(decay_factor, floorf(step / steps_per_decay));
     float scale = 1.f;
     if (clip > 0.f) {
         float norm = sqrtf(*gsq);
@@ -417,10 +428,10 @@ at::Tensor grad_sq_norm(std::vector<at::Tensor> grads) {
 
 void adam_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
                std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
-               double lr, double b1, double b2, double eps, int64_t step,
+               at::Tensor step_dev, double lr0, double decay_factor,
+               double steps_per_decay, double b1, double b2, double eps,
                double clip, at::Tensor gsq) {
-    float bc1 = 1.f - powf((float)b1, (float)step);
-    float bc2 = 1.f - powf((float)b2, (float)step);
+    CHECK_GPU(step_dev); CHECK_F32(step_dev);
     hipStream_t s = at::cuda::getCurrentCUDAStream();
     for (size_t i = 0; i < params.size(); ++i) {
         auto& p = params[i];
@@ -431,9 +442,11 @@ void adam_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
                            (float*)p.data_ptr(),
                            (const float*)grads[i].data_ptr(),
                            (float*)ms[i].data_ptr(), (float*)vs[i].data_ptr(),
-                           (const float*)gsq.data_ptr(), n,
-                           (float)lr, (float)b1, (float)b2, (float)eps,
-                           bc1, bc2, (float)clip);
+                           (const float*)gsq.data_ptr(),
+                           (const float*)step_dev.data_ptr(), n,
+                           (float)lr0, (float)decay_factor,
+                           (float)steps_per_decay, (float)b1, (float)b2,
+                           (float)eps, (float)clip);
     }
     HIP_OK(hipGetLastError());
 }

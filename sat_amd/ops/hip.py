@@ -222,11 +222,14 @@ def grad_sq_norm(grads):
     return _C.grad_sq_norm([g.contiguous() for g in grads])
 
 
-def adam_step(params, grads, ms, vs, lr, beta1, beta2, eps, step,
-              clip, grad_sq):
-    """Fused multi-tensor Adam. `grad_sq` is the on-device Σ‖g‖² scalar from
-    grad_sq_norm; the kernel derives scale = clip/max(clip, √grad_sq)."""
+def adam_step(params, grads, ms, vs, step_dev, lr0, decay_factor,
+              steps_per_decay, beta1, beta2, eps, clip, grad_sq):
+    """Fused multi-tensor Adam.  `grad_sq` is the on-device Σ‖g‖² scalar
+    from grad_sq_norm (clip scale derived in-kernel); `step_dev` is the
+    on-device step counter (bias correction + staircase LR decay computed
+    in-kernel) so the whole optimizer step is hipGraph-capturable."""
     _C.adam_step(list(params), [g.contiguous() for g in grads],
-                 list(ms), list(vs),
-                 float(lr), float(beta1), float(beta2), float(eps),
-                 int(step), float(clip), grad_sq)
+                 list(ms), list(vs), step_dev,
+                 float(lr0), float(decay_factor), float(steps_per_decay),
+                 float(beta1), float(beta2), float(eps), float(clip),
+                 grad_sq)

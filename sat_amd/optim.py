@@ -65,14 +65,23 @@ class Optimizer(object):
         if use_hip:
             from .ops import hip
             hip.require()
-            # grad-norm² reduction and clip-scale stay on-device: no host
-            # sync anywhere in the optimizer step (hipGraph-capturable).
+            # Everything stays on-device (grad-norm², clip scale, step
+            # counter, bias correction, LR decay): zero host syncs, so the
+            # whole optimizer step is hipGraph-capturable.
+            if not hasattr(self, 'step_dev') or self.step_dev is None:
+                self.step_dev = torch.zeros(
+                    (), dtype=torch.float32,
+                    device=self.params[0].device)
+                self.step_dev.fill_(float(self.step_count - 1))
+            self.step_dev.add_(1.0)
             gsq = hip.grad_sq_norm(grads)
             hip.adam_step(
                 [p.data for p in self.params], grads,
                 [self.state[p]['m'] for p in self.params],
                 [self.state[p]['v'] for p in self.params],
-                lr, cfg.beta1, cfg.beta2, cfg.epsilon, self.step_count,
+                self.step_dev, cfg.initial_learning_rate,
+                cfg.learning_rate_decay_factor, cfg.num_steps_per_decay,
+                cfg.beta1, cfg.beta2, cfg.epsilon,
                 cfg.clip_gradients, gsq)
             return
 
@@ -133,3 +142,5 @@ class Optimizer(object):
                             self.state[p][k].device))
         if 'optimizer/step_count' in arrays:
             self.step_count = int(float(arrays['optimizer/step_count']))
+            if getattr(self, 'step_dev', None) is not None:
+                self.step_dev.fill_(float(self.step_count))

@@ -187,7 +187,9 @@ def test_adam_fused_matches_eager():
     norm_ref = torch.sqrt(sum((g ** 2).sum() for g in grads))
     assert _rel_err(gsq.sqrt(), norm_ref) < 1e-4
 
-    hip.adam_step(params, grads, ms, vs, lr, b1, b2, eps, 1, clip, gsq)
+    step_dev = torch.ones((), device=DEV)
+    hip.adam_step(params, grads, ms, vs, step_dev, lr, 1.0, 1e5,
+                  b1, b2, eps, clip, gsq)
 
     scale = min(1.0, clip / norm_ref.item())
     for p, g in zip(p_ref, grads):
