@@ -106,7 +106,7 @@ class CaptionGenerator(tnn.Module):
         for t in range(T):
             logits, alpha, memory, output, state_h = self.decoder.step(
                 contexts, contexts_flat, last_word, memory, output,
-                state_h)
+                state_h, salt=t)
             m = masks[:, t]
             cross_entropies.append(
                 ops.masked_softmax_ce(logits, sentences[:, t], m))

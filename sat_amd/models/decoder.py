@@ -84,6 +84,9 @@ class AttentionDecoder(tnn.Module):
         self._lstm_wc = None
         self._lstm_bc = None
         self._att_vc = None
+        # device-side dropout seed for the fused attention tail; advanced
+        # once per forward so hipGraph replays draw fresh masks
+        self._rng = None
 
     # ---- per-forward weight cast cache ----
 
@@ -101,6 +104,12 @@ class AttentionDecoder(tnn.Module):
         self._lstm_bc = self.lstm_b.to(dtype)
         if self.config.num_attend_layers != 1:
             self._att_vc = self.att_fc_2._wc.reshape(-1)
+        dev = self.embedding.device
+        if dev.type == 'cuda':
+            if self._rng is None or self._rng.device != dev:
+                self._rng = torch.randint(
+                    0, 2 ** 31, (), dtype=torch.int64, device=dev)
+            self._rng.add_(1)
 
     def clear_cast(self):
         from .nn import Dense
@@ -124,7 +133,7 @@ class AttentionDecoder(tnn.Module):
         tb = self.nn.dropout(self.init_fc_b1(x))
         return self.init_fc_a2(ta), self.init_fc_b2(tb)
 
-    def attend(self, contexts, contexts_flat, output):
+    def attend(self, contexts, contexts_flat, output, salt=0):
         """Attention: MLP scores + softmax over L + weighted context sum
         (model.py:395-436, :263-264).  Returns (alpha [B,L], context [B,D]).
         """
@@ -138,12 +147,13 @@ class AttentionDecoder(tnn.Module):
             return ops.attention_pool(contexts, l1 + l2)
         t1 = self.att_fc_1a(ctx)                       # [B·L, A]
         t2 = self.att_fc_1b(out)                       # [B, A]
-        t = t1 + t2.repeat_interleave(self.num_ctx, dim=0)
-        t = self.nn.dropout(t)
-        # fused tail: scores GEMV (fc_2, bias-free) + softmax + pool
+        # fused tail: tiled add + dropout + scores GEMV (fc_2, bias-free)
+        # + softmax over L + weighted context sum
         v = self._att_vc if self._att_vc is not None \
-            else self.att_fc_2.weight.reshape(-1).to(t.dtype)
-        return ops.attention_score_pool(t, v, contexts)
+            else self.att_fc_2.weight.reshape(-1).to(t1.dtype)
+        return ops.attention_tail(t1, t2, v, contexts,
+                                  self.nn.fc_drop_rate, self.nn.is_train,
+                                  self._rng, salt)
 
     def decode(self, expanded_output):
         """[B, H+D+E] -> logits [B, V]."""
@@ -157,7 +167,7 @@ class AttentionDecoder(tnn.Module):
     # ---- one decoder step (shared by train loop and beam search) ----
 
     def step(self, contexts, contexts_flat, last_word, last_memory,
-             last_output, last_state_h=None):
+             last_output, last_state_h=None, salt=0):
         """Run attention + LSTM + decode for one step.
 
         `last_output` feeds the attention MLP (it is the DropoutWrapper
@@ -172,7 +182,8 @@ class AttentionDecoder(tnn.Module):
         rate = self.nn.lstm_drop_rate
         training = self.nn.is_train
 
-        alpha, context = self.attend(contexts, contexts_flat, last_output)
+        alpha, context = self.attend(contexts, contexts_flat, last_output,
+                                     salt)
 
         emb = self._emb_c if self._emb_c is not None \
             else self.embedding.to(contexts.dtype)

@@ -57,14 +57,17 @@ def attention_pool(contexts, logits):
     return F.attention_pool(contexts, logits)
 
 
-def attention_score_pool(temp_flat, v, contexts):
-    """Fused attention tail: scores GEMV + softmax over L + context sum.
-    Replaces the reference's N=1 GEMM (attend fc_2, model.py:429-434) +
-    softmax + weighted-sum trio with two fused kernels on GPU."""
-    if _use_hip(contexts, temp_flat):
+def attention_tail(t1, t2, v, contexts, p, training, seed_dev, salt):
+    """Fused attention tail: dropout(t1 + tiled t2) · v -> softmax over L
+    -> weighted context sum.  Replaces the reference's tile/add/dropout/
+    N=1-GEMM/softmax/weighted-sum chain (model.py:425-435, :263-264) with
+    two fused kernels on GPU; counter-based dropout (seed_dev, salt) keeps
+    it hipGraph-safe."""
+    if _use_hip(contexts, t1):
         from . import hip
-        return hip.attention_score_pool(temp_flat, v, contexts)
-    return F.attention_score_pool(temp_flat, v, contexts)
+        return hip.attention_tail(t1, t2, v, contexts,
+                                  p if training else 0.0, seed_dev, salt)
+    return F.attention_tail(t1, t2, v, contexts, p, training)
 
 
 def embedding(ids, table):

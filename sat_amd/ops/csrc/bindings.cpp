@@ -8,8 +8,18 @@ std::vector<at::Tensor> lstm_pointwise_fwd(at::Tensor gates, at::Tensor c,
 std::vector<at::Tensor> lstm_pointwise_bwd(at::Tensor gates, at::Tensor c,
                                            at::Tensor dh, at::Tensor dc,
                                            double fb);
+at::Tensor act_bwd(at::Tensor dy, at::Tensor y, int64_t act);
 at::Tensor attn_score_fwd(at::Tensor temp, at::Tensor v);
 std::vector<at::Tensor> attn_pool_fwd(at::Tensor ctx, at::Tensor logits);
+std::vector<at::Tensor> attn_scores_fused(at::Tensor t1, at::Tensor t2,
+                                          at::Tensor v, at::Tensor seed,
+                                          double p, int64_t salt, int64_t L);
+std::vector<at::Tensor> attn_pool_bwd(at::Tensor ctx, at::Tensor alpha,
+                                      at::Tensor dalpha, at::Tensor dpooled,
+                                      bool need_dctx);
+std::vector<at::Tensor> attn_scores_bwd(at::Tensor tdrop, at::Tensor v,
+                                        at::Tensor dlogits, at::Tensor seed,
+                                        double p, int64_t salt, int64_t L);
 at::Tensor embedding_fwd(at::Tensor ids, at::Tensor table);
 at::Tensor embedding_bwd(at::Tensor ids, at::Tensor dy, int64_t rows);
 std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor labels,
@@ -27,8 +37,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("dense_fwd", &dense_fwd, "MFMA GEMM + bias/act (bf16)");
     m.def("lstm_pointwise_fwd", &lstm_pointwise_fwd);
     m.def("lstm_pointwise_bwd", &lstm_pointwise_bwd);
+    m.def("act_bwd", &act_bwd);
     m.def("attn_score_fwd", &attn_score_fwd);
     m.def("attn_pool_fwd", &attn_pool_fwd);
+    m.def("attn_scores_fused", &attn_scores_fused);
+    m.def("attn_pool_bwd", &attn_pool_bwd);
+    m.def("attn_scores_bwd", &attn_scores_bwd);
     m.def("embedding_fwd", &embedding_fwd);
     m.def("embedding_bwd", &embedding_bwd);
     m.def("ce_fwd", &ce_fwd);

@@ -98,6 +98,37 @@ std::vector<at::Tensor> lstm_pointwise_bwd(at::Tensor gates, at::Tensor c,
 }
 
 // ======================================================================
+// Activation backward: dpre = dy * act'(y) in ONE kernel (the eager chain
+// y.float / 1-y^2 / mul / cast is 4 launches per dense backward).
+// ======================================================================
+
+__global__ void act_bwd_kernel(const bf16* __restrict__ dy,
+                               const bf16* __restrict__ y,
+                               bf16* __restrict__ dpre,
+                               int64_t n, int act) {
+    int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= n) return;
+    float g = bf2f(dy[idx]);
+    float yv = bf2f(y[idx]);
+    if (act == 1) g *= (1.f - yv * yv);        // tanh'
+    else if (act == 2) g *= (yv > 0.f ? 1.f : 0.f);  // relu'
+    dpre[idx] = f2bf(g);
+}
+
+at::Tensor act_bwd(at::Tensor dy, at::Tensor y, int64_t act) {
+    CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_BF16(dy);
+    CHECK_GPU(y); CHECK_CONTIG(y); CHECK_BF16(y);
+    auto dpre = at::empty_like(dy);
+    int64_t n = dy.numel();
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(act_bwd_kernel, dim3(cdiv(n, 256)), dim3(256), 0, s,
+                       (const bf16*)dy.data_ptr(), (const bf16*)y.data_ptr(),
+                       (bf16*)dpre.data_ptr(), n, (int)act);
+    HIP_OK(hipGetLastError());
+    return dpre;
+}
+
+// ======================================================================
 // Attention: scores GEMV (temp[M,A] @ v[A] -> logits[M]) and the fused
 // LDS-staged softmax-over-L + weighted context sum (model.py:435,263-264).
 // ======================================================================
@@ -138,6 +169,238 @@ at::Tensor attn_score_fwd(at::Tensor temp, at::Tensor v) {
 }
 
 #define MAX_L 1024
+
+// counter-based dropout hash: deterministic in (seed, salt, index), so the
+// backward regenerates the mask with zero storage, and the seed lives in a
+// DEVICE scalar the model advances once per step (hipGraph-safe: replays
+// see fresh masks, unlike a host-baked philox offset).
+__device__ __forceinline__ uint32_t mix3(uint32_t a, uint32_t b, uint32_t c) {
+    uint32_t h = a * 0x9E3779B1u ^ b * 0x85EBCA77u ^ c * 0xC2B2AE3Du;
+    h ^= h >> 16; h *= 0x7FEB352Du;
+    h ^= h >> 15; h *= 0x846CA68Bu;
+    h ^= h >> 16;
+    return h;
+}
+
+__device__ __forceinline__ float drop_scale(const int64_t* seed, int salt,
+                                            uint32_t idx, float p) {
+    if (p <= 0.f) return 1.f;
+    uint32_t h = mix3((uint32_t)(*seed), (uint32_t)salt, idx);
+    float u = (h >> 8) * (1.0f / 16777216.0f);
+    return u >= p ? 1.0f / (1.0f - p) : 0.0f;
+}
+
+// ---- fused attention scores: t = dropout(t1 + t2), logits = t·v ----
+// t1: [B·L, A], t2: [B, A], v: [A]; writes tdrop (saved for dv in bwd)
+// and logits [B·L] fp32.  One wave per row.
+
+__global__ void attn_scores_fused_kernel(
+        const bf16* __restrict__ t1, const bf16* __restrict__ t2,
+        const bf16* __restrict__ v, const int64_t* __restrict__ seed,
+        bf16* __restrict__ tdrop, float* __restrict__ logits,
+        int B, int L, int A, float p, int salt) {
+    int row = blockIdx.x * 4 + (threadIdx.x >> 6);
+    if (row >= B * L) return;
+    int lane = threadIdx.x & 63;
+    int b = row / L;
+    const bf16* t1r = t1 + (int64_t)row * A;
+    const bf16* t2r = t2 + (int64_t)b * A;
+    bf16* tdr = tdrop + (int64_t)row * A;
+    float acc = 0.f;
+    for (int a0 = lane * 8; a0 + 8 <= A; a0 += 64 * 8) {
+        bf16x8 x1 = *(const bf16x8*)(t1r + a0);
+        bf16x8 x2 = *(const bf16x8*)(t2r + a0);
+        bf16x8 vv = *(const bf16x8*)(v + a0);
+        bf16x8 out;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+            float t = bf2f(x1[e]) + bf2f(x2[e]);
+            t *= drop_scale(seed, salt, (uint32_t)(row * A + a0 + e), p);
+            out[e] = f2bf(t);
+            acc += t * bf2f(vv[e]);
+        }
+        *(bf16x8*)(tdr + a0) = out;
+    }
+    acc = wave_sum(acc);
+    if (lane == 0) logits[row] = acc;
+}
+
+std::vector<at::Tensor> attn_scores_fused(at::Tensor t1, at::Tensor t2,
+                                          at::Tensor v, at::Tensor seed,
+                                          double p, int64_t salt,
+                                          int64_t L) {
+    CHECK_GPU(t1); CHECK_CONTIG(t1); CHECK_BF16(t1);
+    CHECK_GPU(t2); CHECK_CONTIG(t2); CHECK_BF16(t2);
+    CHECK_GPU(v); CHECK_CONTIG(v); CHECK_BF16(v);
+    int A = t1.size(1);
+    int rows = t1.size(0);
+    int B = rows / (int)L;
+    TORCH_CHECK(A % 8 == 0 && rows % L == 0);
+    auto tdrop = at::empty_like(t1);
+    auto logits = at::empty({B, (int)L},
+                            t1.options().dtype(at::kFloat));
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(attn_scores_fused_kernel, dim3(cdiv(rows, 4)),
+                       dim3(256), 0, s,
+                       (const bf16*)t1.data_ptr(),
+                       (const bf16*)t2.data_ptr(),
+                       (const bf16*)v.data_ptr(),
+                       (const int64_t*)seed.data_ptr(),
+                       (bf16*)tdrop.data_ptr(), (float*)logits.data_ptr(),
+                       B, (int)L, A, (float)p, (int)salt);
+    HIP_OK(hipGetLastError());
+    return {tdrop, logits};
+}
+
+// ---- attention pool backward ----
+// per image: s_l = Σ_d dpooled·ctx[l,d]; da = dα_ext + s;
+// dlogits = α (da − Σ_l α·da); optional dctx = α ⊗ dpooled.
+
+__global__ void attn_pool_bwd_kernel(
+        const bf16* __restrict__ ctx, const float* __restrict__ alpha,
+        const float* __restrict__ dalpha, const bf16* __restrict__ dpooled,
+        float* __restrict__ dlogits, bf16* __restrict__ dctx,
+        int L, int D) {
+    __shared__ float sa[MAX_L];
+    __shared__ float red[4];
+    int b = blockIdx.x;
+    int tid = threadIdx.x;
+    int wid = tid >> 6, lane = tid & 63;
+    const bf16* cb = ctx + (int64_t)b * L * D;
+    const bf16* dp = dpooled + (int64_t)b * D;
+
+    for (int l = wid; l < L; l += 4) {
+        float acc = 0.f;
+        for (int d0 = lane * 8; d0 + 8 <= D; d0 += 64 * 8) {
+            bf16x8 cv = *(const bf16x8*)(cb + (int64_t)l * D + d0);
+            bf16x8 dv = *(const bf16x8*)(dp + d0);
+#pragma unroll
+            for (int e = 0; e < 8; ++e) acc += bf2f(cv[e]) * bf2f(dv[e]);
+        }
+        acc = wave_sum(acc);
+        if (lane == 0) {
+            float da = acc;
+            if (dalpha != nullptr) da += dalpha[(int64_t)b * L + l];
+            sa[l] = da;
+        }
+    }
+    __syncthreads();
+    // dot = Σ α·da
+    float part = 0.f;
+    for (int l = tid; l < L; l += blockDim.x)
+        part += alpha[(int64_t)b * L + l] * sa[l];
+    part = wave_sum(part);
+    if ((tid & 63) == 0) red[wid] = part;
+    __syncthreads();
+    float dot = red[0] + red[1] + red[2] + red[3];
+    for (int l = tid; l < L; l += blockDim.x) {
+        float a = alpha[(int64_t)b * L + l];
+        dlogits[(int64_t)b * L + l] = a * (sa[l] - dot);
+    }
+    if (dctx != nullptr) {
+        for (int l = wid; l < L; l += 4) {
+            float a = alpha[(int64_t)b * L + l];
+            for (int d0 = lane * 8; d0 + 8 <= D; d0 += 64 * 8) {
+                bf16x8 dv = *(const bf16x8*)(dp + d0);
+                bf16x8 out;
+#pragma unroll
+                for (int e = 0; e < 8; ++e)
+                    out[e] = f2bf(a * bf2f(dv[e]));
+                *(bf16x8*)(dctx + (int64_t)b * L * D + (int64_t)l * D + d0)
+                    = out;
+            }
+        }
+    }
+}
+
+std::vector<at::Tensor> attn_pool_bwd(at::Tensor ctx, at::Tensor alpha,
+                                      at::Tensor dalpha, at::Tensor dpooled,
+                                      bool need_dctx) {
+    CHECK_GPU(ctx); CHECK_CONTIG(ctx); CHECK_BF16(ctx);
+    CHECK_GPU(alpha); CHECK_CONTIG(alpha); CHECK_F32(alpha);
+    int B = ctx.size(0), L = ctx.size(1), D = ctx.size(2);
+    TORCH_CHECK(L <= MAX_L && D % 8 == 0);
+    auto dlogits = at::empty({B, L}, alpha.options());
+    at::Tensor dctx;
+    bf16* dctx_ptr = nullptr;
+    if (need_dctx) {
+        dctx = at::empty_like(ctx);
+        dctx_ptr = (bf16*)dctx.data_ptr();
+    } else {
+        dctx = at::empty({0}, ctx.options());
+    }
+    const float* dalpha_ptr = nullptr;
+    if (dalpha.defined() && dalpha.numel() > 0)
+        dalpha_ptr = (const float*)dalpha.data_ptr();
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(attn_pool_bwd_kernel, dim3(B), dim3(256), 0, s,
+                       (const bf16*)ctx.data_ptr(),
+                       (const float*)alpha.data_ptr(), dalpha_ptr,
+                       (const bf16*)dpooled.data_ptr(),
+                       (float*)dlogits.data_ptr(), dctx_ptr, L, D);
+    HIP_OK(hipGetLastError());
+    return {dlogits, dctx};
+}
+
+// ---- attention scores backward ----
+// dt1 = dlogit ⊗ v ⊙ mask/(1-p) (mask regenerated); dt2 = Σ_l dt1;
+// dv = Σ_rows tdrop·dlogit.  Blocks cover l-chunks within one image.
+
+__global__ void attn_scores_bwd_kernel(
+        const bf16* __restrict__ tdrop, const bf16* __restrict__ v,
+        const float* __restrict__ dlogits,
+        const int64_t* __restrict__ seed,
+        bf16* __restrict__ dt1, float* __restrict__ dt2,
+        float* __restrict__ dvf,
+        int B, int L, int A, int lchunk, float p, int salt) {
+    int nchunk = (L + lchunk - 1) / lchunk;
+    int b = blockIdx.x / nchunk;
+    int l0 = (blockIdx.x % nchunk) * lchunk;
+    int l1 = min(L, l0 + lchunk);
+    int tid = threadIdx.x;
+    for (int a = tid; a < A; a += blockDim.x) {
+        float va = bf2f(v[a]);
+        float dv_acc = 0.f;
+        float dt2_acc = 0.f;
+        for (int l = l0; l < l1; ++l) {
+            int64_t row = (int64_t)b * L + l;
+            float dl = dlogits[row];
+            float td = bf2f(tdrop[row * A + a]);
+            dv_acc += td * dl;
+            float dt = dl * va *
+                drop_scale(seed, salt, (uint32_t)(row * A + a), p);
+            dt1[row * A + a] = f2bf(dt);
+            dt2_acc += dt;
+        }
+        atomicAdd(dvf + a, dv_acc);
+        atomicAdd(dt2 + (int64_t)b * A + a, dt2_acc);
+    }
+}
+
+std::vector<at::Tensor> attn_scores_bwd(at::Tensor tdrop, at::Tensor v,
+                                        at::Tensor dlogits, at::Tensor seed,
+                                        double p, int64_t salt, int64_t L) {
+    CHECK_GPU(tdrop); CHECK_CONTIG(tdrop); CHECK_BF16(tdrop);
+    int rows = tdrop.size(0), A = tdrop.size(1);
+    int B = rows / (int)L;
+    auto dt1 = at::empty_like(tdrop);
+    auto dt2 = at::zeros({B, A}, tdrop.options().dtype(at::kFloat));
+    auto dvf = at::zeros({A}, tdrop.options().dtype(at::kFloat));
+    int lchunk = ((int)L + 3) / 4;
+    int nchunk = ((int)L + lchunk - 1) / lchunk;
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(attn_scores_bwd_kernel, dim3(B * nchunk),
+                       dim3(256), 0, s,
+                       (const bf16*)tdrop.data_ptr(),
+                       (const bf16*)v.data_ptr(),
+                       (const float*)dlogits.data_ptr(),
+                       (const int64_t*)seed.data_ptr(),
+                       (bf16*)dt1.data_ptr(), (float*)dt2.data_ptr(),
+                       (float*)dvf.data_ptr(),
+                       B, (int)L, A, lchunk, (float)p, (int)salt);
+    HIP_OK(hipGetLastError());
+    return {dt1, dt2, dvf};
+}
 
 __global__ void attn_pool_kernel(const bf16* __restrict__ ctx,   // [B,L,D]
                                  const float* __restrict__ logits, // [B,L]

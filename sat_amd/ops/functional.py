@@ -54,12 +54,14 @@ def attention_pool(contexts, logits):
     return alpha, context
 
 
-def attention_score_pool(temp_flat, v, contexts):
-    """Fused attention tail (reference model.py:429-435 + :263-264):
-    logits = temp_flat[B·L,A] @ v[A], alpha = softmax over L, context =
-    Σ_l α_l·ctx_l.  Returns (alpha [B,L], context [B,D])."""
+def attention_tail(t1, t2, v, contexts, p, training):
+    """Attention tail (reference model.py:425-435 + :263-264):
+    t = dropout(t1 + tiled t2); logits = t @ v; alpha = softmax over L;
+    context = Σ_l α_l·ctx_l.  Returns (alpha [B,L], context [B,D])."""
     B, L = contexts.shape[0], contexts.shape[1]
-    logits = temp_flat.matmul(v).reshape(B, L)
+    t = t1 + t2.repeat_interleave(L, dim=0)
+    t = dropout(t, p, training)
+    logits = t.matmul(v).reshape(B, L)
     return attention_pool(contexts, logits)
 
 

@@ -67,11 +67,8 @@ class _Dense(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         x, w, y = ctx.saved_tensors
-        if ctx.act == 'tanh':
-            yf = y.float()
-            dpre = (dy.float() * (1 - yf * yf)).to(dy.dtype)
-        elif ctx.act == 'relu':
-            dpre = dy * (y > 0).to(dy.dtype)
+        if ctx.act in ('tanh', 'relu'):
+            dpre = _C.act_bwd(dy.contiguous(), y, _ACT[ctx.act])
         else:
             dpre = dy
         dx = dpre.matmul(w) if ctx.needs_input_grad[0] else None
@@ -115,8 +112,7 @@ def lstm_cell(x, h, c, weight, bias, forget_bias=1.0):
 
 class _AttnPool(torch.autograd.Function):
     """(contexts [B,L,D] bf16, logits [B,L] fp32) -> (alpha fp32, pooled
-    bf16).  Forward is the fused LDS-staged kernel; backward composes
-    hipBLASLt bmm + eager softmax-grad (library GEMMs, not hot)."""
+    bf16).  Fused LDS-staged kernels both directions."""
 
     @staticmethod
     def forward(ctx, contexts, logits):
@@ -127,48 +123,52 @@ class _AttnPool(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dalpha, dpooled):
         contexts, alpha = ctx.saved_tensors
-        # d(alpha)/d(logits) softmax backward + pooled path
-        da = torch.bmm(contexts.float(),
-                       dpooled.float().unsqueeze(2)).squeeze(2)  # [B,L]
-        if dalpha is not None:
-            da = da + dalpha
-        dlogits = alpha * (da - (alpha * da).sum(dim=1, keepdim=True))
-        dctx = None
-        if ctx.needs_input_grad[0]:
-            dctx = (alpha.unsqueeze(2) * dpooled.float().unsqueeze(1)) \
-                .to(contexts.dtype)
-        return dctx, dlogits
+        da = dalpha.contiguous() if dalpha is not None \
+            else torch.empty(0, device=alpha.device, dtype=alpha.dtype)
+        dlogits, dctx = _C.attn_pool_bwd(
+            contexts, alpha, da, dpooled.contiguous(),
+            bool(ctx.needs_input_grad[0]))
+        return (dctx if ctx.needs_input_grad[0] else None), dlogits
 
 
-class _AttnScore(torch.autograd.Function):
-    """temp [M,A] bf16 @ v [A] bf16 -> logits fp32 [M] (GEMV kernel)."""
-
-    @staticmethod
-    def forward(ctx, temp, v):
-        logits = _C.attn_score_fwd(temp.contiguous(), v.contiguous())
-        ctx.save_for_backward(temp, v)
-        return logits
+class _AttnTail(torch.autograd.Function):
+    """The whole attention tail, fused (reference model.py:425-435 +
+    :263-264): t = dropout(t1 + tiled t2), logits = t·v, α = softmax over
+    L, pooled = Σ_l α_l·ctx_l.  Dropout masks are counter-based
+    (seed_dev advances on device once per step -> hipGraph-safe) and
+    regenerated in backward with zero mask storage."""
 
     @staticmethod
-    def backward(ctx, dlogits):
-        temp, v = ctx.saved_tensors
-        dl = dlogits.contiguous()
-        dtemp = None
-        dv = None
-        if ctx.needs_input_grad[0]:
-            dtemp = (dl.unsqueeze(1) * v.float().unsqueeze(0)) \
-                .to(temp.dtype)
-        if ctx.needs_input_grad[1]:
-            dv = temp.float().t().matmul(dl).to(v.dtype)
-        return dtemp, dv
+    def forward(ctx, t1, t2, v, contexts, p, seed_dev, salt):
+        B, L = contexts.shape[0], contexts.shape[1]
+        tdrop, logits = _C.attn_scores_fused(
+            t1.contiguous(), t2.contiguous(), v.contiguous(), seed_dev,
+            float(p), int(salt), L)
+        alpha, pooled = _C.attn_pool_fwd(contexts, logits)
+        ctx.save_for_backward(tdrop, v, alpha, contexts, seed_dev)
+        ctx.p = float(p)
+        ctx.salt = int(salt)
+        ctx.L = L
+        return alpha, pooled
+
+    @staticmethod
+    def backward(ctx, dalpha, dpooled):
+        tdrop, v, alpha, contexts, seed_dev = ctx.saved_tensors
+        da = dalpha.contiguous() if dalpha is not None \
+            else torch.empty(0, device=alpha.device, dtype=alpha.dtype)
+        dlogits, dctx = _C.attn_pool_bwd(
+            contexts, alpha, da, dpooled.contiguous(),
+            bool(ctx.needs_input_grad[3]))
+        dt1, dt2f, dvf = _C.attn_scores_bwd(
+            tdrop, v, dlogits, seed_dev, ctx.p, ctx.salt, ctx.L)
+        return (dt1, dt2f.to(tdrop.dtype), dvf.to(v.dtype),
+                (dctx if ctx.needs_input_grad[3] else None), None, None,
+                None)
 
 
-def attention_score_pool(temp_flat, v, contexts):
-    """Fused attention tail: logits = temp·v, alpha = softmax_L, context =
-    Σ_l α_l ctx_l.  Returns (alpha [B,L] fp32, pooled [B,D])."""
-    logits = _AttnScore.apply(temp_flat, v).reshape(contexts.shape[0],
-                                                    contexts.shape[1])
-    return _AttnPool.apply(contexts, logits)
+def attention_tail(t1, t2, v, contexts, p, seed_dev, salt):
+    """t1: [B·L,A], t2: [B,A], v: [A] -> (alpha [B,L] fp32, pooled [B,D])."""
+    return _AttnTail.apply(t1, t2, v, contexts, p, seed_dev, salt)
 
 
 def attention_pool(contexts, logits):
