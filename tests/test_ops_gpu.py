@@ -131,8 +131,13 @@ def test_attention_tail_fwd_bwd_nodrop():
     (pooled.float() * gp.float()).sum().backward()
     (ra * ga).sum().backward(retain_graph=True)
     (rp * gp.float()).sum().backward()
-    for t, r in zip((t1, t2, v, ctx), refs):
+    for t, r in zip((t1, v, ctx), (refs[0], refs[2], refs[3])):
         assert _rel_err(t.grad, r.grad) < 5e-2
+    # t2's true gradient is ~0 at p=0 (softmax is shift-invariant in the
+    # broadcast t2), so compare absolutely against the dt1 scale instead
+    scale = t1.grad.float().abs().max().item()
+    assert (t2.grad.float() - refs[1].grad).abs().max().item() \
+        < 5e-2 * scale
 
 
 def test_attention_tail_dropout_semantics():
@@ -168,12 +173,15 @@ def test_attention_tail_dropout_semantics():
         t1.detach(), t2.detach(), v.detach(), seed, p, 4, L)
     assert not torch.equal(tdrop, tdrop3)
 
-    # backward regenerates the same mask: dt1 must be 0 wherever the
-    # forward dropped
-    gp = torch.randn_like(pooled)
-    (pooled.float() * gp.float()).sum().backward()
-    assert (t1.grad[~nz] == 0).all()
-    assert (t1.grad != 0).any()
+    # fwd/bwd mask consistency, with exactly recoverable masks:
+    # t = 1 everywhere -> tdrop ∈ {0, 2}; dlogits = 1, v = 1 -> dt1 ∈ {0, 2}
+    ones1 = torch.ones(B * L, A, device=DEV, dtype=torch.bfloat16)
+    zeros2 = torch.zeros(B, A, device=DEV, dtype=torch.bfloat16)
+    vones = torch.ones(A, device=DEV, dtype=torch.bfloat16)
+    td, _ = _C.attn_scores_fused(ones1, zeros2, vones, seed, p, 9, L)
+    dl = torch.ones(B, L, device=DEV)
+    dt1b, _, _ = _C.attn_scores_bwd(td, vones, dl, seed, p, 9, L)
+    assert torch.equal(td != 0, dt1b != 0)
 
 
 def test_attention_pool_small_l():
