@@ -1,131 +1,251 @@
 // dense_fwd: Y[M,N] = act(X[M,K] @ W[N,K]^T + bias), bf16 in/out, fp32 acc.
 //
-// The hand-written MFMA GEMM behind every fc layer of the decoder
+// The hand-written MFMA GEMM family behind every fc layer of the decoder
 // (reference op surface SURVEY.md §2.3: init/attend/decode MLPs, LSTM gate
-// GEMM).  CDNA4 structure per the gfx950 playbook:
-//   * v_mfma_f32_16x16x32_bf16 tiles, fp32 accumulation in AGPRs;
-//   * 128x128 block tile, BK=64, 4 waves of 64x64 each (4x4 fragments);
-//   * operands staged through LDS with +8-element row padding so the
-//     ds_read_b128 fragment reads spread across banks;
-//   * both A and B (weight stored [N,K]) read fragments as contiguous
-//     16-byte chunks — no transposes anywhere;
-//   * fused epilogue: bias add + tanh/relu + bf16 store.
-// Edge tiles (M, N, K not multiples of the tile) are guarded with zero-fill
-// loads / masked stores, so any shape with K%8==0 works.
+// GEMM).  Three shapes of work, three kernels:
+//
+//   * tiled_gemm<2,2,4,4>  — 128x128 tile, the large-M path (attention
+//     projection [B·196,512]x[512,512]);
+//   * tiled_gemm<4,1,2,4>  — 128x64 tile for N <= 512: twice the blocks,
+//     fills the 256-CU / 8-XCD chip at the flagship shapes;
+//   * skinny split-K       — M <= 32 (decode/LSTM GEMMs at batch 32):
+//     one 32-row tile, 16-column wave slabs, K split across blocks with
+//     fp32 atomic accumulation + a fused bias/act epilogue pass.  The
+//     128x128 kernel runs these shapes at 8-40 blocks (latency-bound,
+//     measured 40-60us each); the skinny kernel runs 128-316 blocks.
+//
+// CDNA4 structure per the gfx950 playbook: v_mfma_f32_16x16x32_bf16,
+// fp32 AGPR accumulation, LDS staging with +8 bf16 row padding (bank
+// spread for ds_read_b128 fragment reads), bf16x8 (16 B) global loads,
+// guarded edges so any K%8==0 shape works.
 
 #include "common.h"
 
-#define BM 128
-#define BN 128
 #define BK 64
-#define LDS_STRIDE (BK + 8)   // bf16 elements; row stride 144 B (16B-aligned)
+#define LDS_STRIDE (BK + 8)   // row stride 144 B (16 B aligned)
 
 #define ACT_NONE 0
 #define ACT_TANH 1
 #define ACT_RELU 2
 
+__device__ __forceinline__ float apply_act(float v, int act) {
+    if (act == ACT_TANH) return tanhf(v);
+    if (act == ACT_RELU) return fmaxf(v, 0.f);
+    return v;
+}
+
+// ---------------------------------------------------------------------
+// tiled kernel: BM = WR*FM*16, BN = WC*FN*16; 4 waves (WR*WC == 4)
+// ---------------------------------------------------------------------
+
+template <int WR, int WC, int FM, int FN>
 __global__ __launch_bounds__(256)
-void dense_fwd_kernel(const bf16* __restrict__ A,   // [M,K]
-                      const bf16* __restrict__ W,   // [N,K]
-                      const bf16* __restrict__ bias, // [N] or nullptr
-                      bf16* __restrict__ Y,          // [M,N]
-                      int M, int N, int K, int act) {
+void tiled_gemm_kernel(const bf16* __restrict__ A,    // [M,K]
+                       const bf16* __restrict__ W,    // [N,K]
+                       const bf16* __restrict__ bias, // [N] or null
+                       bf16* __restrict__ Y,          // [M,N]
+                       int M, int N, int K, int act) {
+    constexpr int BM = WR * FM * 16;
+    constexpr int BN = WC * FN * 16;
     __shared__ bf16 As[BM * LDS_STRIDE];
     __shared__ bf16 Bs[BN * LDS_STRIDE];
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
-    const int wave = tid >> 6;           // 0..3
-    const int wr = wave >> 1;            // wave row (0..1) -> 64-row slab
-    const int wc = wave & 1;             // wave col (0..1) -> 64-col slab
+    const int wave = tid >> 6;
+    const int wr = wave / WC;
+    const int wc = wave % WC;
 
     const int bm = blockIdx.y * BM;
     const int bn = blockIdx.x * BN;
 
-    floatx4 acc[4][4];
+    floatx4 acc[FM][FN];
 #pragma unroll
-    for (int i = 0; i < 4; ++i)
+    for (int i = 0; i < FM; ++i)
 #pragma unroll
-        for (int j = 0; j < 4; ++j)
+        for (int j = 0; j < FN; ++j)
             acc[i][j] = floatx4{0.f, 0.f, 0.f, 0.f};
 
-    const int lrow = lane & 15;          // fragment row within 16
-    const int kgrp = lane >> 4;          // 0..3 -> 8-element k chunk
+    const int lrow = lane & 15;
+    const int kgrp = lane >> 4;
+
+    constexpr int A_CHUNKS = BM * BK / 8 / 256;  // bf16x8 chunks per thread
+    constexpr int B_CHUNKS = BN * BK / 8 / 256;
 
     for (int k0 = 0; k0 < K; k0 += BK) {
-        // ---- stage A and B tiles (guarded, zero-filled) ----
-        // 128 rows x 64 cols = 8192 bf16 / 256 threads = 4 chunks of 8
 #pragma unroll
-        for (int i = 0; i < 4; ++i) {
-            int q = tid + 256 * i;       // chunk id 0..1023
-            int row = q >> 3;            // 8 chunks per row
-            int c8 = (q & 7) * 8;        // start col within tile
+        for (int i = 0; i < A_CHUNKS; ++i) {
+            int q = tid + 256 * i;
+            int row = q >> 3;
+            int c8 = (q & 7) * 8;
             int gk = k0 + c8;
-            bf16x8 av = {};
+            bf16x8 v = {};
             int ga = bm + row;
             if (ga < M && gk + 8 <= K) {
-                av = *(const bf16x8*)(A + (int64_t)ga * K + gk);
+                v = *(const bf16x8*)(A + (int64_t)ga * K + gk);
             } else if (ga < M) {
                 for (int e = 0; e < 8; ++e)
-                    if (gk + e < K) av[e] = A[(int64_t)ga * K + gk + e];
+                    if (gk + e < K) v[e] = A[(int64_t)ga * K + gk + e];
             }
-            *(bf16x8*)(As + row * LDS_STRIDE + c8) = av;
-
-            bf16x8 bv = {};
+            *(bf16x8*)(As + row * LDS_STRIDE + c8) = v;
+        }
+#pragma unroll
+        for (int i = 0; i < B_CHUNKS; ++i) {
+            int q = tid + 256 * i;
+            int row = q >> 3;
+            int c8 = (q & 7) * 8;
+            int gk = k0 + c8;
+            bf16x8 v = {};
             int gb = bn + row;
             if (gb < N && gk + 8 <= K) {
-                bv = *(const bf16x8*)(W + (int64_t)gb * K + gk);
+                v = *(const bf16x8*)(W + (int64_t)gb * K + gk);
             } else if (gb < N) {
                 for (int e = 0; e < 8; ++e)
-                    if (gk + e < K) bv[e] = W[(int64_t)gb * K + gk + e];
+                    if (gk + e < K) v[e] = W[(int64_t)gb * K + gk + e];
             }
-            *(bf16x8*)(Bs + row * LDS_STRIDE + c8) = bv;
+            *(bf16x8*)(Bs + row * LDS_STRIDE + c8) = v;
         }
         __syncthreads();
 
-        // ---- MFMA over the two 32-deep k-steps of this tile ----
 #pragma unroll
         for (int kk = 0; kk < BK / 32; ++kk) {
-            bf16x8 a_frag[4], b_frag[4];
+            bf16x8 a_frag[FM], b_frag[FN];
             const int kof = kk * 32 + kgrp * 8;
 #pragma unroll
-            for (int mi = 0; mi < 4; ++mi)
+            for (int mi = 0; mi < FM; ++mi)
                 a_frag[mi] = *(const bf16x8*)(
-                    As + (wr * 64 + mi * 16 + lrow) * LDS_STRIDE + kof);
+                    As + (wr * FM * 16 + mi * 16 + lrow) * LDS_STRIDE + kof);
 #pragma unroll
-            for (int ni = 0; ni < 4; ++ni)
+            for (int ni = 0; ni < FN; ++ni)
                 b_frag[ni] = *(const bf16x8*)(
-                    Bs + (wc * 64 + ni * 16 + lrow) * LDS_STRIDE + kof);
+                    Bs + (wc * FN * 16 + ni * 16 + lrow) * LDS_STRIDE + kof);
 #pragma unroll
-            for (int mi = 0; mi < 4; ++mi)
+            for (int mi = 0; mi < FM; ++mi)
 #pragma unroll
-                for (int ni = 0; ni < 4; ++ni)
+                for (int ni = 0; ni < FN; ++ni)
                     acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         a_frag[mi], b_frag[ni], acc[mi][ni], 0, 0, 0);
         }
         __syncthreads();
     }
 
-    // ---- epilogue: bias + activation + guarded bf16 store ----
-    // C/D lane map for 16x16: col = lane&15, row = (lane>>4)*4 + reg.
+    // epilogue (C/D map: col = lane&15, row = (lane>>4)*4 + reg)
 #pragma unroll
-    for (int ni = 0; ni < 4; ++ni) {
-        int col = bn + wc * 64 + ni * 16 + (lane & 15);
+    for (int ni = 0; ni < FN; ++ni) {
+        int col = bn + wc * FN * 16 + ni * 16 + (lane & 15);
         float bv = (bias != nullptr && col < N) ? bf2f(bias[col]) : 0.f;
 #pragma unroll
-        for (int mi = 0; mi < 4; ++mi) {
+        for (int mi = 0; mi < FM; ++mi) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                int row = bm + wr * 64 + mi * 16 + (lane >> 4) * 4 + r;
-                if (row < M && col < N) {
-                    float v = acc[mi][ni][r] + bv;
-                    if (act == ACT_TANH) v = tanhf(v);
-                    else if (act == ACT_RELU) v = fmaxf(v, 0.f);
-                    Y[(int64_t)row * N + col] = f2bf(v);
-                }
+                int row = bm + wr * FM * 16 + mi * 16 + (lane >> 4) * 4 + r;
+                if (row < M && col < N)
+                    Y[(int64_t)row * N + col] =
+                        f2bf(apply_act(acc[mi][ni][r] + bv, act));
             }
         }
     }
+}
+
+// ---------------------------------------------------------------------
+// skinny kernel: M <= 32; grid (ceil(N/64), SPLITK); 4 waves of 16 cols;
+// A fragments straight from global (x is L2-resident and tiny), W
+// fragments streamed from global.  SPLITK == 1 stores the final bf16;
+// otherwise fp32 atomics into a workspace + separate epilogue.
+// ---------------------------------------------------------------------
+
+__global__ __launch_bounds__(256)
+void skinny_gemm_kernel(const bf16* __restrict__ A,   // [M,K], M <= 32
+                        const bf16* __restrict__ W,   // [N,K]
+                        const bf16* __restrict__ bias,
+                        bf16* __restrict__ Y,         // [M,N] (splitk==1)
+                        float* __restrict__ Yf,       // [M,N] (splitk>1)
+                        int M, int N, int K, int act, int splitk) {
+    const int lane = threadIdx.x & 63;
+    const int wave = threadIdx.x >> 6;
+    const int n0 = blockIdx.x * 64 + wave * 16;
+    if (n0 >= N) return;
+
+    const int kq = blockIdx.y;
+    const int kchunk = ((K / 32 + splitk - 1) / splitk) * 32;
+    const int kbeg = kq * kchunk;
+    const int kend = min(K, kbeg + kchunk);
+
+    const int lrow = lane & 15;
+    const int kgrp = lane >> 4;
+
+    floatx4 acc[2];
+    acc[0] = floatx4{0.f, 0.f, 0.f, 0.f};
+    acc[1] = floatx4{0.f, 0.f, 0.f, 0.f};
+
+    const bool wfull = (n0 + 16 <= N);
+    for (int k = kbeg; k < kend; k += 32) {
+        const int kof = k + kgrp * 8;
+        bf16x8 b_frag = {};
+        if (kof + 8 <= K) {
+            int gb = n0 + lrow;
+            if (wfull || gb < N)
+                b_frag = *(const bf16x8*)(W + (int64_t)gb * K + kof);
+        } else if (kof < K) {
+            int gb = n0 + lrow;
+            if (wfull || gb < N)
+                for (int e = 0; e < 8 && kof + e < K; ++e)
+                    b_frag[e] = W[(int64_t)gb * K + kof + e];
+        }
+        bf16x8 a0 = {}, a1 = {};
+        if (kof + 8 <= K) {
+            if (lrow < M)
+                a0 = *(const bf16x8*)(A + (int64_t)lrow * K + kof);
+            if (16 + lrow < M)
+                a1 = *(const bf16x8*)(A + (int64_t)(16 + lrow) * K + kof);
+        } else if (kof < K) {
+            if (lrow < M)
+                for (int e = 0; e < 8 && kof + e < K; ++e)
+                    a0[e] = A[(int64_t)lrow * K + kof + e];
+            if (16 + lrow < M)
+                for (int e = 0; e < 8 && kof + e < K; ++e)
+                    a1[e] = A[(int64_t)(16 + lrow) * K + kof + e];
+        }
+        acc[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b_frag,
+                                                         acc[0], 0, 0, 0);
+        acc[1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b_frag,
+                                                         acc[1], 0, 0, 0);
+    }
+
+    const int col = n0 + (lane & 15);
+    if (col >= N) return;
+    if (splitk == 1) {
+        float bv = (bias != nullptr) ? bf2f(bias[col]) : 0.f;
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = mi * 16 + (lane >> 4) * 4 + r;
+                if (row < M)
+                    Y[(int64_t)row * N + col] =
+                        f2bf(apply_act(acc[mi][r] + bv, act));
+            }
+    } else {
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = mi * 16 + (lane >> 4) * 4 + r;
+                if (row < M)
+                    atomicAdd(Yf + (int64_t)row * N + col, acc[mi][r]);
+            }
+    }
+}
+
+__global__ void skinny_epilogue_kernel(const float* __restrict__ Yf,
+                                       const bf16* __restrict__ bias,
+                                       bf16* __restrict__ Y,
+                                       int64_t n, int N, int act) {
+    int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= n) return;
+    float v = Yf[idx];
+    if (bias != nullptr) v += bf2f(bias[idx % N]);
+    Y[idx] = f2bf(apply_act(v, act));
 }
 
 at::Tensor dense_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, int64_t act) {
@@ -142,12 +262,53 @@ at::Tensor dense_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, int64_t act) {
         bias_ptr = (const bf16*)bias.data_ptr();
     }
     auto y = at::empty({M, N}, x.options());
-    dim3 grid(cdiv(N, BN), cdiv(M, BM));
     hipStream_t stream = at::cuda::getCurrentCUDAStream();
-    hipLaunchKernelGGL(dense_fwd_kernel, grid, dim3(256), 0, stream,
-                       (const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(),
-                       bias_ptr, (bf16*)y.data_ptr(),
-                       (int)M, (int)N, (int)K, (int)act);
+
+    if (M <= 32 && K % 32 == 0) {
+        // skinny path with split-K chosen to fill the 256-CU chip
+        int nblocks = cdiv(N, 64);
+        int splitk = 1;
+        while (nblocks * splitk < 192 && splitk < 8 &&
+               (int)(K / 32) >= 2 * splitk)
+            splitk *= 2;
+        if (splitk == 1) {
+            hipLaunchKernelGGL(skinny_gemm_kernel, dim3(nblocks, 1),
+                               dim3(256), 0, stream,
+                               (const bf16*)x.data_ptr(),
+                               (const bf16*)w.data_ptr(), bias_ptr,
+                               (bf16*)y.data_ptr(), nullptr,
+                               (int)M, (int)N, (int)K, (int)act, 1);
+        } else {
+            auto yf = at::zeros({M, N}, x.options().dtype(at::kFloat));
+            hipLaunchKernelGGL(skinny_gemm_kernel, dim3(nblocks, splitk),
+                               dim3(256), 0, stream,
+                               (const bf16*)x.data_ptr(),
+                               (const bf16*)w.data_ptr(), nullptr,
+                               nullptr, (float*)yf.data_ptr(),
+                               (int)M, (int)N, (int)K, (int)act, splitk);
+            int64_t n = M * N;
+            hipLaunchKernelGGL(skinny_epilogue_kernel,
+                               dim3(cdiv(n, 256)), dim3(256), 0, stream,
+                               (const float*)yf.data_ptr(), bias_ptr,
+                               (bf16*)y.data_ptr(), n, (int)N, (int)act);
+        }
+    } else if (N <= 512) {
+        dim3 grid(cdiv(N, 64), cdiv(M, 128));
+        hipLaunchKernelGGL((tiled_gemm_kernel<4, 1, 2, 4>), grid, dim3(256),
+                           0, stream,
+                           (const bf16*)x.data_ptr(),
+                           (const bf16*)w.data_ptr(), bias_ptr,
+                           (bf16*)y.data_ptr(), (int)M, (int)N, (int)K,
+                           (int)act);
+    } else {
+        dim3 grid(cdiv(N, 128), cdiv(M, 128));
+        hipLaunchKernelGGL((tiled_gemm_kernel<2, 2, 4, 4>), grid, dim3(256),
+                           0, stream,
+                           (const bf16*)x.data_ptr(),
+                           (const bf16*)w.data_ptr(), bias_ptr,
+                           (bf16*)y.data_ptr(), (int)M, (int)N, (int)K,
+                           (int)act);
+    }
     HIP_OK(hipGetLastError());
     return y;
 }
