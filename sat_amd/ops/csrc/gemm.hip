@@ -226,13 +226,15 @@ void skinny_gemm_kernel(const bf16* __restrict__ A,   // [M,K], M <= 32
                         f2bf(apply_act(acc[mi][r] + bv, act));
             }
     } else {
+        // slab store: slice kq owns Yf[kq]; no atomics, no pre-zeroing
+        float* slab = Yf + (int64_t)kq * M * N;
 #pragma unroll
         for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 int row = mi * 16 + (lane >> 4) * 4 + r;
                 if (row < M)
-                    atomicAdd(Yf + (int64_t)row * N + col, acc[mi][r]);
+                    slab[(int64_t)row * N + col] = acc[mi][r];
             }
     }
 }
@@ -240,10 +242,12 @@ void skinny_gemm_kernel(const bf16* __restrict__ A,   // [M,K], M <= 32
 __global__ void skinny_epilogue_kernel(const float* __restrict__ Yf,
                                        const bf16* __restrict__ bias,
                                        bf16* __restrict__ Y,
-                                       int64_t n, int N, int act) {
+                                       int64_t n, int N, int act,
+                                       int splitk) {
     int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (idx >= n) return;
-    float v = Yf[idx];
+    float v = 0.f;
+    for (int k = 0; k < splitk; ++k) v += Yf[k * n + idx];
     if (bias != nullptr) v += bf2f(bias[idx % N]);
     Y[idx] = f2bf(apply_act(v, act));
 }
@@ -279,7 +283,8 @@ at::Tensor dense_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, int64_t act) {
                                (bf16*)y.data_ptr(), nullptr,
                                (int)M, (int)N, (int)K, (int)act, 1);
         } else {
-            auto yf = at::zeros({M, N}, x.options().dtype(at::kFloat));
+            auto yf = at::empty({splitk, M, N},
+                                x.options().dtype(at::kFloat));
             hipLaunchKernelGGL(skinny_gemm_kernel, dim3(nblocks, splitk),
                                dim3(256), 0, stream,
                                (const bf16*)x.data_ptr(),
@@ -290,7 +295,8 @@ at::Tensor dense_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, int64_t act) {
             hipLaunchKernelGGL(skinny_epilogue_kernel,
                                dim3(cdiv(n, 256)), dim3(256), 0, stream,
                                (const float*)yf.data_ptr(), bias_ptr,
-                               (bf16*)y.data_ptr(), n, (int)N, (int)act);
+                               (bf16*)y.data_ptr(), n, (int)N, (int)act,
+                               splitk);
         }
     } else if (N <= 512) {
         dim3 grid(cdiv(N, 64), cdiv(M, 128));

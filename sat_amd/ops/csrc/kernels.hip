@@ -253,63 +253,73 @@ std::vector<at::Tensor> attn_scores_fused(at::Tensor t1, at::Tensor t2,
 }
 
 // ---- attention pool backward ----
-// per image: s_l = Σ_d dpooled·ctx[l,d]; da = dα_ext + s;
-// dlogits = α (da − Σ_l α·da); optional dctx = α ⊗ dpooled.
+// phase 1, grid (B, 4): s_l = Σ_d dpooled·ctx[l,d] for an l-chunk (wave
+// per row, coalesced), plus dctx = α ⊗ dpooled for the same rows.
+// phase 2, grid (B): dlogits = α (dα_ext + s − Σ_l α·(dα_ext + s)).
 
-__global__ void attn_pool_bwd_kernel(
+__global__ void attn_pool_bwd_p1_kernel(
         const bf16* __restrict__ ctx, const float* __restrict__ alpha,
-        const float* __restrict__ dalpha, const bf16* __restrict__ dpooled,
-        float* __restrict__ dlogits, bf16* __restrict__ dctx,
+        const bf16* __restrict__ dpooled,
+        float* __restrict__ sbuf, bf16* __restrict__ dctx,
         int L, int D) {
-    __shared__ float sa[MAX_L];
-    __shared__ float red[4];
     int b = blockIdx.x;
-    int tid = threadIdx.x;
-    int wid = tid >> 6, lane = tid & 63;
+    int nchunk = gridDim.y;
+    int lchunk = (L + nchunk - 1) / nchunk;
+    int l0 = blockIdx.y * lchunk;
+    int l1 = min(L, l0 + lchunk);
+    int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
     const bf16* cb = ctx + (int64_t)b * L * D;
     const bf16* dp = dpooled + (int64_t)b * D;
 
-    for (int l = wid; l < L; l += 4) {
+    for (int l = l0 + wid; l < l1; l += 4) {
         float acc = 0.f;
+        float a = alpha[(int64_t)b * L + l];
         for (int d0 = lane * 8; d0 + 8 <= D; d0 += 64 * 8) {
             bf16x8 cv = *(const bf16x8*)(cb + (int64_t)l * D + d0);
             bf16x8 dv = *(const bf16x8*)(dp + d0);
+            if (dctx != nullptr) {
+                bf16x8 out;
 #pragma unroll
-            for (int e = 0; e < 8; ++e) acc += bf2f(cv[e]) * bf2f(dv[e]);
+                for (int e = 0; e < 8; ++e) {
+                    float dpe = bf2f(dv[e]);
+                    acc += bf2f(cv[e]) * dpe;
+                    out[e] = f2bf(a * dpe);
+                }
+                *(bf16x8*)(dctx + (int64_t)b * L * D + (int64_t)l * D + d0)
+                    = out;
+            } else {
+#pragma unroll
+                for (int e = 0; e < 8; ++e)
+                    acc += bf2f(cv[e]) * bf2f(dv[e]);
+            }
         }
         acc = wave_sum(acc);
-        if (lane == 0) {
-            float da = acc;
-            if (dalpha != nullptr) da += dalpha[(int64_t)b * L + l];
-            sa[l] = da;
-        }
+        if (lane == 0) sbuf[(int64_t)b * L + l] = acc;
     }
-    __syncthreads();
-    // dot = Σ α·da
+}
+
+__global__ void attn_pool_bwd_p2_kernel(
+        const float* __restrict__ alpha, const float* __restrict__ dalpha,
+        const float* __restrict__ sbuf, float* __restrict__ dlogits,
+        int L) {
+    __shared__ float red[4];
+    int b = blockIdx.x;
+    int tid = threadIdx.x;
     float part = 0.f;
-    for (int l = tid; l < L; l += blockDim.x)
-        part += alpha[(int64_t)b * L + l] * sa[l];
+    for (int l = tid; l < L; l += blockDim.x) {
+        float da = sbuf[(int64_t)b * L + l];
+        if (dalpha != nullptr) da += dalpha[(int64_t)b * L + l];
+        part += alpha[(int64_t)b * L + l] * da;
+    }
     part = wave_sum(part);
-    if ((tid & 63) == 0) red[wid] = part;
+    if ((tid & 63) == 0) red[tid >> 6] = part;
     __syncthreads();
     float dot = red[0] + red[1] + red[2] + red[3];
     for (int l = tid; l < L; l += blockDim.x) {
-        float a = alpha[(int64_t)b * L + l];
-        dlogits[(int64_t)b * L + l] = a * (sa[l] - dot);
-    }
-    if (dctx != nullptr) {
-        for (int l = wid; l < L; l += 4) {
-            float a = alpha[(int64_t)b * L + l];
-            for (int d0 = lane * 8; d0 + 8 <= D; d0 += 64 * 8) {
-                bf16x8 dv = *(const bf16x8*)(dp + d0);
-                bf16x8 out;
-#pragma unroll
-                for (int e = 0; e < 8; ++e)
-                    out[e] = f2bf(a * bf2f(dv[e]));
-                *(bf16x8*)(dctx + (int64_t)b * L * D + (int64_t)l * D + d0)
-                    = out;
-            }
-        }
+        float da = sbuf[(int64_t)b * L + l];
+        if (dalpha != nullptr) da += dalpha[(int64_t)b * L + l];
+        dlogits[(int64_t)b * L + l] =
+            alpha[(int64_t)b * L + l] * (da - dot);
     }
 }
 
@@ -321,6 +331,7 @@ std::vector<at::Tensor> attn_pool_bwd(at::Tensor ctx, at::Tensor alpha,
     int B = ctx.size(0), L = ctx.size(1), D = ctx.size(2);
     TORCH_CHECK(L <= MAX_L && D % 8 == 0);
     auto dlogits = at::empty({B, L}, alpha.options());
+    auto sbuf = at::empty({B, L}, alpha.options());
     at::Tensor dctx;
     bf16* dctx_ptr = nullptr;
     if (need_dctx) {
@@ -333,11 +344,15 @@ std::vector<at::Tensor> attn_pool_bwd(at::Tensor ctx, at::Tensor alpha,
     if (dalpha.defined() && dalpha.numel() > 0)
         dalpha_ptr = (const float*)dalpha.data_ptr();
     hipStream_t s = at::cuda::getCurrentCUDAStream();
-    hipLaunchKernelGGL(attn_pool_bwd_kernel, dim3(B), dim3(256), 0, s,
+    hipLaunchKernelGGL(attn_pool_bwd_p1_kernel, dim3(B, 4), dim3(256), 0, s,
                        (const bf16*)ctx.data_ptr(),
-                       (const float*)alpha.data_ptr(), dalpha_ptr,
+                       (const float*)alpha.data_ptr(),
                        (const bf16*)dpooled.data_ptr(),
-                       (float*)dlogits.data_ptr(), dctx_ptr, L, D);
+                       (float*)sbuf.data_ptr(), dctx_ptr, L, D);
+    hipLaunchKernelGGL(attn_pool_bwd_p2_kernel, dim3(B), dim3(256), 0, s,
+                       (const float*)alpha.data_ptr(), dalpha_ptr,
+                       (const float*)sbuf.data_ptr(),
+                       (float*)dlogits.data_ptr(), L);
     HIP_OK(hipGetLastError());
     return {dlogits, dctx};
 }
@@ -358,22 +373,32 @@ __global__ void attn_scores_bwd_kernel(
     int l0 = (blockIdx.x % nchunk) * lchunk;
     int l1 = min(L, l0 + lchunk);
     int tid = threadIdx.x;
-    for (int a = tid; a < A; a += blockDim.x) {
-        float va = bf2f(v[a]);
-        float dv_acc = 0.f;
-        float dt2_acc = 0.f;
-        for (int l = l0; l < l1; ++l) {
-            int64_t row = (int64_t)b * L + l;
-            float dl = dlogits[row];
+    // l outer / a inner: every load+store is a coalesced row sweep; each
+    // thread owns columns {tid, tid+256, ...} so dv/dt2 partials stay in
+    // registers until one atomicAdd per column at the end.
+    constexpr int MAX_AC = 8;   // supports A up to 2048
+    float dv_acc[MAX_AC] = {};
+    float dt2_acc[MAX_AC] = {};
+    for (int l = l0; l < l1; ++l) {
+        int64_t row = (int64_t)b * L + l;
+        float dl = dlogits[row];
+        for (int ai = 0; ai < MAX_AC; ++ai) {
+            int a = tid + ai * (int)blockDim.x;
+            if (a >= A) break;
+            float va = bf2f(v[a]);
             float td = bf2f(tdrop[row * A + a]);
-            dv_acc += td * dl;
+            dv_acc[ai] += td * dl;
             float dt = dl * va *
                 drop_scale(seed, salt, (uint32_t)(row * A + a), p);
             dt1[row * A + a] = f2bf(dt);
-            dt2_acc += dt;
+            dt2_acc[ai] += dt;
         }
-        atomicAdd(dvf + a, dv_acc);
-        atomicAdd(dt2 + (int64_t)b * A + a, dt2_acc);
+    }
+    for (int ai = 0; ai < MAX_AC; ++ai) {
+        int a = tid + ai * (int)blockDim.x;
+        if (a >= A) break;
+        atomicAdd(dvf + a, dv_acc[ai]);
+        atomicAdd(dt2 + (int64_t)b * A + a, dt2_acc[ai]);
     }
 }
 
@@ -382,6 +407,7 @@ std::vector<at::Tensor> attn_scores_bwd(at::Tensor tdrop, at::Tensor v,
                                         double p, int64_t salt, int64_t L) {
     CHECK_GPU(tdrop); CHECK_CONTIG(tdrop); CHECK_BF16(tdrop);
     int rows = tdrop.size(0), A = tdrop.size(1);
+    TORCH_CHECK(A <= 2048, "attn_scores_bwd supports A <= 2048");
     int B = rows / (int)L;
     auto dt1 = at::empty_like(tdrop);
     auto dt2 = at::zeros({B, A}, tdrop.options().dtype(at::kFloat));
@@ -402,17 +428,12 @@ std::vector<at::Tensor> attn_scores_bwd(at::Tensor tdrop, at::Tensor v,
     return {dt1, dt2, dvf};
 }
 
-__global__ void attn_pool_kernel(const bf16* __restrict__ ctx,   // [B,L,D]
-                                 const float* __restrict__ logits, // [B,L]
-                                 float* __restrict__ alpha,       // [B,L]
-                                 bf16* __restrict__ pooled,       // [B,D]
-                                 int L, int D) {
+__global__ void attn_softmax_kernel(const float* __restrict__ logits,
+                                    float* __restrict__ alpha, int L) {
     __shared__ float sa[MAX_L];
     __shared__ float red[8];
     int b = blockIdx.x;
     int tid = threadIdx.x;
-
-    // softmax over L (L <= MAX_L), staged in LDS
     float lmax = -1e30f;
     for (int l = tid; l < L; l += blockDim.x) {
         float x = logits[(int64_t)b * L + l];
@@ -425,7 +446,6 @@ __global__ void attn_pool_kernel(const bf16* __restrict__ ctx,   // [B,L,D]
     float m = -1e30f;
     for (int w = 0; w < (int)(blockDim.x >> 6); ++w) m = fmaxf(m, red[w]);
     __syncthreads();
-
     float lsum = 0.f;
     for (int l = tid; l < L; l += blockDim.x) {
         float e = __expf(sa[l] - m);
@@ -438,21 +458,36 @@ __global__ void attn_pool_kernel(const bf16* __restrict__ ctx,   // [B,L,D]
     float z = 0.f;
     for (int w = 0; w < (int)(blockDim.x >> 6); ++w) z += red[w];
     float inv = 1.0f / z;
-    __syncthreads();
-    for (int l = tid; l < L; l += blockDim.x) {
-        sa[l] *= inv;
-        alpha[(int64_t)b * L + l] = sa[l];
-    }
-    __syncthreads();
+    for (int l = tid; l < L; l += blockDim.x)
+        alpha[(int64_t)b * L + l] = sa[l] * inv;
+}
 
-    // pooled[d] = sum_l alpha[l] * ctx[l,d]; threads own columns
+// pooled[b, d] = sum_l alpha[b,l] * ctx[b,l,d]; grid (B, ceil(D/128));
+// 256 threads sweep two l-rows x 128 columns per iteration (coalesced).
+__global__ void attn_pool_sum_kernel(const bf16* __restrict__ ctx,
+                                     const float* __restrict__ alpha,
+                                     bf16* __restrict__ pooled,
+                                     int L, int D) {
+    __shared__ float sa[MAX_L];
+    int b = blockIdx.x;
+    int d0 = blockIdx.y * 128;
+    int tid = threadIdx.x;
+    for (int l = tid; l < L; l += blockDim.x)
+        sa[l] = alpha[(int64_t)b * L + l];
+    __syncthreads();
+    int d = d0 + (tid & 127);
+    int loff = tid >> 7;            // 0 or 1
+    if (d >= D) return;
     const bf16* cb = ctx + (int64_t)b * L * D;
-    for (int d = tid; d < D; d += blockDim.x) {
-        float acc = 0.f;
-        for (int l = 0; l < L; ++l)
-            acc += sa[l] * bf2f(cb[(int64_t)l * D + d]);
-        pooled[(int64_t)b * D + d] = f2bf(acc);
-    }
+    float acc = 0.f;
+    for (int l = loff; l < L; l += 2)
+        acc += sa[l] * bf2f(cb[(int64_t)l * D + d]);
+    // combine the two l-phases through LDS (reuse sa tail as scratch)
+    __shared__ float part[256];
+    part[tid] = acc;
+    __syncthreads();
+    if (loff == 0)
+        pooled[(int64_t)b * D + d] = f2bf(part[tid] + part[tid + 128]);
 }
 
 std::vector<at::Tensor> attn_pool_fwd(at::Tensor ctx, at::Tensor logits) {
@@ -463,11 +498,14 @@ std::vector<at::Tensor> attn_pool_fwd(at::Tensor ctx, at::Tensor logits) {
     auto alpha = at::empty({B, L}, ctx.options().dtype(at::kFloat));
     auto pooled = at::empty({B, D}, ctx.options());
     hipStream_t s = at::cuda::getCurrentCUDAStream();
-    hipLaunchKernelGGL(attn_pool_kernel, dim3(B), dim3(256), 0, s,
-                       (const bf16*)ctx.data_ptr(),
+    hipLaunchKernelGGL(attn_softmax_kernel, dim3(B), dim3(256), 0, s,
                        (const float*)logits.data_ptr(),
-                       (float*)alpha.data_ptr(), (bf16*)pooled.data_ptr(),
-                       L, D);
+                       (float*)alpha.data_ptr(), L);
+    hipLaunchKernelGGL(attn_pool_sum_kernel, dim3(B, cdiv(D, 128)),
+                       dim3(256), 0, s,
+                       (const bf16*)ctx.data_ptr(),
+                       (const float*)alpha.data_ptr(),
+                       (bf16*)pooled.data_ptr(), L, D);
     HIP_OK(hipGetLastError());
     return {alpha, pooled};
 }
