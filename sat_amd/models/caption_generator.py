@@ -57,6 +57,17 @@ class CaptionGenerator(tnn.Module):
                     == 'bf16')
                 else torch.float32)
 
+    def _use_bptt(self, contexts):
+        """The fused BPTT path covers the reference default architecture
+        (2-layer attend/decode MLPs) on GPU bf16."""
+        cfg = self.config
+        return (contexts.is_cuda
+                and contexts.dtype == torch.bfloat16
+                and getattr(cfg, 'use_hip_kernels', True)
+                and getattr(cfg, 'use_bptt', True)
+                and cfg.num_attend_layers == 2
+                and cfg.num_decode_layers == 2)
+
     def compute_contexts(self, images):
         """images: [B,3,224,224] float -> contexts [B,L,D].
 
@@ -95,33 +106,45 @@ class CaptionGenerator(tnn.Module):
         initial_memory, initial_output = self.decoder.initialize(
             context_mean)
 
-        memory, output = initial_memory, initial_output
-        state_h = output
-        last_word = torch.zeros(B, dtype=torch.int64,
-                                device=images.device)
-
-        cross_entropies = []
-        masked_alphas = []
-        num_correct = []
-        for t in range(T):
-            logits, alpha, memory, output, state_h = self.decoder.step(
-                contexts, contexts_flat, last_word, memory, output,
-                state_h, salt=t)
-            m = masks[:, t]
-            cross_entropies.append(
-                ops.masked_softmax_ce(logits, sentences[:, t], m))
-            masked_alphas.append(alpha.float() * m.unsqueeze(1))
-            with torch.no_grad():
-                pred = logits.argmax(dim=1)
-                num_correct.append(((pred == sentences[:, t]).float()
-                                    * m).sum())
-            last_word = sentences[:, t]
-
         mask_sum = masks.sum()
-        cross_entropy_loss = torch.stack(cross_entropies, dim=1).sum() \
-            / mask_sum
+        if self._use_bptt(contexts):
+            # fused hand-written BPTT over all T steps (sat_amd.models.bptt)
+            from .bptt import run_decoder_bptt
+            ce, attentions, predictions = run_decoder_bptt(
+                self.decoder, contexts, initial_memory, initial_output,
+                sentences, masks)
+            cross_entropy_loss = ce.sum() / mask_sum
+            with torch.no_grad():
+                num_correct = [((predictions == sentences).float()
+                                * masks).sum()]
+        else:
+            memory, output = initial_memory, initial_output
+            state_h = output
+            last_word = torch.zeros(B, dtype=torch.int64,
+                                    device=images.device)
 
-        attentions = torch.stack(masked_alphas, dim=2).sum(dim=2)  # [B,L]
+            cross_entropies = []
+            masked_alphas = []
+            num_correct = []
+            for t in range(T):
+                logits, alpha, memory, output, state_h = \
+                    self.decoder.step(
+                        contexts, contexts_flat, last_word, memory,
+                        output, state_h, salt=t)
+                m = masks[:, t]
+                cross_entropies.append(
+                    ops.masked_softmax_ce(logits, sentences[:, t], m))
+                masked_alphas.append(alpha.float() * m.unsqueeze(1))
+                with torch.no_grad():
+                    pred = logits.argmax(dim=1)
+                    num_correct.append(((pred == sentences[:, t]).float()
+                                        * m).sum())
+                last_word = sentences[:, t]
+
+            cross_entropy_loss = torch.stack(
+                cross_entropies, dim=1).sum() / mask_sum
+            attentions = torch.stack(masked_alphas, dim=2).sum(dim=2)
+
         diffs = 1.0 - attentions
         attention_loss = cfg.attention_loss_factor \
             * 0.5 * (diffs ** 2).sum() / (B * self.num_ctx)

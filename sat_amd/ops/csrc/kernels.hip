@@ -190,6 +190,49 @@ __device__ __forceinline__ float drop_scale(const int64_t* seed, int salt,
     return u >= p ? 1.0f / (1.0f - p) : 0.0f;
 }
 
+// ---- generic counter-based dropout (used by the hand-written BPTT) ----
+// Same hash as the attention tail: y = x * mask(seed,salt,idx)/(1-p).
+// Forward and backward are the same kernel (masks regenerate exactly).
+
+__global__ void hash_dropout_kernel(const bf16* __restrict__ x,
+                                    const int64_t* __restrict__ seed,
+                                    bf16* __restrict__ y,
+                                    int64_t n, float p, int salt) {
+    int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = idx * 8; i < n; i += stride * 8) {
+        if (i + 8 <= n) {
+            bf16x8 v = *(const bf16x8*)(x + i);
+            bf16x8 o;
+#pragma unroll
+            for (int e = 0; e < 8; ++e)
+                o[e] = f2bf(bf2f(v[e]) *
+                            drop_scale(seed, salt, (uint32_t)(i + e), p));
+            *(bf16x8*)(y + i) = o;
+        } else {
+            for (int64_t j = i; j < n; ++j)
+                y[j] = f2bf(bf2f(x[j]) *
+                            drop_scale(seed, salt, (uint32_t)j, p));
+        }
+    }
+}
+
+at::Tensor hash_dropout(at::Tensor x, at::Tensor seed, double p,
+                        int64_t salt) {
+    CHECK_GPU(x); CHECK_CONTIG(x); CHECK_BF16(x);
+    if (p <= 0.0) return x;
+    auto y = at::empty_like(x);
+    int64_t n = x.numel();
+    int blocks = (int)std::min<int64_t>(cdiv(n, 256 * 8), 4096);
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(hash_dropout_kernel, dim3(blocks), dim3(256), 0, s,
+                       (const bf16*)x.data_ptr(),
+                       (const int64_t*)seed.data_ptr(),
+                       (bf16*)y.data_ptr(), n, (float)p, (int)salt);
+    HIP_OK(hipGetLastError());
+    return y;
+}
+
 // ---- fused attention scores: t = dropout(t1 + t2), logits = t·v ----
 // t1: [B·L, A], t2: [B, A], v: [A]; writes tdrop (saved for dv in bwd)
 // and logits [B·L] fp32.  One wave per row.
