@@ -110,8 +110,8 @@ class DecoderBPTT(torch.autograd.Function):
             HD[sl] = hdrop
             logits = _C.dense_fwd(hdrop, wd2, bd2, ACT_NONE)
             LOGITS[sl] = logits
-            losses_t, lse_t = _C.ce_fwd(logits, sentences[:, t],
-                                        masks[:, t])
+            losses_t, lse_t = _C.ce_fwd(logits, labels_cat[sl],
+                                        masks_cat[sl])
             CE[sl] = losses_t
             LSE[sl] = lse_t
             preds.append(logits.argmax(dim=1))
@@ -131,7 +131,7 @@ class DecoderBPTT(torch.autograd.Function):
             memory = c_new
             output = out_t
             state_h = sth_t
-            last_word = sentences[:, t]
+            last_word = labels_cat[sl]
 
         ce = CE.reshape(T, B).t().contiguous()          # [B,T]
         predictions = torch.stack(preds, dim=1)          # [B,T]
