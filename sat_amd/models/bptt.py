@@ -66,18 +66,21 @@ class DecoderBPTT(torch.autograd.Function):
         CE = torch.empty(T * B, dtype=torch.float32, device=dev)
 
         t1s, t2s, tdrops, alphas = [], [], [], []
-        gates_l, cprev_l, hid_l, pooled_l, emb_l, out_l = [], [], [], [], \
-            [], []
+        gates_l, cprev_l, hid_l = [], [], []
         preds = []
 
         labels_cat = sentences.t().reshape(-1)          # [T·B] step-major
         masks_cat = masks.t().reshape(-1).contiguous()  # [T·B] float
 
         memory = init_memory
-        output = init_output
         state_h = init_output
         last_word = torch.zeros(B, dtype=torch.int64, device=dev)
         attn_acc = torch.zeros(B, L, dtype=torch.float32, device=dev)
+        empty_b = _EMPTY_B(dev)
+
+        # attend input for step 0 (later steps' come fused out of
+        # expand_fuse at step t-1)
+        _C.hash_dropout_out(init_output, seed, p_fc, 1, ODROP[0:B])
 
         for t in range(T):
             s = t * 16
@@ -85,31 +88,25 @@ class DecoderBPTT(torch.autograd.Function):
 
             cdrop = _drop(ctx_flat, seed, p_fc, s + 0)
             t1 = _C.dense_fwd(cdrop, w1a, b1a, ACT_TANH)
-            od = _drop(output, seed, p_fc, s + 1)
-            ODROP[sl] = od
-            t2 = _C.dense_fwd(od, w1b, b1b, ACT_TANH)
+            t2 = _C.dense_fwd(ODROP[sl], w1b, b1b, ACT_TANH)
             tdrop, att_logits = _C.attn_scores_fused(
                 t1, t2, v, seed, p_fc, s + 2, L)
             alpha, pooled = _C.attn_pool_fwd(contexts, att_logits)
 
             embt = _C.embedding_fwd(last_word, emb)
-            x = torch.cat([pooled, embt], dim=1)
-            xdrop = _drop(x, seed, p_lstm, s + 3)
-            xh = torch.cat([xdrop, state_h], dim=1)
-            XH[sl] = xh
-            gates = _C.dense_fwd(xh, wl, bl, ACT_NONE)
+            _C.lstm_in_fuse(pooled, embt, state_h, seed, p_lstm, s + 3,
+                            XH[sl])
+            gates = _C.dense_fwd(XH[sl], wl, bl, ACT_NONE)
             h_raw, c_new = _C.lstm_pointwise_fwd(gates, memory, 1.0)
-            out_t = _drop(h_raw, seed, p_lstm, s + 4)
-            sth_t = _drop(h_raw, seed, p_lstm, s + 5)
-
-            expanded = torch.cat([out_t, pooled, embt], dim=1)
-            expdrop = _drop(expanded, seed, p_fc, s + 6)
-            EXPD[sl] = expdrop
-            hid = _C.dense_fwd(expdrop, wd1, bd1, ACT_TANH)
-            hdrop = _drop(hid, seed, p_fc, s + 7)
-            HD[sl] = hdrop
-            logits = _C.dense_fwd(hdrop, wd2, bd2, ACT_NONE)
-            LOGITS[sl] = logits
+            od_next = ODROP[(t + 1) * B:(t + 2) * B] if t + 1 < T \
+                else empty_b
+            out_t, sth_t = _C.expand_fuse(
+                h_raw, pooled, embt, seed, EXPD[sl], od_next,
+                p_lstm, p_fc, s)
+            hid = _C.dense_fwd(EXPD[sl], wd1, bd1, ACT_TANH)
+            _C.hash_dropout_out(hid, seed, p_fc, s + 7, HD[sl])
+            logits = _C.dense_fwd_out(HD[sl], wd2, bd2, ACT_NONE,
+                                      LOGITS[sl])
             losses_t, lse_t = _C.ce_fwd(logits, labels_cat[sl],
                                         masks_cat[sl])
             CE[sl] = losses_t
@@ -124,12 +121,8 @@ class DecoderBPTT(torch.autograd.Function):
             gates_l.append(gates)
             cprev_l.append(memory)
             hid_l.append(hid)
-            pooled_l.append(pooled)
-            emb_l.append(embt)
-            out_l.append(out_t)
 
             memory = c_new
-            output = out_t
             state_h = sth_t
             last_word = labels_cat[sl]
 
@@ -141,12 +134,11 @@ class DecoderBPTT(torch.autograd.Function):
             wd2, bd2, seed, XH, EXPD, HD, ODROP, LOGITS, LSE,
             labels_cat, masks_cat, masks)
         ctx_ag.saved_lists = (t1s, t2s, tdrops, alphas, gates_l, cprev_l,
-                              hid_l, pooled_l, emb_l, out_l)
+                              hid_l)
         ctx_ag.dims = (B, L, D, T, A, H, E, V, Dd, I)
         ctx_ag.p_fc = p_fc
         ctx_ag.p_lstm = p_lstm
         ctx_ag.train_cnn = train_cnn
-        ctx_ag.init_output = init_output
         ctx_ag.mark_non_differentiable(predictions)
         return ce, attn_acc, predictions
 
@@ -155,8 +147,8 @@ class DecoderBPTT(torch.autograd.Function):
         (contexts, emb, w1a, b1a, w1b, b1b, v, wl, bl, wd1, bd1,
          wd2, bd2, seed, XH, EXPD, HD, ODROP, LOGITS, LSE,
          labels_cat, masks_cat, masks) = ctx_ag.saved_tensors
-        (t1s, t2s, tdrops, alphas, gates_l, cprev_l, hid_l, pooled_l,
-         emb_l, out_l) = ctx_ag.saved_lists
+        (t1s, t2s, tdrops, alphas, gates_l, cprev_l,
+         hid_l) = ctx_ag.saved_lists
         B, L, D, T, A, H, E, V, Dd, I = ctx_ag.dims
         p_fc = ctx_ag.p_fc
         p_lstm = ctx_ag.p_lstm
@@ -198,33 +190,21 @@ class DecoderBPTT(torch.autograd.Function):
         for t in range(T - 1, -1, -1):
             s = t * 16
             sl = slice(t * B, (t + 1) * B)
-            dexpanded = _drop(DEXPD[sl].contiguous(), seed, p_fc, s + 6)
-            dout_dec = dexpanded[:, :H]
-            dpool_dec = dexpanded[:, H:H + D]
-            demb_dec = dexpanded[:, H + D:]
-
-            d_out_total = (dout_dec + d_out_carry).contiguous()
-            dh_raw = _drop(d_out_total, seed, p_lstm, s + 4)
-            if p_lstm > 0.0:
-                dh_raw = dh_raw + _drop(d_sth_carry, seed, p_lstm, s + 5)
-            else:
-                dh_raw = dh_raw + d_sth_carry
-            dgates, dc_prev = _C.lstm_pointwise_bwd(
-                gates_l[t], cprev_l[t], dh_raw.contiguous(), dc_carry, 1.0)
-            DG[sl] = dgates
+            dh_raw, dpool_dec, demb_dec = _C.dexp_fuse(
+                DEXPD[sl], d_out_carry, d_sth_carry, seed, p_fc, p_lstm,
+                s, D, E)
+            dgates, dc_prev = _C.lstm_pointwise_bwd_out(
+                gates_l[t], cprev_l[t], dh_raw, dc_carry, 1.0, DG[sl])
             dxh = _C.dense_fwd(dgates, wl_t, _EMPTY_B(dev), ACT_NONE)
-            d_sth_carry = dxh[:, I:].contiguous()
-            dx = _drop(dxh[:, :I].contiguous(), seed, p_lstm, s + 3)
-            dpool_lstm = dx[:, :D]
-            demb_lstm = dx[:, D:]
-            dpooled = (dpool_dec + dpool_lstm).contiguous()
-            DEMB[sl] = demb_dec + demb_lstm
+            dpooled, d_sth_carry = _C.dx_fuse(
+                dxh, dpool_dec, demb_dec, seed, DEMB[sl], p_lstm,
+                s + 3, H)
             dc_carry = dc_prev
 
             dalpha_t = d_attn * masks[:, t].unsqueeze(1)
             dlog_att, dctx_t = _C.attn_pool_bwd(
                 contexts, alphas[t], dalpha_t.contiguous(),
-                dpooled.to(torch.bfloat16), need_dctx)
+                dpooled, need_dctx)
             if need_dctx:
                 dctx_acc += dctx_t
             dt1, dt2f, dvf = _C.attn_scores_bwd(

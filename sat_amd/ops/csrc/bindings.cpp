@@ -3,6 +3,28 @@
 #include <vector>
 
 at::Tensor dense_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, int64_t act);
+at::Tensor dense_fwd_out(at::Tensor x, at::Tensor w, at::Tensor bias,
+                         int64_t act, at::Tensor out);
+std::vector<at::Tensor> lstm_pointwise_bwd_out(at::Tensor gates,
+                                               at::Tensor c, at::Tensor dh,
+                                               at::Tensor dc, double fb,
+                                               at::Tensor dgates_out);
+void lstm_in_fuse(at::Tensor pooled, at::Tensor emb, at::Tensor sth,
+                  at::Tensor seed, double p, int64_t salt, at::Tensor xh);
+std::vector<at::Tensor> expand_fuse(at::Tensor h_raw, at::Tensor pooled,
+                                    at::Tensor emb, at::Tensor seed,
+                                    at::Tensor expdrop, at::Tensor od_next,
+                                    double p_lstm, double p_fc, int64_t s);
+std::vector<at::Tensor> dexp_fuse(at::Tensor dexpd, at::Tensor d_out_carry,
+                                  at::Tensor d_sth_carry, at::Tensor seed,
+                                  double p_fc, double p_lstm, int64_t s,
+                                  int64_t D, int64_t E);
+std::vector<at::Tensor> dx_fuse(at::Tensor dxh, at::Tensor dpool_dec,
+                                at::Tensor demb_dec, at::Tensor seed,
+                                at::Tensor demb_out, double p,
+                                int64_t salt, int64_t H);
+void hash_dropout_out(at::Tensor x, at::Tensor seed, double p,
+                      int64_t salt, at::Tensor out);
 std::vector<at::Tensor> lstm_pointwise_fwd(at::Tensor gates, at::Tensor c,
                                            double fb);
 std::vector<at::Tensor> lstm_pointwise_bwd(at::Tensor gates, at::Tensor c,
@@ -37,6 +59,13 @@ void adam_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("dense_fwd", &dense_fwd, "MFMA GEMM + bias/act (bf16)");
+    m.def("dense_fwd_out", &dense_fwd_out);
+    m.def("lstm_pointwise_bwd_out", &lstm_pointwise_bwd_out);
+    m.def("lstm_in_fuse", &lstm_in_fuse);
+    m.def("expand_fuse", &expand_fuse);
+    m.def("dexp_fuse", &dexp_fuse);
+    m.def("dx_fuse", &dx_fuse);
+    m.def("hash_dropout_out", &hash_dropout_out);
     m.def("lstm_pointwise_fwd", &lstm_pointwise_fwd);
     m.def("lstm_pointwise_bwd", &lstm_pointwise_bwd);
     m.def("act_bwd", &act_bwd);

@@ -252,7 +252,8 @@ __global__ void skinny_epilogue_kernel(const float* __restrict__ Yf,
     Y[idx] = f2bf(apply_act(v, act));
 }
 
-at::Tensor dense_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, int64_t act) {
+at::Tensor dense_fwd_out(at::Tensor x, at::Tensor w, at::Tensor bias,
+                         int64_t act, at::Tensor out) {
     CHECK_GPU(x); CHECK_CONTIG(x); CHECK_BF16(x);
     CHECK_GPU(w); CHECK_CONTIG(w); CHECK_BF16(w);
     TORCH_CHECK(x.dim() == 2 && w.dim() == 2, "dense_fwd expects 2-D inputs");
@@ -265,7 +266,14 @@ at::Tensor dense_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, int64_t act) {
         TORCH_CHECK(bias.numel() == N, "bias size mismatch");
         bias_ptr = (const bf16*)bias.data_ptr();
     }
-    auto y = at::empty({M, N}, x.options());
+    at::Tensor y;
+    if (out.defined() && out.numel() > 0) {
+        CHECK_CONTIG(out); CHECK_BF16(out);
+        TORCH_CHECK(out.numel() == M * N, "out size mismatch");
+        y = out;
+    } else {
+        y = at::empty({M, N}, x.options());
+    }
     hipStream_t stream = at::cuda::getCurrentCUDAStream();
 
     if (M <= 32 && K % 32 == 0) {
@@ -317,4 +325,9 @@ at::Tensor dense_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, int64_t act) {
     }
     HIP_OK(hipGetLastError());
     return y;
+}
+
+at::Tensor dense_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
+                     int64_t act) {
+    return dense_fwd_out(x, w, bias, act, at::Tensor());
 }

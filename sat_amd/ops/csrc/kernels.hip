@@ -77,12 +77,17 @@ std::vector<at::Tensor> lstm_pointwise_fwd(at::Tensor gates, at::Tensor c,
     return {h_out, c_out};
 }
 
-std::vector<at::Tensor> lstm_pointwise_bwd(at::Tensor gates, at::Tensor c,
-                                           at::Tensor dh, at::Tensor dc,
-                                           double fb) {
+std::vector<at::Tensor> lstm_pointwise_bwd_out(at::Tensor gates,
+                                               at::Tensor c, at::Tensor dh,
+                                               at::Tensor dc, double fb,
+                                               at::Tensor dgates_out) {
     CHECK_GPU(gates); CHECK_CONTIG(gates); CHECK_BF16(gates);
     int B = c.size(0), H = c.size(1);
-    auto dgates = at::empty_like(gates);
+    at::Tensor dgates;
+    if (dgates_out.defined() && dgates_out.numel() > 0)
+        dgates = dgates_out;
+    else
+        dgates = at::empty_like(gates);
     auto dc_prev = at::empty_like(c);
     const bf16* dc_ptr = nullptr;
     if (dc.defined() && dc.numel() > 0) dc_ptr = (const bf16*)dc.data_ptr();
@@ -95,6 +100,12 @@ std::vector<at::Tensor> lstm_pointwise_bwd(at::Tensor gates, at::Tensor c,
                        B, H, (float)fb);
     HIP_OK(hipGetLastError());
     return {dgates, dc_prev};
+}
+
+std::vector<at::Tensor> lstm_pointwise_bwd(at::Tensor gates, at::Tensor c,
+                                           at::Tensor dh, at::Tensor dc,
+                                           double fb) {
+    return lstm_pointwise_bwd_out(gates, c, dh, dc, fb, at::Tensor());
 }
 
 // ======================================================================
@@ -182,10 +193,10 @@ __device__ __forceinline__ uint32_t mix3(uint32_t a, uint32_t b, uint32_t c) {
     return h;
 }
 
-__device__ __forceinline__ float drop_scale(const int64_t* seed, int salt,
+__device__ __forceinline__ float drop_scale(uint32_t seed, int salt,
                                             uint32_t idx, float p) {
     if (p <= 0.f) return 1.f;
-    uint32_t h = mix3((uint32_t)(*seed), (uint32_t)salt, idx);
+    uint32_t h = mix3(seed, (uint32_t)salt, idx);
     float u = (h >> 8) * (1.0f / 16777216.0f);
     return u >= p ? 1.0f / (1.0f - p) : 0.0f;
 }
@@ -195,9 +206,10 @@ __device__ __forceinline__ float drop_scale(const int64_t* seed, int salt,
 // Forward and backward are the same kernel (masks regenerate exactly).
 
 __global__ void hash_dropout_kernel(const bf16* __restrict__ x,
-                                    const int64_t* __restrict__ seed,
+                                    const int64_t* __restrict__ seed_p,
                                     bf16* __restrict__ y,
                                     int64_t n, float p, int salt) {
+    const uint32_t seed = (uint32_t)(*seed_p);
     int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t i = idx * 8; i < n; i += stride * 8) {
@@ -239,9 +251,10 @@ at::Tensor hash_dropout(at::Tensor x, at::Tensor seed, double p,
 
 __global__ void attn_scores_fused_kernel(
         const bf16* __restrict__ t1, const bf16* __restrict__ t2,
-        const bf16* __restrict__ v, const int64_t* __restrict__ seed,
+        const bf16* __restrict__ v, const int64_t* __restrict__ seed_p,
         bf16* __restrict__ tdrop, float* __restrict__ logits,
         int B, int L, int A, float p, int salt) {
+    const uint32_t seed = (uint32_t)(*seed_p);
     int row = blockIdx.x * 4 + (threadIdx.x >> 6);
     if (row >= B * L) return;
     int lane = threadIdx.x & 63;
@@ -407,10 +420,11 @@ std::vector<at::Tensor> attn_pool_bwd(at::Tensor ctx, at::Tensor alpha,
 __global__ void attn_scores_bwd_kernel(
         const bf16* __restrict__ tdrop, const bf16* __restrict__ v,
         const float* __restrict__ dlogits,
-        const int64_t* __restrict__ seed,
+        const int64_t* __restrict__ seed_p,
         bf16* __restrict__ dt1, float* __restrict__ dt2,
         float* __restrict__ dvf,
         int B, int L, int A, int lchunk, float p, int salt) {
+    const uint32_t seed = (uint32_t)(*seed_p);
     int nchunk = (L + lchunk - 1) / lchunk;
     int b = blockIdx.x / nchunk;
     int l0 = (blockIdx.x % nchunk) * lchunk;

@@ -29,6 +29,7 @@ setup(
                 os.path.join(CSRC, 'bindings.cpp'),
                 os.path.join(CSRC, 'gemm.hip'),
                 os.path.join(CSRC, 'kernels.hip'),
+                os.path.join(CSRC, 'bptt_fuse.hip'),
             ],
             extra_compile_args={
                 'cxx': ['-O3', '-std=c++17'],
