@@ -62,8 +62,6 @@ class DecoderBPTT(torch.autograd.Function):
         HD = torch.empty(T * B, Dd, dtype=torch.bfloat16, device=dev)
         ODROP = torch.empty(T * B, H, dtype=torch.bfloat16, device=dev)
         LOGITS = torch.empty(T * B, V, dtype=torch.bfloat16, device=dev)
-        LSE = torch.empty(T * B, dtype=torch.float32, device=dev)
-        CE = torch.empty(T * B, dtype=torch.float32, device=dev)
 
         t1s, t2s, tdrops, alphas = [], [], [], []
         gates_l, cprev_l, hid_l = [], [], []
@@ -75,7 +73,6 @@ class DecoderBPTT(torch.autograd.Function):
         memory = init_memory
         state_h = init_output
         last_word = torch.zeros(B, dtype=torch.int64, device=dev)
-        attn_acc = torch.zeros(B, L, dtype=torch.float32, device=dev)
         empty_b = _EMPTY_B(dev)
 
         # attend input for step 0 (later steps' come fused out of
@@ -105,14 +102,7 @@ class DecoderBPTT(torch.autograd.Function):
                 p_lstm, p_fc, s)
             hid = _C.dense_fwd(EXPD[sl], wd1, bd1, ACT_TANH)
             _C.hash_dropout_out(hid, seed, p_fc, s + 7, HD[sl])
-            logits = _C.dense_fwd_out(HD[sl], wd2, bd2, ACT_NONE,
-                                      LOGITS[sl])
-            losses_t, lse_t = _C.ce_fwd(logits, labels_cat[sl],
-                                        masks_cat[sl])
-            CE[sl] = losses_t
-            LSE[sl] = lse_t
-            preds.append(logits.argmax(dim=1))
-            attn_acc += alpha * masks[:, t].unsqueeze(1)
+            _C.dense_fwd_out(HD[sl], wd2, bd2, ACT_NONE, LOGITS[sl])
 
             t1s.append(t1)
             t2s.append(t2)
@@ -126,8 +116,14 @@ class DecoderBPTT(torch.autograd.Function):
             state_h = sth_t
             last_word = labels_cat[sl]
 
-        ce = CE.reshape(T, B).t().contiguous()          # [B,T]
-        predictions = torch.stack(preds, dim=1)          # [B,T]
+        # batched loss / argmax / attention accumulation (one pass each)
+        CE, LSE = _C.ce_fwd(LOGITS, labels_cat, masks_cat)
+        ce = CE.reshape(T, B).t().contiguous()           # [B,T]
+        predictions = LOGITS.reshape(T, B, V).argmax(dim=2) \
+            .t().contiguous()                            # [B,T]
+        alpha_stack = torch.stack(alphas)                # [T,B,L]
+        attn_acc = (alpha_stack
+                    * masks.t().reshape(T, B, 1)).sum(dim=0)
 
         ctx_ag.save_for_backward(
             contexts, emb, w1a, b1a, w1b, b1b, v, wl, bl, wd1, bd1,
@@ -207,9 +203,8 @@ class DecoderBPTT(torch.autograd.Function):
                 dpooled, need_dctx)
             if need_dctx:
                 dctx_acc += dctx_t
-            dt1, dt2f, dvf = _C.attn_scores_bwd(
-                tdrops[t], v, dlog_att, seed, p_fc, s + 2, L)
-            dv_acc += dvf
+            dt1, dt2f, _dv = _C.attn_scores_bwd_acc(
+                tdrops[t], v, dlog_att, seed, p_fc, s + 2, L, dv_acc)
 
             dpre1b = _C.act_bwd(dt2f.to(torch.bfloat16), t2s[t], ACT_TANH)
             DPRE1B[sl] = dpre1b

@@ -44,6 +44,11 @@ std::vector<at::Tensor> attn_pool_bwd(at::Tensor ctx, at::Tensor alpha,
 std::vector<at::Tensor> attn_scores_bwd(at::Tensor tdrop, at::Tensor v,
                                         at::Tensor dlogits, at::Tensor seed,
                                         double p, int64_t salt, int64_t L);
+std::vector<at::Tensor> attn_scores_bwd_acc(at::Tensor tdrop, at::Tensor v,
+                                            at::Tensor dlogits,
+                                            at::Tensor seed, double p,
+                                            int64_t salt, int64_t L,
+                                            at::Tensor dv_acc);
 at::Tensor embedding_fwd(at::Tensor ids, at::Tensor table);
 at::Tensor embedding_bwd(at::Tensor ids, at::Tensor dy, int64_t rows);
 std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor labels,
@@ -75,6 +80,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("attn_scores_fused", &attn_scores_fused);
     m.def("attn_pool_bwd", &attn_pool_bwd);
     m.def("attn_scores_bwd", &attn_scores_bwd);
+    m.def("attn_scores_bwd_acc", &attn_scores_bwd_acc);
     m.def("embedding_fwd", &embedding_fwd);
     m.def("embedding_bwd", &embedding_bwd);
     m.def("ce_fwd", &ce_fwd);

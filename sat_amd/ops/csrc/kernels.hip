@@ -459,16 +459,22 @@ __global__ void attn_scores_bwd_kernel(
     }
 }
 
-std::vector<at::Tensor> attn_scores_bwd(at::Tensor tdrop, at::Tensor v,
-                                        at::Tensor dlogits, at::Tensor seed,
-                                        double p, int64_t salt, int64_t L) {
+std::vector<at::Tensor> attn_scores_bwd_acc(at::Tensor tdrop, at::Tensor v,
+                                            at::Tensor dlogits,
+                                            at::Tensor seed, double p,
+                                            int64_t salt, int64_t L,
+                                            at::Tensor dv_acc) {
     CHECK_GPU(tdrop); CHECK_CONTIG(tdrop); CHECK_BF16(tdrop);
     int rows = tdrop.size(0), A = tdrop.size(1);
     TORCH_CHECK(A <= 2048, "attn_scores_bwd supports A <= 2048");
     int B = rows / (int)L;
     auto dt1 = at::empty_like(tdrop);
     auto dt2 = at::zeros({B, A}, tdrop.options().dtype(at::kFloat));
-    auto dvf = at::zeros({A}, tdrop.options().dtype(at::kFloat));
+    at::Tensor dvf;
+    if (dv_acc.defined() && dv_acc.numel() > 0)
+        dvf = dv_acc;   // caller-owned accumulator (atomicAdd accumulates)
+    else
+        dvf = at::zeros({A}, tdrop.options().dtype(at::kFloat));
     int lchunk = ((int)L + 3) / 4;
     int nchunk = ((int)L + lchunk - 1) / lchunk;
     hipStream_t s = at::cuda::getCurrentCUDAStream();
@@ -483,6 +489,13 @@ std::vector<at::Tensor> attn_scores_bwd(at::Tensor tdrop, at::Tensor v,
                        B, (int)L, A, lchunk, (float)p, (int)salt);
     HIP_OK(hipGetLastError());
     return {dt1, dt2, dvf};
+}
+
+std::vector<at::Tensor> attn_scores_bwd(at::Tensor tdrop, at::Tensor v,
+                                        at::Tensor dlogits, at::Tensor seed,
+                                        double p, int64_t salt, int64_t L) {
+    return attn_scores_bwd_acc(tdrop, v, dlogits, seed, p, salt, L,
+                               at::Tensor());
 }
 
 __global__ void attn_softmax_kernel(const float* __restrict__ logits,
