@@ -54,6 +54,12 @@ class DecoderBPTT(torch.autograd.Function):
         dev = contexts.device
 
         ctx_flat = contexts.reshape(B * L, D)
+        # all T context-dropout slabs in one kernel (salt t*16+0); the
+        # buffer also feeds the batched dW_1a GEMM in backward
+        if p_fc > 0.0:
+            CDROP = _C.hash_dropout_steps(ctx_flat, seed, p_fc, 0, 16, T)
+        else:
+            CDROP = None
 
         # forward-side batched buffers (consumed by backward's batched dW)
         XH = torch.empty(T * B, I + H, dtype=torch.bfloat16, device=dev)
@@ -83,7 +89,7 @@ class DecoderBPTT(torch.autograd.Function):
             s = t * 16
             sl = slice(t * B, (t + 1) * B)
 
-            cdrop = _drop(ctx_flat, seed, p_fc, s + 0)
+            cdrop = CDROP[t] if CDROP is not None else ctx_flat
             t1 = _C.dense_fwd(cdrop, w1a, b1a, ACT_TANH)
             t2 = _C.dense_fwd(ODROP[sl], w1b, b1b, ACT_TANH)
             tdrop, att_logits = _C.attn_scores_fused(
@@ -129,6 +135,7 @@ class DecoderBPTT(torch.autograd.Function):
             contexts, emb, w1a, b1a, w1b, b1b, v, wl, bl, wd1, bd1,
             wd2, bd2, seed, XH, EXPD, HD, ODROP, LOGITS, LSE,
             labels_cat, masks_cat, masks)
+        ctx_ag.cdrop = CDROP
         ctx_ag.saved_lists = (t1s, t2s, tdrops, alphas, gates_l, cprev_l,
                               hid_l)
         ctx_ag.dims = (B, L, D, T, A, H, E, V, Dd, I)
@@ -173,9 +180,10 @@ class DecoderBPTT(torch.autograd.Function):
         DG = torch.empty(T * B, 4 * H, dtype=torch.bfloat16, device=dev)
         DEMB = torch.empty(T * B, E, dtype=torch.bfloat16, device=dev)
         DPRE1B = torch.empty(T * B, A, dtype=torch.bfloat16, device=dev)
+        DPRE1A = torch.empty(T * B * L, A, dtype=torch.bfloat16,
+                             device=dev)
+        CDROP = ctx_ag.cdrop
 
-        dW1a = torch.zeros_like(w1a, dtype=torch.float32)
-        db1a = torch.zeros_like(b1a, dtype=torch.float32)
         dv_acc = torch.zeros_like(v, dtype=torch.float32)
         dctx_acc = torch.zeros_like(contexts) if need_dctx else None
 
@@ -211,16 +219,21 @@ class DecoderBPTT(torch.autograd.Function):
             dodrop = _C.dense_fwd(dpre1b, w1b_t, _EMPTY_B(dev), ACT_NONE)
             d_out_carry = _drop(dodrop, seed, p_fc, s + 1)
 
-            dpre1a = _C.act_bwd(dt1, t1s[t], ACT_TANH)
-            cdrop = _drop(ctx_flat, seed, p_fc, s + 0)
-            dW1a += dpre1a.t().matmul(cdrop).float()
-            db1a += dpre1a.float().sum(0)
+            sl_a = slice(t * B * L, (t + 1) * B * L)
+            _C.act_bwd_out(dt1, t1s[t], ACT_TANH, DPRE1A[sl_a])
             if need_dctx:
-                dcd = dpre1a.matmul(w1a)
+                dcd = DPRE1A[sl_a].matmul(w1a)
                 dctx_acc += _drop(dcd, seed, p_fc, s + 0) \
                     .reshape(B, L, D)
 
         # ---- batched weight grads ----
+        if CDROP is not None:
+            dW1a = DPRE1A.t().matmul(CDROP.reshape(T * B * L, D))
+        else:
+            # p_fc == 0: every step saw the same ctx_flat
+            dW1a = DPRE1A.reshape(T, B * L, A).sum(0).t() \
+                .matmul(ctx_flat)
+        db1a = DPRE1A.float().sum(0)
         dWl = DG.t().matmul(XH)
         dbl = DG.float().sum(0)
         dWd2 = DL.t().matmul(HD)

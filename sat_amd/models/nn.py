@@ -124,8 +124,17 @@ class Conv2d(tnn.Module):
     def forward(self, x):
         w = self.weight.to(x.dtype)
         b = self.bias.to(x.dtype) if self.bias is not None else None
-        x = _pad_same(x, self.kernel_size, self.stride)
-        y = torch.nn.functional.conv2d(x, w, b, stride=self.stride)
+        k, st = self.kernel_size, self.stride
+        ph = _same_pad(x.shape[2], k, st)
+        pw = _same_pad(x.shape[3], k, st)
+        if ph % 2 == 0 and pw % 2 == 0:
+            # symmetric SAME: no pad-copy kernel, pad inside the conv
+            y = torch.nn.functional.conv2d(x, w, b, stride=st,
+                                           padding=(ph // 2, pw // 2))
+        else:
+            x = torch.nn.functional.pad(
+                x, (pw // 2, pw - pw // 2, ph // 2, ph - ph // 2))
+            y = torch.nn.functional.conv2d(x, w, b, stride=st)
         if self.activation == 'relu':
             y = torch.relu(y)
         return y

@@ -25,6 +25,10 @@ std::vector<at::Tensor> dx_fuse(at::Tensor dxh, at::Tensor dpool_dec,
                                 int64_t salt, int64_t H);
 void hash_dropout_out(at::Tensor x, at::Tensor seed, double p,
                       int64_t salt, at::Tensor out);
+at::Tensor hash_dropout_steps(at::Tensor x, at::Tensor seed, double p,
+                              int64_t salt_base, int64_t salt_stride,
+                              int64_t T);
+void act_bwd_out(at::Tensor dy, at::Tensor y, int64_t act, at::Tensor out);
 std::vector<at::Tensor> lstm_pointwise_fwd(at::Tensor gates, at::Tensor c,
                                            double fb);
 std::vector<at::Tensor> lstm_pointwise_bwd(at::Tensor gates, at::Tensor c,
@@ -71,6 +75,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("dexp_fuse", &dexp_fuse);
     m.def("dx_fuse", &dx_fuse);
     m.def("hash_dropout_out", &hash_dropout_out);
+    m.def("hash_dropout_steps", &hash_dropout_steps);
+    m.def("act_bwd_out", &act_bwd_out);
     m.def("lstm_pointwise_fwd", &lstm_pointwise_fwd);
     m.def("lstm_pointwise_bwd", &lstm_pointwise_bwd);
     m.def("act_bwd", &act_bwd);

@@ -265,3 +265,65 @@ void hash_dropout_out(at::Tensor x, at::Tensor seed, double p,
                        (bf16*)out.data_ptr(), n, (float)p, (int)salt);
     HIP_OK(hipGetLastError());
 }
+
+// ---- generate all T step-dropouts of one tensor into a [T, n] buffer ----
+// slab t uses salt = salt_base + t*salt_stride with slab-local indices,
+// exactly matching T separate hash_dropout calls.
+
+__global__ void hash_dropout_steps_kernel(const bf16* __restrict__ x,
+                                          const int64_t* __restrict__ seed_p,
+                                          bf16* __restrict__ y,
+                                          int64_t n, int T, float p,
+                                          int salt_base, int salt_stride) {
+    const uint32_t seed = (uint32_t)(*seed_p);
+    int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t total = n * T;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = idx; i < total; i += stride) {
+        int t = (int)(i / n);
+        int64_t j = i % n;
+        y[i] = f2bf(bf2f(x[j]) *
+                    dscale(seed, salt_base + t * salt_stride,
+                           (uint32_t)j, p));
+    }
+}
+
+at::Tensor hash_dropout_steps(at::Tensor x, at::Tensor seed, double p,
+                              int64_t salt_base, int64_t salt_stride,
+                              int64_t T) {
+    int64_t n = x.numel();
+    auto y = at::empty({T, x.size(0), x.size(1)}, x.options());
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    int blocks = (int)std::min<int64_t>(cdiv(n * T, 256), 16384);
+    hipLaunchKernelGGL(hash_dropout_steps_kernel, dim3(blocks), dim3(256),
+                       0, s,
+                       (const bf16*)x.data_ptr(),
+                       (const int64_t*)seed.data_ptr(),
+                       (bf16*)y.data_ptr(), n, (int)T, (float)p,
+                       (int)salt_base, (int)salt_stride);
+    HIP_OK(hipGetLastError());
+    return y;
+}
+
+__global__ void act_bwd_out_kernel(const bf16* __restrict__ dy,
+                                   const bf16* __restrict__ y,
+                                   bf16* __restrict__ dpre,
+                                   int64_t n, int act) {
+    int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= n) return;
+    float g = bf2f(dy[idx]);
+    float yv = bf2f(y[idx]);
+    if (act == 1) g *= (1.f - yv * yv);
+    else if (act == 2) g *= (yv > 0.f ? 1.f : 0.f);
+    dpre[idx] = f2bf(g);
+}
+
+void act_bwd_out(at::Tensor dy, at::Tensor y, int64_t act, at::Tensor out) {
+    int64_t n = dy.numel();
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(act_bwd_out_kernel, dim3(cdiv(n, 256)), dim3(256),
+                       0, s,
+                       (const bf16*)dy.data_ptr(), (const bf16*)y.data_ptr(),
+                       (bf16*)out.data_ptr(), n, (int)act);
+    HIP_OK(hipGetLastError());
+}

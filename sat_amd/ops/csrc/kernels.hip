@@ -424,38 +424,47 @@ __global__ void attn_scores_bwd_kernel(
         bf16* __restrict__ dt1, float* __restrict__ dt2,
         float* __restrict__ dvf,
         int B, int L, int A, int lchunk, float p, int salt) {
+    // one wave per row stream (bf16x8 vector traffic); each lane owns the
+    // fixed column set {lane*8..lane*8+7} + multiples of 512, so dv/dt2
+    // partials accumulate in registers across the wave's rows.
     const uint32_t seed = (uint32_t)(*seed_p);
     int nchunk = (L + lchunk - 1) / lchunk;
     int b = blockIdx.x / nchunk;
     int l0 = (blockIdx.x % nchunk) * lchunk;
     int l1 = min(L, l0 + lchunk);
-    int tid = threadIdx.x;
-    // l outer / a inner: every load+store is a coalesced row sweep; each
-    // thread owns columns {tid, tid+256, ...} so dv/dt2 partials stay in
-    // registers until one atomicAdd per column at the end.
-    constexpr int MAX_AC = 8;   // supports A up to 2048
-    float dv_acc[MAX_AC] = {};
-    float dt2_acc[MAX_AC] = {};
-    for (int l = l0; l < l1; ++l) {
+    int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+
+    constexpr int MAX_CH = 4;        // A <= 2048
+    int nch = A / 512;               // bf16x8 chunks per lane (A % 512 == 0)
+    float dv_acc[MAX_CH][8] = {};
+    float dt2_acc[MAX_CH][8] = {};
+
+    for (int l = l0 + wid; l < l1; l += 4) {
         int64_t row = (int64_t)b * L + l;
         float dl = dlogits[row];
-        for (int ai = 0; ai < MAX_AC; ++ai) {
-            int a = tid + ai * (int)blockDim.x;
-            if (a >= A) break;
-            float va = bf2f(v[a]);
-            float td = bf2f(tdrop[row * A + a]);
-            dv_acc[ai] += td * dl;
-            float dt = dl * va *
-                drop_scale(seed, salt, (uint32_t)(row * A + a), p);
-            dt1[row * A + a] = f2bf(dt);
-            dt2_acc[ai] += dt;
+        for (int ch = 0; ch < nch; ++ch) {
+            int a0 = ch * 512 + lane * 8;
+            bf16x8 td = *(const bf16x8*)(tdrop + row * A + a0);
+            bf16x8 vv = *(const bf16x8*)(v + a0);
+            bf16x8 o;
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {
+                dv_acc[ch][e] += bf2f(td[e]) * dl;
+                float dt = dl * bf2f(vv[e]) *
+                    drop_scale(seed, salt, (uint32_t)(row * A + a0 + e), p);
+                o[e] = f2bf(dt);
+                dt2_acc[ch][e] += dt;
+            }
+            *(bf16x8*)(dt1 + row * A + a0) = o;
         }
     }
-    for (int ai = 0; ai < MAX_AC; ++ai) {
-        int a = tid + ai * (int)blockDim.x;
-        if (a >= A) break;
-        atomicAdd(dvf + a, dv_acc[ai]);
-        atomicAdd(dt2 + (int64_t)b * A + a, dt2_acc[ai]);
+    for (int ch = 0; ch < nch; ++ch) {
+        int a0 = ch * 512 + lane * 8;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+            atomicAdd(dvf + a0 + e, dv_acc[ch][e]);
+            atomicAdd(dt2 + (int64_t)b * A + a0 + e, dt2_acc[ch][e]);
+        }
     }
 }
 
@@ -466,7 +475,8 @@ std::vector<at::Tensor> attn_scores_bwd_acc(at::Tensor tdrop, at::Tensor v,
                                             at::Tensor dv_acc) {
     CHECK_GPU(tdrop); CHECK_CONTIG(tdrop); CHECK_BF16(tdrop);
     int rows = tdrop.size(0), A = tdrop.size(1);
-    TORCH_CHECK(A <= 2048, "attn_scores_bwd supports A <= 2048");
+    TORCH_CHECK(A % 512 == 0 && A <= 2048,
+                "attn_scores_bwd requires A % 512 == 0, A <= 2048");
     int B = rows / (int)L;
     auto dt1 = at::empty_like(tdrop);
     auto dt2 = at::zeros({B, A}, tdrop.options().dtype(at::kFloat));
@@ -538,26 +548,39 @@ __global__ void attn_pool_sum_kernel(const bf16* __restrict__ ctx,
                                      const float* __restrict__ alpha,
                                      bf16* __restrict__ pooled,
                                      int L, int D) {
+    // grid (B, ceil(D/128)); 256 threads = 16 l-streams x (16 threads x
+    // bf16x8) covering 128 columns; partials combined through LDS.
     __shared__ float sa[MAX_L];
+    __shared__ float part[16][128 + 4];
     int b = blockIdx.x;
     int d0 = blockIdx.y * 128;
     int tid = threadIdx.x;
     for (int l = tid; l < L; l += blockDim.x)
         sa[l] = alpha[(int64_t)b * L + l];
     __syncthreads();
-    int d = d0 + (tid & 127);
-    int loff = tid >> 7;            // 0 or 1
-    if (d >= D) return;
-    const bf16* cb = ctx + (int64_t)b * L * D;
-    float acc = 0.f;
-    for (int l = loff; l < L; l += 2)
-        acc += sa[l] * bf2f(cb[(int64_t)l * D + d]);
-    // combine the two l-phases through LDS (reuse sa tail as scratch)
-    __shared__ float part[256];
-    part[tid] = acc;
+    int dg = tid & 15;               // column-chunk within the 128
+    int lg = tid >> 4;               // l-stream 0..15
+    int d = d0 + dg * 8;
+    float acc[8] = {};
+    if (d < D) {
+        const bf16* cb = ctx + (int64_t)b * L * D;
+        for (int l = lg; l < L; l += 16) {
+            bf16x8 cv = *(const bf16x8*)(cb + (int64_t)l * D + d);
+            float a = sa[l];
+#pragma unroll
+            for (int e = 0; e < 8; ++e) acc[e] += a * bf2f(cv[e]);
+        }
+    }
+#pragma unroll
+    for (int e = 0; e < 8; ++e) part[lg][dg * 8 + e] = acc[e];
     __syncthreads();
-    if (loff == 0)
-        pooled[(int64_t)b * D + d] = f2bf(part[tid] + part[tid + 128]);
+    // threads 0..127 finalize one column each
+    if (tid < 128 && d0 + tid < D) {
+        float s = 0.f;
+#pragma unroll
+        for (int g = 0; g < 16; ++g) s += part[g][tid];
+        pooled[(int64_t)b * D + d0 + tid] = f2bf(s);
+    }
 }
 
 std::vector<at::Tensor> attn_pool_fwd(at::Tensor ctx, at::Tensor logits) {
