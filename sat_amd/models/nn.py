@@ -125,6 +125,22 @@ class Conv2d(tnn.Module):
         w = self.weight.to(x.dtype)
         b = self.bias.to(x.dtype) if self.bias is not None else None
         k, st = self.kernel_size, self.stride
+        if (x.is_cuda and x.dtype == torch.bfloat16
+                and not torch.is_grad_enabled()
+                and k == 3 and st == 1 and w.shape[1] == 3
+                and w.shape[0] % 8 == 0 and w.shape[0] <= 128
+                and x.is_contiguous(memory_format=torch.channels_last)):
+            # 3-channel first layer: MIOpen NHWC bf16 falls back to a
+            # 2.4 ms CK kernel here; the direct kernel is memory-bound
+            from ..ops import hip
+            if hip.available():
+                from sat_amd import _C
+                y = _C.conv3_fwd(
+                    x, w.contiguous(),
+                    b if b is not None
+                    else torch.empty(0, dtype=x.dtype, device=x.device),
+                    self.activation == 'relu')
+                return y
         ph = _same_pad(x.shape[2], k, st)
         pw = _same_pad(x.shape[3], k, st)
         if ph % 2 == 0 and pw % 2 == 0:

@@ -3,6 +3,8 @@
 #include <vector>
 
 at::Tensor dense_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, int64_t act);
+at::Tensor conv3_fwd(at::Tensor input, at::Tensor weight, at::Tensor bias,
+                     bool relu);
 at::Tensor dense_fwd_out(at::Tensor x, at::Tensor w, at::Tensor bias,
                          int64_t act, at::Tensor out);
 std::vector<at::Tensor> lstm_pointwise_bwd_out(at::Tensor gates,
@@ -68,6 +70,8 @@ void adam_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("dense_fwd", &dense_fwd, "MFMA GEMM + bias/act (bf16)");
+    m.def("conv3_fwd", &conv3_fwd,
+          "direct NHWC conv for 3-channel 3x3/s1 (VGG conv1_1)");
     m.def("dense_fwd_out", &dense_fwd_out);
     m.def("lstm_pointwise_bwd_out", &lstm_pointwise_bwd_out);
     m.def("lstm_in_fuse", &lstm_in_fuse);

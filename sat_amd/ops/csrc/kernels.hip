@@ -434,16 +434,19 @@ __global__ void attn_scores_bwd_kernel(
     int l1 = min(L, l0 + lchunk);
     int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
 
+    // fully unrolled compile-time ch loop (a runtime bound would make the
+    // accumulator arrays dynamically indexed -> scratch spill, 3x slower)
     constexpr int MAX_CH = 4;        // A <= 2048
-    int nch = A / 512;               // bf16x8 chunks per lane (A % 512 == 0)
     float dv_acc[MAX_CH][8] = {};
     float dt2_acc[MAX_CH][8] = {};
 
     for (int l = l0 + wid; l < l1; l += 4) {
         int64_t row = (int64_t)b * L + l;
         float dl = dlogits[row];
-        for (int ch = 0; ch < nch; ++ch) {
+#pragma unroll
+        for (int ch = 0; ch < MAX_CH; ++ch) {
             int a0 = ch * 512 + lane * 8;
+            if (a0 >= A) break;
             bf16x8 td = *(const bf16x8*)(tdrop + row * A + a0);
             bf16x8 vv = *(const bf16x8*)(v + a0);
             bf16x8 o;
@@ -458,8 +461,10 @@ __global__ void attn_scores_bwd_kernel(
             *(bf16x8*)(dt1 + row * A + a0) = o;
         }
     }
-    for (int ch = 0; ch < nch; ++ch) {
+#pragma unroll
+    for (int ch = 0; ch < MAX_CH; ++ch) {
         int a0 = ch * 512 + lane * 8;
+        if (a0 >= A) break;
 #pragma unroll
         for (int e = 0; e < 8; ++e) {
             atomicAdd(dvf + a0 + e, dv_acc[ch][e]);

@@ -30,6 +30,7 @@ setup(
                 os.path.join(CSRC, 'gemm.hip'),
                 os.path.join(CSRC, 'kernels.hip'),
                 os.path.join(CSRC, 'bptt_fuse.hip'),
+                os.path.join(CSRC, 'conv3.hip'),
             ],
             extra_compile_args={
                 'cxx': ['-O3', '-std=c++17'],

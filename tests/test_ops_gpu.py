@@ -270,3 +270,30 @@ def test_fused_optimizer_in_training(tiny_config):
     torch.cuda.synchronize()
     assert not torch.equal(before, p.detach())
     assert torch.isfinite(p).all()
+
+
+def test_conv3_fwd_matches_torch():
+    """Direct NHWC conv1_1 kernel vs torch.conv2d (fp32 reference)."""
+    from sat_amd import _C
+    torch.manual_seed(7)
+    B, H, W, C = 4, 57, 61, 64  # odd sizes exercise the halo guards
+    x = torch.randn(B, 3, H, W).to(DEV, torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    w = (torch.randn(C, 3, 3, 3) * 0.2).to(DEV, torch.bfloat16)
+    b = torch.randn(C).to(DEV, torch.bfloat16)
+    y = _C.conv3_fwd(x, w, b, True)
+    ref = torch.relu(torch.nn.functional.conv2d(
+        x.float(), w.float(), b.float(), padding=1))
+    assert y.shape == ref.shape
+    assert _rel_err(y, ref) < 2e-2
+
+
+def test_conv3_fwd_no_bias_no_relu():
+    from sat_amd import _C
+    x = torch.randn(2, 3, 16, 16).to(DEV, torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    w = (torch.randn(8, 3, 3, 3) * 0.2).to(DEV, torch.bfloat16)
+    y = _C.conv3_fwd(x, w, torch.empty(0, dtype=torch.bfloat16,
+                                       device=DEV), False)
+    ref = torch.nn.functional.conv2d(x.float(), w.float(), padding=1)
+    assert _rel_err(y, ref) < 2e-2
