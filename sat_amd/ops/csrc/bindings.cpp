@@ -62,6 +62,12 @@ std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor labels,
 at::Tensor ce_bwd(at::Tensor logits, at::Tensor labels, at::Tensor mask,
                   at::Tensor lse, at::Tensor dloss);
 at::Tensor grad_sq_norm(std::vector<at::Tensor> grads);
+at::Tensor sq_norm_mt(at::Tensor desc, at::Tensor cum, int64_t n_tensors,
+                      int64_t total);
+void adam_step_mt(at::Tensor desc, at::Tensor cum, int64_t n_tensors,
+                  int64_t total, at::Tensor step_dev, double lr0,
+                  double decay_factor, double steps_per_decay, double b1,
+                  double b2, double eps, double clip, at::Tensor gsq);
 void adam_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
                std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
                at::Tensor step_dev, double lr0, double decay_factor,
@@ -96,5 +102,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("ce_fwd", &ce_fwd);
     m.def("ce_bwd", &ce_bwd);
     m.def("grad_sq_norm", &grad_sq_norm);
+    m.def("sq_norm_mt", &sq_norm_mt);
+    m.def("adam_step_mt", &adam_step_mt);
     m.def("adam_step", &adam_step);
 }

@@ -20,6 +20,8 @@
 // guarded edges so any K%8==0 shape works.
 
 #include "common.h"
+#include <map>
+#include <mutex>
 
 #define BK 64
 #define LDS_STRIDE (BK + 8)   // row stride 144 B (16 B aligned)
@@ -158,13 +160,15 @@ __global__ __launch_bounds__(256)
 void skinny_gemm_kernel(const bf16* __restrict__ A,   // [M,K], M <= 32
                         const bf16* __restrict__ W,   // [N,K]
                         const bf16* __restrict__ bias,
-                        bf16* __restrict__ Y,         // [M,N] (splitk==1)
-                        float* __restrict__ Yf,       // [M,N] (splitk>1)
+                        bf16* __restrict__ Y,         // [M,N]
+                        float* __restrict__ Yf,       // slabs (splitk>1)
+                        uint32_t* __restrict__ cnt,   // per-tile monotonic
                         int M, int N, int K, int act, int splitk) {
+    __shared__ uint32_t last_flag;    // the ONLY shared object
     const int lane = threadIdx.x & 63;
     const int wave = threadIdx.x >> 6;
     const int n0 = blockIdx.x * 64 + wave * 16;
-    if (n0 >= N) return;
+    const bool active = (n0 < N);
 
     const int kq = blockIdx.y;
     const int kchunk = ((K / 32 + splitk - 1) / splitk) * 32;
@@ -179,42 +183,44 @@ void skinny_gemm_kernel(const bf16* __restrict__ A,   // [M,K], M <= 32
     acc[1] = floatx4{0.f, 0.f, 0.f, 0.f};
 
     const bool wfull = (n0 + 16 <= N);
-    for (int k = kbeg; k < kend; k += 32) {
-        const int kof = k + kgrp * 8;
-        bf16x8 b_frag = {};
-        if (kof + 8 <= K) {
-            int gb = n0 + lrow;
-            if (wfull || gb < N)
-                b_frag = *(const bf16x8*)(W + (int64_t)gb * K + kof);
-        } else if (kof < K) {
-            int gb = n0 + lrow;
-            if (wfull || gb < N)
-                for (int e = 0; e < 8 && kof + e < K; ++e)
-                    b_frag[e] = W[(int64_t)gb * K + kof + e];
+    if (active) {
+        for (int k = kbeg; k < kend; k += 32) {
+            const int kof = k + kgrp * 8;
+            bf16x8 b_frag = {};
+            if (kof + 8 <= K) {
+                int gb = n0 + lrow;
+                if (wfull || gb < N)
+                    b_frag = *(const bf16x8*)(W + (int64_t)gb * K + kof);
+            } else if (kof < K) {
+                int gb = n0 + lrow;
+                if (wfull || gb < N)
+                    for (int e = 0; e < 8 && kof + e < K; ++e)
+                        b_frag[e] = W[(int64_t)gb * K + kof + e];
+            }
+            bf16x8 a0 = {}, a1 = {};
+            if (kof + 8 <= K) {
+                if (lrow < M)
+                    a0 = *(const bf16x8*)(A + (int64_t)lrow * K + kof);
+                if (16 + lrow < M)
+                    a1 = *(const bf16x8*)(A + (int64_t)(16 + lrow) * K + kof);
+            } else if (kof < K) {
+                if (lrow < M)
+                    for (int e = 0; e < 8 && kof + e < K; ++e)
+                        a0[e] = A[(int64_t)lrow * K + kof + e];
+                if (16 + lrow < M)
+                    for (int e = 0; e < 8 && kof + e < K; ++e)
+                        a1[e] = A[(int64_t)(16 + lrow) * K + kof + e];
+            }
+            acc[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b_frag,
+                                                             acc[0], 0, 0, 0);
+            acc[1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b_frag,
+                                                             acc[1], 0, 0, 0);
         }
-        bf16x8 a0 = {}, a1 = {};
-        if (kof + 8 <= K) {
-            if (lrow < M)
-                a0 = *(const bf16x8*)(A + (int64_t)lrow * K + kof);
-            if (16 + lrow < M)
-                a1 = *(const bf16x8*)(A + (int64_t)(16 + lrow) * K + kof);
-        } else if (kof < K) {
-            if (lrow < M)
-                for (int e = 0; e < 8 && kof + e < K; ++e)
-                    a0[e] = A[(int64_t)lrow * K + kof + e];
-            if (16 + lrow < M)
-                for (int e = 0; e < 8 && kof + e < K; ++e)
-                    a1[e] = A[(int64_t)(16 + lrow) * K + kof + e];
-        }
-        acc[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b_frag,
-                                                         acc[0], 0, 0, 0);
-        acc[1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b_frag,
-                                                         acc[1], 0, 0, 0);
     }
 
     const int col = n0 + (lane & 15);
-    if (col >= N) return;
     if (splitk == 1) {
+        if (!active || col >= N) return;
         float bv = (bias != nullptr) ? bf2f(bias[col]) : 0.f;
 #pragma unroll
         for (int mi = 0; mi < 2; ++mi)
@@ -225,8 +231,12 @@ void skinny_gemm_kernel(const bf16* __restrict__ A,   // [M,K], M <= 32
                     Y[(int64_t)row * N + col] =
                         f2bf(apply_act(acc[mi][r] + bv, act));
             }
-    } else {
-        // slab store: slice kq owns Yf[kq]; no atomics, no pre-zeroing
+        return;
+    }
+
+    // ---- split-K slab store + in-launch combine (agent-scope
+    // release/acquire hand-off, monotonic per-tile ticket counter) ----
+    if (active && col < N) {
         float* slab = Yf + (int64_t)kq * M * N;
 #pragma unroll
         for (int mi = 0; mi < 2; ++mi)
@@ -236,6 +246,34 @@ void skinny_gemm_kernel(const bf16* __restrict__ A,   // [M,K], M <= 32
                 if (row < M)
                     slab[(int64_t)row * N + col] = acc[mi][r];
             }
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        uint32_t old = __hip_atomic_fetch_add(
+            &cnt[blockIdx.x], 1u, __ATOMIC_RELAXED,
+            __HIP_MEMORY_SCOPE_AGENT);
+        last_flag = (((old + 1) % (uint32_t)splitk) == 0u) ? 1u : 0u;
+    }
+    __syncthreads();
+    if (last_flag == 0u) return;
+    if (threadIdx.x == 0)
+        __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+    __syncthreads();
+
+    // reducer: this block sums all slabs for its 64-column tile
+    int ncols = min(64, N - blockIdx.x * 64);
+    int nel = M * ncols;
+    for (int i = threadIdx.x; i < nel; i += (int)blockDim.x) {
+        int row = i / ncols;
+        int c = blockIdx.x * 64 + (i % ncols);
+        float vsum = 0.f;
+        for (int q = 0; q < splitk; ++q)
+            vsum += Yf[(int64_t)q * M * N + (int64_t)row * N + c];
+        if (bias != nullptr) vsum += bf2f(bias[c]);
+        Y[(int64_t)row * N + c] = f2bf(apply_act(vsum, act));
     }
 }
 
@@ -288,24 +326,38 @@ at::Tensor dense_fwd_out(at::Tensor x, at::Tensor w, at::Tensor bias,
                                dim3(256), 0, stream,
                                (const bf16*)x.data_ptr(),
                                (const bf16*)w.data_ptr(), bias_ptr,
-                               (bf16*)y.data_ptr(), nullptr,
+                               (bf16*)y.data_ptr(), nullptr, nullptr,
                                (int)M, (int)N, (int)K, (int)act, 1);
         } else {
             auto yf = at::empty({splitk, M, N},
                                 x.options().dtype(at::kFloat));
+            // persistent per-device monotonic tile counters: the ticket
+            // check is modulo splitk, so no per-call zeroing is needed
+            static std::mutex cnt_mu;
+            static std::map<int, at::Tensor> cnt_map;
+            at::Tensor cnt;
+            {
+                std::lock_guard<std::mutex> g(cnt_mu);
+                int devi = x.device().index();
+                auto it = cnt_map.find(devi);
+                if (it == cnt_map.end() ||
+                    it->second.numel() < nblocks) {
+                    cnt = at::zeros({std::max(nblocks, 1024)},
+                                    x.options().dtype(at::kInt));
+                    cnt_map[devi] = cnt;
+                } else {
+                    cnt = it->second;
+                }
+            }
             hipLaunchKernelGGL(skinny_gemm_kernel, dim3(nblocks, splitk),
                                dim3(256), 0, stream,
                                (const bf16*)x.data_ptr(),
-                               (const bf16*)w.data_ptr(), nullptr,
-                               nullptr, (float*)yf.data_ptr(),
+                               (const bf16*)w.data_ptr(), bias_ptr,
+                               (bf16*)y.data_ptr(), (float*)yf.data_ptr(),
+                               (uint32_t*)cnt.data_ptr(),
                                (int)M, (int)N, (int)K, (int)act, splitk);
-            int64_t n = M * N;
-            hipLaunchKernelGGL(skinny_epilogue_kernel,
-                               dim3(cdiv(n, 256)), dim3(256), 0, stream,
-                               (const float*)yf.data_ptr(), bias_ptr,
-                               (bf16*)y.data_ptr(), n, (int)N, (int)act,
-                               splitk);
         }
+    } else if (N <= 512) {
     } else if (N <= 512) {
         dim3 grid(cdiv(N, 64), cdiv(M, 128));
         hipLaunchKernelGGL((tiled_gemm_kernel<4, 1, 2, 4>), grid, dim3(256),

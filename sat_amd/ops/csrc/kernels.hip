@@ -417,6 +417,7 @@ std::vector<at::Tensor> attn_pool_bwd(at::Tensor ctx, at::Tensor alpha,
 // dt1 = dlogit ⊗ v ⊙ mask/(1-p) (mask regenerated); dt2 = Σ_l dt1;
 // dv = Σ_rows tdrop·dlogit.  Blocks cover l-chunks within one image.
 
+template <int NCH>
 __global__ void attn_scores_bwd_kernel(
         const bf16* __restrict__ tdrop, const bf16* __restrict__ v,
         const float* __restrict__ dlogits,
@@ -424,9 +425,9 @@ __global__ void attn_scores_bwd_kernel(
         bf16* __restrict__ dt1, float* __restrict__ dt2,
         float* __restrict__ dvf,
         int B, int L, int A, int lchunk, float p, int salt) {
-    // one wave per row stream (bf16x8 vector traffic); each lane owns the
-    // fixed column set {lane*8..lane*8+7} + multiples of 512, so dv/dt2
-    // partials accumulate in registers across the wave's rows.
+    // one wave per row stream (bf16x8 traffic); lane owns fixed columns
+    // {lane*8 + 512*ch}; NCH = A/512 is a template arg so the partial
+    // arrays are statically indexed (registers, not scratch).
     const uint32_t seed = (uint32_t)(*seed_p);
     int nchunk = (L + lchunk - 1) / lchunk;
     int b = blockIdx.x / nchunk;
@@ -434,19 +435,15 @@ __global__ void attn_scores_bwd_kernel(
     int l1 = min(L, l0 + lchunk);
     int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
 
-    // fully unrolled compile-time ch loop (a runtime bound would make the
-    // accumulator arrays dynamically indexed -> scratch spill, 3x slower)
-    constexpr int MAX_CH = 4;        // A <= 2048
-    float dv_acc[MAX_CH][8] = {};
-    float dt2_acc[MAX_CH][8] = {};
+    float dv_acc[NCH][8] = {};
+    float dt2_acc[NCH][8] = {};
 
     for (int l = l0 + wid; l < l1; l += 4) {
         int64_t row = (int64_t)b * L + l;
         float dl = dlogits[row];
 #pragma unroll
-        for (int ch = 0; ch < MAX_CH; ++ch) {
+        for (int ch = 0; ch < NCH; ++ch) {
             int a0 = ch * 512 + lane * 8;
-            if (a0 >= A) break;
             bf16x8 td = *(const bf16x8*)(tdrop + row * A + a0);
             bf16x8 vv = *(const bf16x8*)(v + a0);
             bf16x8 o;
@@ -462,9 +459,8 @@ __global__ void attn_scores_bwd_kernel(
         }
     }
 #pragma unroll
-    for (int ch = 0; ch < MAX_CH; ++ch) {
+    for (int ch = 0; ch < NCH; ++ch) {
         int a0 = ch * 512 + lane * 8;
-        if (a0 >= A) break;
 #pragma unroll
         for (int e = 0; e < 8; ++e) {
             atomicAdd(dvf + a0 + e, dv_acc[ch][e]);
@@ -493,15 +489,23 @@ std::vector<at::Tensor> attn_scores_bwd_acc(at::Tensor tdrop, at::Tensor v,
     int lchunk = ((int)L + 3) / 4;
     int nchunk = ((int)L + lchunk - 1) / lchunk;
     hipStream_t s = at::cuda::getCurrentCUDAStream();
-    hipLaunchKernelGGL(attn_scores_bwd_kernel, dim3(B * nchunk),
-                       dim3(256), 0, s,
-                       (const bf16*)tdrop.data_ptr(),
-                       (const bf16*)v.data_ptr(),
-                       (const float*)dlogits.data_ptr(),
-                       (const int64_t*)seed.data_ptr(),
-                       (bf16*)dt1.data_ptr(), (float*)dt2.data_ptr(),
-                       (float*)dvf.data_ptr(),
-                       B, (int)L, A, lchunk, (float)p, (int)salt);
+#define LAUNCH_SB(NCH) \
+    hipLaunchKernelGGL((attn_scores_bwd_kernel<NCH>), dim3(B * nchunk), \
+                       dim3(256), 0, s, \
+                       (const bf16*)tdrop.data_ptr(), \
+                       (const bf16*)v.data_ptr(), \
+                       (const float*)dlogits.data_ptr(), \
+                       (const int64_t*)seed.data_ptr(), \
+                       (bf16*)dt1.data_ptr(), (float*)dt2.data_ptr(), \
+                       (float*)dvf.data_ptr(), \
+                       B, (int)L, A, lchunk, (float)p, (int)salt)
+    switch (A / 512) {
+        case 1: LAUNCH_SB(1); break;
+        case 2: LAUNCH_SB(2); break;
+        case 3: LAUNCH_SB(3); break;
+        default: LAUNCH_SB(4); break;
+    }
+#undef LAUNCH_SB
     HIP_OK(hipGetLastError());
     return {dt1, dt2, dvf};
 }
@@ -847,5 +851,120 @@ void adam_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
                            (float)steps_per_decay, (float)b1, (float)b2,
                            (float)eps, (float)clip);
     }
+    HIP_OK(hipGetLastError());
+}
+
+// ======================================================================
+// Multi-tensor Adam / grad-norm: ONE launch over the concatenation of all
+// parameter tensors.  desc: [n,4] int64 device pointers (p,g,m,v); cum:
+// [n+1] int64 cumulative numels.  Tensor lookup = binary search in LDS.
+// ======================================================================
+
+__global__ void sq_norm_mt_kernel(const int64_t* __restrict__ desc,
+                                  const int64_t* __restrict__ cum,
+                                  int n_tensors, int64_t total,
+                                  float* __restrict__ out) {
+    extern __shared__ int64_t scum[];
+    for (int i = threadIdx.x; i <= n_tensors; i += blockDim.x)
+        scum[i] = cum[i];
+    __syncthreads();
+    float acc = 0.f;
+    int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = idx; i < total; i += stride) {
+        int lo = 0, hi = n_tensors;
+        while (hi - lo > 1) {
+            int mid = (lo + hi) >> 1;
+            if (i >= scum[mid]) lo = mid; else hi = mid;
+        }
+        const float* g = (const float*)desc[lo * 4 + 1];
+        float gv = g[i - scum[lo]];
+        acc += gv * gv;
+    }
+    acc = wave_sum(acc);
+    __shared__ float red[4];
+    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = acc;
+    __syncthreads();
+    if (threadIdx.x == 0)
+        atomicAdd(out, red[0] + red[1] + red[2] + red[3]);
+}
+
+at::Tensor sq_norm_mt(at::Tensor desc, at::Tensor cum, int64_t n_tensors,
+                      int64_t total) {
+    auto out = at::zeros({1}, at::TensorOptions()
+                                  .dtype(at::kFloat)
+                                  .device(desc.device()));
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    int blocks = (int)std::min<int64_t>(cdiv(total, 256 * 8), 2048);
+    size_t shm = (n_tensors + 1) * sizeof(int64_t);
+    hipLaunchKernelGGL(sq_norm_mt_kernel, dim3(blocks), dim3(256), shm, s,
+                       (const int64_t*)desc.data_ptr(),
+                       (const int64_t*)cum.data_ptr(),
+                       (int)n_tensors, total, (float*)out.data_ptr());
+    HIP_OK(hipGetLastError());
+    return out;
+}
+
+__global__ void adam_mt_kernel(const int64_t* __restrict__ desc,
+                               const int64_t* __restrict__ cum,
+                               int n_tensors, int64_t total,
+                               const float* __restrict__ gsq,
+                               const float* __restrict__ step_dev,
+                               float lr0, float decay_factor,
+                               float steps_per_decay, float b1, float b2,
+                               float eps, float clip) {
+    extern __shared__ int64_t scum[];
+    for (int i = threadIdx.x; i <= n_tensors; i += blockDim.x)
+        scum[i] = cum[i];
+    __syncthreads();
+    float step = *step_dev;
+    float bc1 = 1.f - powf(b1, step);
+    float bc2 = 1.f - powf(b2, step);
+    float lr = lr0;
+    if (decay_factor < 1.f)
+        lr = lr0 * powf(decay_factor, floorf(step / steps_per_decay));
+    float scale = 1.f;
+    if (clip > 0.f) {
+        float norm = sqrtf(*gsq);
+        if (norm > clip) scale = clip / norm;
+    }
+    int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = idx; i < total; i += stride) {
+        int lo = 0, hi = n_tensors;
+        while (hi - lo > 1) {
+            int mid = (lo + hi) >> 1;
+            if (i >= scum[mid]) lo = mid; else hi = mid;
+        }
+        int64_t off = i - scum[lo];
+        float* p = (float*)desc[lo * 4 + 0];
+        const float* g = (const float*)desc[lo * 4 + 1];
+        float* m = (float*)desc[lo * 4 + 2];
+        float* v = (float*)desc[lo * 4 + 3];
+        float gi = g[off] * scale;
+        float mi = b1 * m[off] + (1.f - b1) * gi;
+        float vi = b2 * v[off] + (1.f - b2) * gi * gi;
+        m[off] = mi;
+        v[off] = vi;
+        p[off] -= lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+    }
+}
+
+void adam_step_mt(at::Tensor desc, at::Tensor cum, int64_t n_tensors,
+                  int64_t total, at::Tensor step_dev, double lr0,
+                  double decay_factor, double steps_per_decay, double b1,
+                  double b2, double eps, double clip, at::Tensor gsq) {
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    int blocks = (int)std::min<int64_t>(cdiv(total, 256 * 4), 4096);
+    size_t shm = (n_tensors + 1) * sizeof(int64_t);
+    hipLaunchKernelGGL(adam_mt_kernel, dim3(blocks), dim3(256), shm, s,
+                       (const int64_t*)desc.data_ptr(),
+                       (const int64_t*)cum.data_ptr(),
+                       (int)n_tensors, total,
+                       (const float*)gsq.data_ptr(),
+                       (const float*)step_dev.data_ptr(),
+                       (float)lr0, (float)decay_factor,
+                       (float)steps_per_decay, (float)b1, (float)b2,
+                       (float)eps, (float)clip);
     HIP_OK(hipGetLastError());
 }

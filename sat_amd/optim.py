@@ -63,23 +63,46 @@ class Optimizer(object):
         use_hip = (len(self.params) > 0 and self.params[0].is_cuda
                    and self.kind == 'Adam')
         if use_hip:
+            from sat_amd import _C
             from .ops import hip
             hip.require()
             # Everything stays on-device (grad-norm², clip scale, step
             # counter, bias correction, LR decay): zero host syncs, so the
-            # whole optimizer step is hipGraph-capturable.
+            # whole optimizer step is hipGraph-capturable.  All tensors go
+            # through ONE multi-tensor kernel via a device pointer table,
+            # rebuilt only when any data pointer changes.
             if not hasattr(self, 'step_dev') or self.step_dev is None:
                 self.step_dev = torch.zeros(
                     (), dtype=torch.float32,
                     device=self.params[0].device)
                 self.step_dev.fill_(float(self.step_count - 1))
             self.step_dev.add_(1.0)
-            gsq = hip.grad_sq_norm(grads)
-            hip.adam_step(
-                [p.data for p in self.params], grads,
-                [self.state[p]['m'] for p in self.params],
-                [self.state[p]['v'] for p in self.params],
-                self.step_dev, cfg.initial_learning_rate,
+            ptrs = tuple(
+                (p.data.data_ptr(), g.data_ptr(),
+                 self.state[p]['m'].data_ptr(),
+                 self.state[p]['v'].data_ptr())
+                for p, g in zip(self.params, grads))
+            if getattr(self, '_mt_ptrs', None) != ptrs:
+                dev = self.params[0].device
+                self._mt_ptrs = ptrs
+                self._mt_desc = torch.tensor(
+                    [list(row) for row in ptrs],
+                    dtype=torch.int64).to(dev)
+                numels = [p.numel() for p in self.params]
+                cum = [0]
+                for n in numels:
+                    cum.append(cum[-1] + n)
+                self._mt_cum = torch.tensor(
+                    cum, dtype=torch.int64).to(dev)
+                self._mt_total = cum[-1]
+                for g in grads:
+                    assert g.is_contiguous() and g.dtype == torch.float32
+            gsq = _C.sq_norm_mt(self._mt_desc, self._mt_cum,
+                                len(self.params), self._mt_total)
+            _C.adam_step_mt(
+                self._mt_desc, self._mt_cum, len(self.params),
+                self._mt_total, self.step_dev,
+                cfg.initial_learning_rate,
                 cfg.learning_rate_decay_factor, cfg.num_steps_per_decay,
                 cfg.beta1, cfg.beta2, cfg.epsilon,
                 cfg.clip_gradients, gsq)
