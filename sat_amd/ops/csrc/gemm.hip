@@ -358,7 +358,6 @@ at::Tensor dense_fwd_out(at::Tensor x, at::Tensor w, at::Tensor bias,
                                (int)M, (int)N, (int)K, (int)act, splitk);
         }
     } else if (N <= 512) {
-    } else if (N <= 512) {
         dim3 grid(cdiv(N, 64), cdiv(M, 128));
         hipLaunchKernelGGL((tiled_gemm_kernel<4, 1, 2, 4>), grid, dim3(256),
                            0, stream,
