@@ -247,6 +247,7 @@ void skinny_gemm_kernel(const bf16* __restrict__ A,   // [M,K], M <= 32
                     slab[(int64_t)row * N + col] = acc[mi][r];
             }
     }
+    if (cnt == nullptr) return;   // slab mode: a separate epilogue reduces
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
     if (threadIdx.x == 0) {
@@ -331,31 +332,18 @@ at::Tensor dense_fwd_out(at::Tensor x, at::Tensor w, at::Tensor bias,
         } else {
             auto yf = at::empty({splitk, M, N},
                                 x.options().dtype(at::kFloat));
-            // persistent per-device monotonic tile counters: the ticket
-            // check is modulo splitk, so no per-call zeroing is needed
-            static std::mutex cnt_mu;
-            static std::map<int, at::Tensor> cnt_map;
-            at::Tensor cnt;
-            {
-                std::lock_guard<std::mutex> g(cnt_mu);
-                int devi = x.device().index();
-                auto it = cnt_map.find(devi);
-                if (it == cnt_map.end() ||
-                    it->second.numel() < nblocks) {
-                    cnt = at::zeros({std::max(nblocks, 1024)},
-                                    x.options().dtype(at::kInt));
-                    cnt_map[devi] = cnt;
-                } else {
-                    cnt = it->second;
-                }
-            }
             hipLaunchKernelGGL(skinny_gemm_kernel, dim3(nblocks, splitk),
                                dim3(256), 0, stream,
                                (const bf16*)x.data_ptr(),
-                               (const bf16*)w.data_ptr(), bias_ptr,
-                               (bf16*)y.data_ptr(), (float*)yf.data_ptr(),
-                               (uint32_t*)cnt.data_ptr(),
+                               (const bf16*)w.data_ptr(), nullptr,
+                               nullptr, (float*)yf.data_ptr(), nullptr,
                                (int)M, (int)N, (int)K, (int)act, splitk);
+            int64_t n = M * N;
+            hipLaunchKernelGGL(skinny_epilogue_kernel,
+                               dim3(cdiv(n, 256)), dim3(256), 0, stream,
+                               (const float*)yf.data_ptr(), bias_ptr,
+                               (bf16*)y.data_ptr(), n, (int)N, (int)act,
+                               splitk);
         }
     } else if (N <= 512) {
         dim3 grid(cdiv(N, 64), cdiv(M, 128));

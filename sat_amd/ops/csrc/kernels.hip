@@ -193,6 +193,28 @@ __device__ __forceinline__ uint32_t mix3(uint32_t a, uint32_t b, uint32_t c) {
     return h;
 }
 
+// 8 masks from two hashes (one per 4 bytes): ~4x less ALU than 8
+// per-element hashes.  Used by the attention scores fwd/bwd PAIR only —
+// both sides must derive masks identically.
+__device__ __forceinline__ void drop_scale8(uint32_t seed, int salt,
+                                            uint32_t idx8, float p,
+                                            float* sc) {
+    if (p <= 0.f) {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) sc[e] = 1.f;
+        return;
+    }
+    uint32_t pq = (uint32_t)(p * 256.0f);
+    float inv = 1.0f / (1.0f - p);
+    uint32_t h1 = mix3(seed, (uint32_t)salt, idx8);
+    uint32_t h2 = mix3(h1, (uint32_t)salt ^ 0xA5A5A5A5u, idx8);
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+        sc[e] = ((h1 >> (e * 8)) & 0xFFu) >= pq ? inv : 0.f;
+        sc[4 + e] = ((h2 >> (e * 8)) & 0xFFu) >= pq ? inv : 0.f;
+    }
+}
+
 __device__ __forceinline__ float drop_scale(uint32_t seed, int salt,
                                             uint32_t idx, float p) {
     if (p <= 0.f) return 1.f;
@@ -268,10 +290,11 @@ __global__ void attn_scores_fused_kernel(
         bf16x8 x2 = *(const bf16x8*)(t2r + a0);
         bf16x8 vv = *(const bf16x8*)(v + a0);
         bf16x8 out;
+        float sc[8];
+        drop_scale8(seed, salt, (uint32_t)((row * A + a0) >> 3), p, sc);
 #pragma unroll
         for (int e = 0; e < 8; ++e) {
-            float t = bf2f(x1[e]) + bf2f(x2[e]);
-            t *= drop_scale(seed, salt, (uint32_t)(row * A + a0 + e), p);
+            float t = (bf2f(x1[e]) + bf2f(x2[e])) * sc[e];
             out[e] = f2bf(t);
             acc += t * bf2f(vv[e]);
         }
@@ -447,11 +470,12 @@ __global__ void attn_scores_bwd_kernel(
             bf16x8 td = *(const bf16x8*)(tdrop + row * A + a0);
             bf16x8 vv = *(const bf16x8*)(v + a0);
             bf16x8 o;
+            float sc[8];
+            drop_scale8(seed, salt, (uint32_t)((row * A + a0) >> 3), p, sc);
 #pragma unroll
             for (int e = 0; e < 8; ++e) {
                 dv_acc[ch][e] += bf2f(td[e]) * dl;
-                float dt = dl * bf2f(vv[e]) *
-                    drop_scale(seed, salt, (uint32_t)(row * A + a0 + e), p);
+                float dt = dl * bf2f(vv[e]) * sc[e];
                 o[e] = f2bf(dt);
                 dt2_acc[ch][e] += dt;
             }
@@ -486,7 +510,7 @@ std::vector<at::Tensor> attn_scores_bwd_acc(at::Tensor tdrop, at::Tensor v,
         dvf = dv_acc;   // caller-owned accumulator (atomicAdd accumulates)
     else
         dvf = at::zeros({A}, tdrop.options().dtype(at::kFloat));
-    int lchunk = ((int)L + 3) / 4;
+    int lchunk = ((int)L + 15) / 16;   // ~16 chunks/image -> 512 blocks
     int nchunk = ((int)L + lchunk - 1) / lchunk;
     hipStream_t s = at::cuda::getCurrentCUDAStream();
 #define LAUNCH_SB(NCH) \
