@@ -448,9 +448,12 @@ __global__ void attn_scores_bwd_kernel(
         bf16* __restrict__ dt1, float* __restrict__ dt2,
         float* __restrict__ dvf,
         int B, int L, int A, int lchunk, float p, int salt) {
-    // one wave per row stream (bf16x8 traffic); lane owns fixed columns
-    // {lane*8 + 512*ch}; NCH = A/512 is a template arg so the partial
-    // arrays are statically indexed (registers, not scratch).
+    // wave-per-row bf16x8 streaming; per-wave column partials are combined
+    // through LDS and flushed with ONE atomic per thread-column — a lane
+    // issuing 8 consecutive atomics serializes ~40x (measured 420us vs
+    // 10us for this shape).
+    __shared__ float sdv[4][NCH * 512];
+    __shared__ float sd2[4][NCH * 512];
     const uint32_t seed = (uint32_t)(*seed_p);
     int nchunk = (L + lchunk - 1) / lchunk;
     int b = blockIdx.x / nchunk;
@@ -487,9 +490,16 @@ __global__ void attn_scores_bwd_kernel(
         int a0 = ch * 512 + lane * 8;
 #pragma unroll
         for (int e = 0; e < 8; ++e) {
-            atomicAdd(dvf + a0 + e, dv_acc[ch][e]);
-            atomicAdd(dt2 + (int64_t)b * A + a0 + e, dt2_acc[ch][e]);
+            sdv[wid][a0 + e] = dv_acc[ch][e];
+            sd2[wid][a0 + e] = dt2_acc[ch][e];
         }
+    }
+    __syncthreads();
+    for (int a = threadIdx.x; a < A; a += (int)blockDim.x) {
+        float dv_s = sdv[0][a] + sdv[1][a] + sdv[2][a] + sdv[3][a];
+        float d2_s = sd2[0][a] + sd2[1][a] + sd2[2][a] + sd2[3][a];
+        atomicAdd(dvf + a, dv_s);
+        atomicAdd(dt2 + (int64_t)b * A + a, d2_s);
     }
 }
 
@@ -510,7 +520,7 @@ std::vector<at::Tensor> attn_scores_bwd_acc(at::Tensor tdrop, at::Tensor v,
         dvf = dv_acc;   // caller-owned accumulator (atomicAdd accumulates)
     else
         dvf = at::zeros({A}, tdrop.options().dtype(at::kFloat));
-    int lchunk = ((int)L + 15) / 16;   // ~16 chunks/image -> 512 blocks
+    int lchunk = ((int)L + 3) / 4;     // 4 chunks/image
     int nchunk = ((int)L + lchunk - 1) / lchunk;
     hipStream_t s = at::cuda::getCurrentCUDAStream();
 #define LAUNCH_SB(NCH) \
