@@ -141,16 +141,32 @@ class Conv2d(tnn.Module):
                     else torch.empty(0, dtype=x.dtype, device=x.device),
                     self.activation == 'relu')
                 return y
+        # frozen GPU path: fused NHWC bias+ReLU kernel after the MIOpen
+        # conv instead of two separate eager elementwise passes
+        fuse_epi = (x.is_cuda and x.dtype == torch.bfloat16
+                    and not torch.is_grad_enabled() and b is not None
+                    and self.weight.shape[0] % 8 == 0)
+        conv_bias = None if fuse_epi else b
         ph = _same_pad(x.shape[2], k, st)
         pw = _same_pad(x.shape[3], k, st)
         if ph % 2 == 0 and pw % 2 == 0:
             # symmetric SAME: no pad-copy kernel, pad inside the conv
-            y = torch.nn.functional.conv2d(x, w, b, stride=st,
+            y = torch.nn.functional.conv2d(x, w, conv_bias, stride=st,
                                            padding=(ph // 2, pw // 2))
         else:
             x = torch.nn.functional.pad(
                 x, (pw // 2, pw - pw // 2, ph // 2, ph - ph // 2))
-            y = torch.nn.functional.conv2d(x, w, b, stride=st)
+            y = torch.nn.functional.conv2d(x, w, conv_bias, stride=st)
+        if fuse_epi and y.is_contiguous(
+                memory_format=torch.channels_last):
+            from ..ops import hip
+            if hip.available():
+                from sat_amd import _C
+                _C.bias_act_nhwc(y, b, self.activation == 'relu')
+                return y
+            y = y + b.reshape(1, -1, 1, 1)
+        elif fuse_epi:
+            y = y + b.reshape(1, -1, 1, 1)
         if self.activation == 'relu':
             y = torch.relu(y)
         return y
@@ -180,6 +196,15 @@ class MaxPool2d(tnn.Module):
         self.stride = stride
 
     def forward(self, x):
+        if (x.is_cuda and x.dtype == torch.bfloat16
+                and not torch.is_grad_enabled()
+                and self.kernel_size == 2 and self.stride == 2
+                and x.shape[1] % 8 == 0
+                and x.is_contiguous(memory_format=torch.channels_last)):
+            from ..ops import hip
+            if hip.available():
+                from sat_amd import _C
+                return _C.maxpool2x2_nhwc(x)
         x = _pad_same(x, self.kernel_size, self.stride)
         return torch.nn.functional.max_pool2d(
             x, self.kernel_size, self.stride)

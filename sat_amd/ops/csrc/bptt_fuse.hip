@@ -275,16 +275,22 @@ __global__ void hash_dropout_steps_kernel(const bf16* __restrict__ x,
                                           bf16* __restrict__ y,
                                           int64_t n, int T, float p,
                                           int salt_base, int salt_stride) {
+    // n % 8 == 0 (checked by the wrapper): bf16x8 vector traffic
     const uint32_t seed = (uint32_t)(*seed_p);
-    int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t i8 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
     int64_t total = n * T;
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t i = idx; i < total; i += stride) {
-        int t = (int)(i / n);
-        int64_t j = i % n;
-        y[i] = f2bf(bf2f(x[j]) *
-                    dscale(seed, salt_base + t * salt_stride,
-                           (uint32_t)j, p));
+    int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
+    for (; i8 < total; i8 += stride) {
+        int t = (int)(i8 / n);
+        int64_t j = i8 % n;
+        int salt = salt_base + t * salt_stride;
+        bf16x8 v = *(const bf16x8*)(x + j);
+        bf16x8 o;
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+            o[e] = f2bf(bf2f(v[e]) * dscale(seed, salt,
+                                            (uint32_t)(j + e), p));
+        *(bf16x8*)(y + i8) = o;
     }
 }
 
@@ -292,9 +298,10 @@ at::Tensor hash_dropout_steps(at::Tensor x, at::Tensor seed, double p,
                               int64_t salt_base, int64_t salt_stride,
                               int64_t T) {
     int64_t n = x.numel();
+    TORCH_CHECK(n % 8 == 0, "hash_dropout_steps needs numel % 8 == 0");
     auto y = at::empty({T, x.size(0), x.size(1)}, x.options());
     hipStream_t s = at::cuda::getCurrentCUDAStream();
-    int blocks = (int)std::min<int64_t>(cdiv(n * T, 256), 16384);
+    int blocks = (int)std::min<int64_t>(cdiv(n * T, 256 * 8), 16384);
     hipLaunchKernelGGL(hash_dropout_steps_kernel, dim3(blocks), dim3(256),
                        0, s,
                        (const bf16*)x.data_ptr(),

@@ -102,3 +102,100 @@ at::Tensor conv3_fwd(at::Tensor input, at::Tensor weight, at::Tensor bias,
     HIP_OK(hipGetLastError());
     return out;
 }
+
+// ---- fused NHWC bias + ReLU (replaces torch's separate bias-add and
+// clamp kernels after each MIOpen conv: 2x ~23us -> 1x ~12us per layer) ----
+
+__global__ void bias_act_nhwc_kernel(bf16* __restrict__ y,
+                                     const bf16* __restrict__ bias,
+                                     int64_t n, int C, int relu) {
+    int64_t i8 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+    if (i8 >= n) return;
+    bf16x8 v = *(const bf16x8*)(y + i8);
+    int c0 = (int)(i8 % C);
+    const bf16x8 bv = *(const bf16x8*)(bias + c0);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+        float t = bf2f(v[e]) + bf2f(bv[e]);
+        if (relu) t = fmaxf(t, 0.f);
+        v[e] = f2bf(t);
+    }
+    *(bf16x8*)(y + i8) = v;
+}
+
+void bias_act_nhwc(at::Tensor y, at::Tensor bias, bool relu) {
+    // y: channels_last [B,C,H,W]; per-pixel channel runs are contiguous
+    CHECK_GPU(y); CHECK_BF16(y);
+    int C = y.size(1);
+    TORCH_CHECK(C % 8 == 0, "C must be a multiple of 8");
+    TORCH_CHECK(y.is_contiguous(at::MemoryFormat::ChannelsLast));
+    int64_t n = y.numel();
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(bias_act_nhwc_kernel, dim3(cdiv(n / 8, 256)),
+                       dim3(256), 0, s,
+                       (bf16*)y.data_ptr(),
+                       (const bf16*)bias.contiguous().data_ptr(),
+                       n, C, relu ? 1 : 0);
+    HIP_OK(hipGetLastError());
+}
+
+// ---- direct NHWC 2x2/s2 max pool (torch's nhwc maxpool measured 74us on
+// [32,64,112,112]; traffic floor is ~8us) ----
+
+__global__ void maxpool2x2_nhwc_kernel(const bf16* __restrict__ in,
+                                       bf16* __restrict__ out,
+                                       int B, int H, int W, int C,
+                                       int Ho, int Wo) {
+    int64_t i8 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+    int64_t n = (int64_t)B * Ho * Wo * C;
+    if (i8 >= n) return;
+    int c = (int)(i8 % C);
+    int64_t pix = i8 / C;
+    int xo = (int)(pix % Wo);
+    int64_t t = pix / Wo;
+    int yo = (int)(t % Ho);
+    int b = (int)(t / Ho);
+    int y0 = yo * 2, x0 = xo * 2;
+    const bf16* base = in + (((int64_t)b * H + y0) * W + x0) * C + c;
+    bf16x8 m = *(const bf16x8*)base;
+    bf16x8 v;
+    if (x0 + 1 < W) {
+        v = *(const bf16x8*)(base + C);
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+            if (bf2f(v[e]) > bf2f(m[e])) m[e] = v[e];
+    }
+    if (y0 + 1 < H) {
+        v = *(const bf16x8*)(base + (int64_t)W * C);
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+            if (bf2f(v[e]) > bf2f(m[e])) m[e] = v[e];
+        if (x0 + 1 < W) {
+            v = *(const bf16x8*)(base + (int64_t)W * C + C);
+#pragma unroll
+            for (int e = 0; e < 8; ++e)
+                if (bf2f(v[e]) > bf2f(m[e])) m[e] = v[e];
+        }
+    }
+    *(bf16x8*)(out + i8) = m;
+}
+
+at::Tensor maxpool2x2_nhwc(at::Tensor input) {
+    CHECK_GPU(input); CHECK_BF16(input);
+    TORCH_CHECK(input.is_contiguous(at::MemoryFormat::ChannelsLast));
+    int B = input.size(0), C = input.size(1), H = input.size(2),
+        W = input.size(3);
+    TORCH_CHECK(C % 8 == 0);
+    int Ho = (H + 1) / 2, Wo = (W + 1) / 2;
+    auto out = at::empty({B, C, Ho, Wo},
+                         input.options()
+                             .memory_format(at::MemoryFormat::ChannelsLast));
+    int64_t n = (int64_t)B * Ho * Wo * C;
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(maxpool2x2_nhwc_kernel, dim3(cdiv(n / 8, 256)),
+                       dim3(256), 0, s,
+                       (const bf16*)input.data_ptr(),
+                       (bf16*)out.data_ptr(), B, H, W, C, Ho, Wo);
+    HIP_OK(hipGetLastError());
+    return out;
+}

@@ -297,3 +297,26 @@ def test_conv3_fwd_no_bias_no_relu():
                                        device=DEV), False)
     ref = torch.nn.functional.conv2d(x.float(), w.float(), padding=1)
     assert _rel_err(y, ref) < 2e-2
+
+
+def test_bias_act_nhwc():
+    from sat_amd import _C
+    torch.manual_seed(8)
+    y = torch.randn(2, 16, 7, 9).to(DEV, torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    b = torch.randn(16).to(DEV, torch.bfloat16)
+    ref = torch.relu(y.float() + b.float().reshape(1, -1, 1, 1))
+    _C.bias_act_nhwc(y, b, True)
+    assert _rel_err(y, ref) < 1e-2
+
+
+def test_maxpool2x2_nhwc():
+    from sat_amd import _C
+    for H, W in [(8, 8), (7, 9)]:
+        x = torch.randn(2, 16, H, W).to(DEV, torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+        y = _C.maxpool2x2_nhwc(x)
+        ref = torch.nn.functional.max_pool2d(
+            x.float(), 2, 2, ceil_mode=True)
+        assert y.shape == ref.shape
+        assert _rel_err(y, ref) < 1e-3
