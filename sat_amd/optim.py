@@ -95,8 +95,14 @@ class Optimizer(object):
                 self._mt_cum = torch.tensor(
                     cum, dtype=torch.int64).to(dev)
                 self._mt_total = cum[-1]
-                for g in grads:
-                    assert g.is_contiguous() and g.dtype == torch.float32
+                for g, p0 in zip(grads, self.params):
+                    # dense storage in any layout (channels_last conv
+                    # grads included) is fine: p/g/m/v share the layout
+                    # and the kernel iterates flat storage order
+                    assert g.dtype == torch.float32
+                    assert (g.is_contiguous()
+                            or g.is_contiguous(
+                                memory_format=torch.channels_last))
             gsq = _C.sq_norm_mt(self._mt_desc, self._mt_cum,
                                 len(self.params), self._mt_total)
             _C.adam_step_mt(
