@@ -84,8 +84,8 @@ class _Bottleneck(tnn.Module):
             sc = self.shortcut_bn(self.shortcut(x))
         else:
             sc = x
-        y = torch.relu(self.bn_a(self.conv_a(x)))
-        y = torch.relu(self.bn_b(self.conv_b(y)))
+        y = self.bn_a(self.conv_a(x), relu=True)
+        y = self.bn_b(self.conv_b(y), relu=True)
         y = self.bn_c(self.conv_c(y))
         return torch.relu(y + sc)
 
@@ -114,7 +114,7 @@ class ResNet50(tnn.Module):
 
     def forward(self, images):
         """images: [B,3,224,224] -> contexts [B,49,2048]."""
-        x = torch.relu(self.bn1(self.conv1(images)))
+        x = self.bn1(self.conv1(images), relu=True)
         x = self.pool1(x)
         for blk in self._blocks:
             x = blk(x)

@@ -6,6 +6,8 @@ at::Tensor dense_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, int64_t act);
 at::Tensor conv3_fwd(at::Tensor input, at::Tensor weight, at::Tensor bias,
                      bool relu);
 void bias_act_nhwc(at::Tensor y, at::Tensor bias, bool relu);
+void scale_bias_act_nhwc(at::Tensor y, at::Tensor scale, at::Tensor shift,
+                         bool relu);
 at::Tensor maxpool2x2_nhwc(at::Tensor input);
 at::Tensor dense_fwd_out(at::Tensor x, at::Tensor w, at::Tensor bias,
                          int64_t act, at::Tensor out);
@@ -79,6 +81,7 @@ void adam_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("dense_fwd", &dense_fwd, "MFMA GEMM + bias/act (bf16)");
     m.def("bias_act_nhwc", &bias_act_nhwc);
+    m.def("scale_bias_act_nhwc", &scale_bias_act_nhwc);
     m.def("maxpool2x2_nhwc", &maxpool2x2_nhwc);
     m.def("conv3_fwd", &conv3_fwd,
           "direct NHWC conv for 3-channel 3x3/s1 (VGG conv1_1)");

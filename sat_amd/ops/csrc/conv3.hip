@@ -199,3 +199,43 @@ at::Tensor maxpool2x2_nhwc(at::Tensor input) {
     HIP_OK(hipGetLastError());
     return out;
 }
+
+// ---- frozen BatchNorm as one NHWC scale/shift (+optional ReLU) kernel:
+// y = x*scale + shift with scale = gamma/sqrt(var+eps), shift = beta -
+// mean*scale, both precomputed once on the host side ----
+
+__global__ void scale_bias_act_nhwc_kernel(bf16* __restrict__ y,
+                                           const bf16* __restrict__ scale,
+                                           const bf16* __restrict__ shift,
+                                           int64_t n, int C, int relu) {
+    int64_t i8 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+    if (i8 >= n) return;
+    bf16x8 v = *(const bf16x8*)(y + i8);
+    int c0 = (int)(i8 % C);
+    const bf16x8 sv = *(const bf16x8*)(scale + c0);
+    const bf16x8 bv = *(const bf16x8*)(shift + c0);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+        float t = bf2f(v[e]) * bf2f(sv[e]) + bf2f(bv[e]);
+        if (relu) t = fmaxf(t, 0.f);
+        v[e] = f2bf(t);
+    }
+    *(bf16x8*)(y + i8) = v;
+}
+
+void scale_bias_act_nhwc(at::Tensor y, at::Tensor scale, at::Tensor shift,
+                         bool relu) {
+    CHECK_GPU(y); CHECK_BF16(y);
+    int C = y.size(1);
+    TORCH_CHECK(C % 8 == 0, "C must be a multiple of 8");
+    TORCH_CHECK(y.is_contiguous(at::MemoryFormat::ChannelsLast));
+    int64_t n = y.numel();
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(scale_bias_act_nhwc_kernel, dim3(cdiv(n / 8, 256)),
+                       dim3(256), 0, s,
+                       (bf16*)y.data_ptr(),
+                       (const bf16*)scale.data_ptr(),
+                       (const bf16*)shift.data_ptr(),
+                       n, C, relu ? 1 : 0);
+    HIP_OK(hipGetLastError());
+}
