@@ -44,6 +44,10 @@ class GraphedTrainStep(object):
         self.optimizer.step()
         return out
 
+    def _rng_tensor(self):
+        dec = getattr(self.model, 'decoder', None)
+        return getattr(dec, '_rng', None) if dec is not None else None
+
     def _snapshot(self):
         opt = self.optimizer
         snap = {
@@ -52,6 +56,9 @@ class GraphedTrainStep(object):
                       for i, p in enumerate(opt.params)},
             'step_count': opt.step_count,
         }
+        rng = self._rng_tensor()
+        if rng is not None:
+            snap['rng'] = rng.clone()
         return snap
 
     def _restore(self, snap):
@@ -65,6 +72,9 @@ class GraphedTrainStep(object):
         opt.step_count = snap['step_count']
         if getattr(opt, 'step_dev', None) is not None:
             opt.step_dev.fill_(float(opt.step_count))
+        rng = self._rng_tensor()
+        if rng is not None and 'rng' in snap:
+            rng.copy_(snap['rng'])
 
     def _capture(self, images, sentences, masks):
         self.static_in = (images.clone(), sentences.clone(), masks.clone())
