@@ -38,7 +38,47 @@ def _make_caption(rng):
     return ' '.join(words[:-1]) + '.'
 
 
-def make_coco_dict(num_images, caps_per_image, seed, prefix=0):
+# ---- learnable mode: the caption is a deterministic function of the ----
+# image.  Slot choices drawn from RandomState(image_id) pick both the
+# caption words and the image's horizontal color bands, so a model that
+# reads the bands through the (random, frozen) CNN features can genuinely
+# generalize to held-out image ids — the honest synthetic analog of the
+# reference's COCO BLEU protocol.
+
+_SLOTS = [_ADJS, _NOUNS, _VERBS, _PREPS, _ADJS, _NOUNS]
+
+
+def _learnable_choices(image_id):
+    rng = np.random.RandomState(image_id % (2 ** 31))
+    return [int(rng.randint(len(pool))) for pool in _SLOTS]
+
+
+def learnable_caption(image_id):
+    c = _learnable_choices(image_id)
+    return 'a %s %s %s %s a %s %s.' % (
+        _SLOTS[0][c[0]], _SLOTS[1][c[1]], _SLOTS[2][c[2]],
+        _SLOTS[3][c[3]], _SLOTS[4][c[4]], _SLOTS[5][c[5]])
+
+
+def learnable_image(image_id, shape=(224, 224, 3)):
+    """Horizontal bands encode the slot choices (values sized like
+    mean-subtracted pixels)."""
+    h, w, _ = shape
+    img = np.zeros(shape, dtype=np.float32)
+    band = h // len(_SLOTS)
+    c = _learnable_choices(image_id)
+    for i, (choice, pool) in enumerate(zip(c, _SLOTS)):
+        frac = (choice + 1) / (len(pool) + 1)
+        y0, y1 = i * band, (i + 1) * band if i + 1 < len(_SLOTS) else h
+        img[y0:y1, :, 0] = 200.0 * frac - 100.0
+        img[y0:y1, :, 1] = 200.0 * ((choice * 7) % len(pool)) \
+            / len(pool) - 100.0
+        img[y0:y1, :, 2] = 100.0 * np.sin(choice + i)
+    return img
+
+
+def make_coco_dict(num_images, caps_per_image, seed, prefix=0,
+                   mode='noise'):
     rng = np.random.RandomState(seed)
 
     class _R:  # adapter: RandomState with python-like choice/random
@@ -56,29 +96,38 @@ def make_coco_dict(num_images, caps_per_image, seed, prefix=0):
         images.append({'id': iid,
                        'file_name': 'synthetic://%d' % iid})
         for _ in range(caps_per_image):
+            cap = (learnable_caption(iid) if mode == 'learnable'
+                   else _make_caption(r))
             annotations.append({'id': ann_id, 'image_id': iid,
-                                'caption': _make_caption(r)})
+                                'caption': cap})
             ann_id += 1
     return {'images': images, 'annotations': annotations}
 
 
-def make_coco(num_images, caps_per_image, seed, prefix=0):
+def make_coco(num_images, caps_per_image, seed, prefix=0, mode='noise'):
     coco = COCO()
-    coco.dataset = make_coco_dict(num_images, caps_per_image, seed, prefix)
+    coco.dataset = make_coco_dict(num_images, caps_per_image, seed, prefix,
+                                  mode)
     coco.process_dataset()
     coco.createIndex()
     return coco
 
 
 class SyntheticImageLoader(object):
-    """Deterministic random 'images' for synthetic:// files."""
+    """Deterministic 'images' for synthetic:// files.
 
-    def __init__(self, image_shape=(224, 224, 3), seed=0):
+    mode 'noise': id-seeded gaussian noise (pipeline tests).
+    mode 'learnable': band-coded images matching learnable_caption(id)."""
+
+    def __init__(self, image_shape=(224, 224, 3), seed=0, mode='noise'):
         self.image_shape = tuple(image_shape)
         self.seed = seed
+        self.mode = mode
 
     def load_image(self, image_file):
         iid = int(str(image_file).split('://')[-1])
+        if self.mode == 'learnable':
+            return learnable_image(iid, self.image_shape)
         rng = np.random.RandomState((self.seed * 1000003 + iid) % (2 ** 31))
         return rng.randn(*self.image_shape).astype(np.float32) * 50.0
 
@@ -95,7 +144,8 @@ def _vocab_from_coco(config, coco):
 
 def prepare_train_data(config):
     n = getattr(config, 'synthetic_num_images', 640)
-    coco = make_coco(n, 1, config.seed)
+    mode = getattr(config, 'synthetic_mode', 'noise')
+    coco = make_coco(n, 1, config.seed, mode=mode)
     coco.filter_by_cap_len(config.max_caption_length)
     vocabulary = _vocab_from_coco(config, coco)
     coco.filter_by_words(set(vocabulary.words))
@@ -114,7 +164,8 @@ def prepare_train_data(config):
 
 def prepare_eval_data(config):
     n = min(getattr(config, 'synthetic_num_images', 640), 64)
-    coco = make_coco(n, 5, config.seed + 1, prefix=1)
+    mode = getattr(config, 'synthetic_mode', 'noise')
+    coco = make_coco(n, 5, config.seed + 1, prefix=1, mode=mode)
     if config.max_eval_ann_num:
         ann_ids = list(coco.anns.keys())[:config.max_eval_ann_num]
         image_ids = list(dict.fromkeys(
@@ -124,7 +175,7 @@ def prepare_eval_data(config):
     image_files = [coco.imgs[i]['file_name'] for i in image_ids]
     # build vocabulary from a train-shaped corpus so idx<->word matches train
     train_coco = make_coco(getattr(config, 'synthetic_num_images', 640), 1,
-                           config.seed)
+                           config.seed, mode=mode)
     vocabulary = _vocab_from_coco(config, train_coco)
     dataset = DataSet(image_ids, image_files, config.batch_size)
     return coco, dataset, vocabulary

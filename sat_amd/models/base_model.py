@@ -54,7 +54,8 @@ class BaseModel(object):
         self.image_shape = [224, 224, 3]
         if getattr(config, 'synthetic_data', False):
             self.image_loader = SyntheticImageLoader(
-                self.image_shape, getattr(config, 'seed', 0))
+                self.image_shape, getattr(config, 'seed', 0),
+                getattr(config, 'synthetic_mode', 'noise'))
         else:
             self.image_loader = ImageLoader(None, self.image_shape)
 
