@@ -119,17 +119,28 @@ class SyntheticImageLoader(object):
     mode 'noise': id-seeded gaussian noise (pipeline tests).
     mode 'learnable': band-coded images matching learnable_caption(id)."""
 
-    def __init__(self, image_shape=(224, 224, 3), seed=0, mode='noise'):
+    def __init__(self, image_shape=(224, 224, 3), seed=0, mode='noise',
+                 cache_limit=2048):
         self.image_shape = tuple(image_shape)
         self.seed = seed
         self.mode = mode
+        self._cache = {}
+        self._cache_limit = cache_limit
 
     def load_image(self, image_file):
         iid = int(str(image_file).split('://')[-1])
+        img = self._cache.get(iid)
+        if img is not None:
+            return img
         if self.mode == 'learnable':
-            return learnable_image(iid, self.image_shape)
-        rng = np.random.RandomState((self.seed * 1000003 + iid) % (2 ** 31))
-        return rng.randn(*self.image_shape).astype(np.float32) * 50.0
+            img = learnable_image(iid, self.image_shape)
+        else:
+            rng = np.random.RandomState(
+                (self.seed * 1000003 + iid) % (2 ** 31))
+            img = rng.randn(*self.image_shape).astype(np.float32) * 50.0
+        if len(self._cache) < self._cache_limit:
+            self._cache[iid] = img
+        return img
 
     def load_images(self, image_files):
         return np.stack([self.load_image(f) for f in image_files], axis=0)
