@@ -66,7 +66,16 @@ class CaptionGenerator(tnn.Module):
                 and getattr(cfg, 'use_hip_kernels', True)
                 and getattr(cfg, 'use_bptt', True)
                 and cfg.num_attend_layers == 2
-                and cfg.num_decode_layers == 2)
+                and cfg.num_decode_layers == 2
+                # kernel shape contracts (fall back to the per-op loop
+                # for exotic dims): bf16x8 K-tiles + scores-bwd A chunks
+                and cfg.dim_attend_layer % 512 == 0
+                and cfg.dim_attend_layer <= 2048
+                and self.dim_ctx % 8 == 0
+                and cfg.dim_embedding % 8 == 0
+                and cfg.num_lstm_units % 8 == 0
+                and cfg.dim_decode_layer % 8 == 0
+                and self.num_ctx <= 1024)
 
     def compute_contexts(self, images):
         """images: [B,3,224,224] float -> contexts [B,L,D].

@@ -35,7 +35,7 @@ def _use_hip(*tensors):
 
 def dense(x, weight, bias=None, activation=None):
     """act(x @ weight.T + bias). x: [M,K], weight: [N,K] (torch layout)."""
-    if _use_hip(x, weight):
+    if _use_hip(x, weight) and x.shape[-1] % 8 == 0:
         from . import hip
         return hip.dense(x, weight, bias, activation)
     return F.dense(x, weight, bias, activation)
@@ -43,7 +43,7 @@ def dense(x, weight, bias=None, activation=None):
 
 def lstm_cell(x, h, c, weight, bias, forget_bias=1.0):
     """One TF-semantics LSTMCell step; returns (h', c')."""
-    if _use_hip(x, weight):
+    if _use_hip(x, weight) and (x.shape[1] + h.shape[1]) % 8 == 0:
         from . import hip
         return hip.lstm_cell(x, h, c, weight, bias, forget_bias)
     return F.lstm_cell(x, h, c, weight, bias, forget_bias)
@@ -51,7 +51,8 @@ def lstm_cell(x, h, c, weight, bias, forget_bias=1.0):
 
 def attention_pool(contexts, logits):
     """softmax over L + weighted context sum; returns (alpha, context)."""
-    if _use_hip(contexts, logits):
+    if _use_hip(contexts) and contexts.shape[1] <= 1024 \
+            and contexts.shape[2] % 8 == 0:
         from . import hip
         return hip.attention_pool(contexts, logits)
     return F.attention_pool(contexts, logits)
@@ -63,7 +64,9 @@ def attention_tail(t1, t2, v, contexts, p, training, seed_dev, salt):
     N=1-GEMM/softmax/weighted-sum chain (model.py:425-435, :263-264) with
     two fused kernels on GPU; counter-based dropout (seed_dev, salt) keeps
     it hipGraph-safe."""
-    if _use_hip(contexts, t1):
+    if (_use_hip(contexts, t1) and t1.shape[1] % 512 == 0
+            and t1.shape[1] <= 2048 and contexts.shape[1] <= 1024
+            and contexts.shape[2] % 8 == 0):
         from . import hip
         return hip.attention_tail(t1, t2, v, contexts,
                                   p if training else 0.0, seed_dev, salt)
