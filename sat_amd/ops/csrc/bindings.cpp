@@ -9,6 +9,8 @@ void bias_act_nhwc(at::Tensor y, at::Tensor bias, bool relu);
 void scale_bias_act_nhwc(at::Tensor y, at::Tensor scale, at::Tensor shift,
                          bool relu);
 at::Tensor maxpool2x2_nhwc(at::Tensor input);
+at::Tensor conv_igemm_fwd(at::Tensor input, at::Tensor w_ohwi,
+                          at::Tensor bias, bool relu);
 at::Tensor dense_fwd_out(at::Tensor x, at::Tensor w, at::Tensor bias,
                          int64_t act, at::Tensor out);
 std::vector<at::Tensor> lstm_pointwise_bwd_out(at::Tensor gates,
@@ -83,6 +85,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("bias_act_nhwc", &bias_act_nhwc);
     m.def("scale_bias_act_nhwc", &scale_bias_act_nhwc);
     m.def("maxpool2x2_nhwc", &maxpool2x2_nhwc);
+    m.def("conv_igemm_fwd", &conv_igemm_fwd);
     m.def("conv3_fwd", &conv3_fwd,
           "direct NHWC conv for 3-channel 3x3/s1 (VGG conv1_1)");
     m.def("dense_fwd_out", &dense_fwd_out);

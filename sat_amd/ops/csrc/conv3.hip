@@ -14,6 +14,9 @@
 #include "common.h"
 
 #define MAX_COUT 128
+#ifndef LDS_STRIDE
+#define LDS_STRIDE 72
+#endif
 
 __global__ __launch_bounds__(256)
 void conv3_fwd_kernel(const bf16* __restrict__ in,   // [B,H,W,3]
@@ -238,4 +241,165 @@ void scale_bias_act_nhwc(at::Tensor y, at::Tensor scale, at::Tensor shift,
                        (const bf16*)shift.data_ptr(),
                        n, C, relu ? 1 : 0);
     HIP_OK(hipGetLastError());
+}
+
+// ---- implicit-GEMM NHWC 3x3/s1/pad1 convolution (Cin % 64 == 0) ----
+//
+// The conv as a GEMM: M = B*H*W output pixels, N = Cout, K = 9*Cin with
+// k ordered (dy, dx, ci), ci fastest.  Weights are repacked once to OHWI
+// ([Cout][3][3][Cin]), which IS the [N,K] row-major B operand.  The A
+// operand is the input gathered on the fly: each k-tile fixes (dy,dx), so
+// every row's address is its cached pixel base plus ONE scalar delta;
+// only the edge-validity mask differs per row.  Structure is the tiled
+// MFMA GEMM (128x64 tile, 4 waves, LDS +8 padding); bias+ReLU fused in
+// the epilogue.
+
+__global__ __launch_bounds__(256)
+void conv_igemm_kernel(const bf16* __restrict__ in,    // [B,H,W,Cin]
+                       const bf16* __restrict__ w,     // [Cout,9*Cin] OHWI
+                       const bf16* __restrict__ bias,  // [Cout] or null
+                       bf16* __restrict__ out,         // [B,H,W,Cout]
+                       int M, int Hh, int Ww, int Cin, int Cout,
+                       int relu) {
+    constexpr int BM = 128, BN = 64, BKc = 64;
+    __shared__ bf16 As[BM * LDS_STRIDE];
+    __shared__ bf16 Bs[BN * LDS_STRIDE];
+    __shared__ int rbase[BM];
+    __shared__ short rys[BM], rxs[BM];
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;           // 4 waves, 32 rows each
+    const int bm = blockIdx.y * BM;
+    const int bn = blockIdx.x * BN;
+
+    // per-row pixel coordinates + base offsets (once per block)
+    for (int r = tid; r < BM; r += blockDim.x) {
+        int m = bm + r;
+        if (m < M) {
+            int b = m / (Hh * Ww);
+            int yx = m % (Hh * Ww);
+            int y = yx / Ww, x = yx % Ww;
+            rbase[r] = ((b * Hh + y) * Ww + x) * Cin;
+            rys[r] = (short)y;
+            rxs[r] = (short)x;
+        } else {
+            rbase[r] = 0;
+            rys[r] = -1;     // never valid
+            rxs[r] = -1;
+        }
+    }
+    __syncthreads();
+
+    floatx4 acc[2][4];
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+            acc[i][j] = floatx4{0.f, 0.f, 0.f, 0.f};
+
+    const int lrow = lane & 15;
+    const int kgrp = lane >> 4;
+    const int KT = 9 * (Cin / BKc);
+
+    for (int kt = 0; kt < KT; ++kt) {
+        const int dxy = kt / (Cin / BKc);
+        const int ci0 = (kt % (Cin / BKc)) * BKc;
+        const int dy = dxy / 3 - 1, dx = dxy % 3 - 1;
+        const int delta = (dy * Ww + dx) * Cin + ci0;
+
+        // stage A: 128 rows x 64 ci = 4 bf16x8 chunks per thread
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            int q = tid + 256 * i;
+            int row = q >> 3;
+            int c8 = (q & 7) * 8;
+            int yy = rys[row] + dy, xx = rxs[row] + dx;
+            bf16x8 v = {};
+            if (yy >= 0 && yy < Hh && xx >= 0 && xx < Ww)
+                v = *(const bf16x8*)(in + rbase[row] + delta + c8);
+            *(bf16x8*)(As + row * LDS_STRIDE + c8) = v;
+        }
+        // stage B: 64 rows (Cout) x 64 k = 2 chunks per thread
+#pragma unroll
+        for (int i = 0; i < 2; ++i) {
+            int q = tid + 256 * i;
+            int row = q >> 3;
+            int c8 = (q & 7) * 8;
+            bf16x8 v = {};
+            int gb = bn + row;
+            if (gb < Cout)
+                v = *(const bf16x8*)(
+                    w + (int64_t)gb * 9 * Cin + dxy * Cin + ci0 + c8);
+            *(bf16x8*)(Bs + row * LDS_STRIDE + c8) = v;
+        }
+        __syncthreads();
+
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk) {
+            bf16x8 a_frag[2], b_frag[4];
+            const int kof = kk * 32 + kgrp * 8;
+#pragma unroll
+            for (int mi = 0; mi < 2; ++mi)
+                a_frag[mi] = *(const bf16x8*)(
+                    As + (wave * 32 + mi * 16 + lrow) * LDS_STRIDE + kof);
+#pragma unroll
+            for (int ni = 0; ni < 4; ++ni)
+                b_frag[ni] = *(const bf16x8*)(
+                    Bs + (ni * 16 + lrow) * LDS_STRIDE + kof);
+#pragma unroll
+            for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+                for (int ni = 0; ni < 4; ++ni)
+                    acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a_frag[mi], b_frag[ni], acc[mi][ni], 0, 0, 0);
+        }
+        __syncthreads();
+    }
+
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+        int col = bn + ni * 16 + (lane & 15);
+        float bv = (bias != nullptr && col < Cout) ? bf2f(bias[col]) : 0.f;
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = bm + wave * 32 + mi * 16 + (lane >> 4) * 4 + r;
+                if (row < M && col < Cout) {
+                    float v = acc[mi][ni][r] + bv;
+                    if (relu) v = fmaxf(v, 0.f);
+                    out[(int64_t)row * Cout + col] = f2bf(v);
+                }
+            }
+        }
+    }
+}
+
+at::Tensor conv_igemm_fwd(at::Tensor input, at::Tensor w_ohwi,
+                          at::Tensor bias, bool relu) {
+    CHECK_GPU(input); CHECK_BF16(input);
+    CHECK_GPU(w_ohwi); CHECK_CONTIG(w_ohwi); CHECK_BF16(w_ohwi);
+    TORCH_CHECK(input.is_contiguous(at::MemoryFormat::ChannelsLast));
+    int B = input.size(0), Cin = input.size(1), Hh = input.size(2),
+        Ww = input.size(3);
+    int Cout = w_ohwi.size(0);
+    TORCH_CHECK(Cin % 64 == 0, "conv_igemm: Cin % 64 == 0");
+    TORCH_CHECK(w_ohwi.numel() == (int64_t)Cout * 9 * Cin);
+    auto out = at::empty({B, Cout, Hh, Ww},
+                         input.options()
+                             .memory_format(at::MemoryFormat::ChannelsLast));
+    const bf16* bias_ptr = nullptr;
+    if (bias.defined() && bias.numel() > 0)
+        bias_ptr = (const bf16*)bias.contiguous().data_ptr();
+    int M = B * Hh * Ww;
+    dim3 grid(cdiv(Cout, 64), cdiv(M, 128));
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(conv_igemm_kernel, grid, dim3(256), 0, s,
+                       (const bf16*)input.data_ptr(),
+                       (const bf16*)w_ohwi.data_ptr(), bias_ptr,
+                       (bf16*)out.data_ptr(), M, Hh, Ww, Cin, Cout,
+                       relu ? 1 : 0);
+    HIP_OK(hipGetLastError());
+    return out;
 }
