@@ -254,6 +254,7 @@ void scale_bias_act_nhwc(at::Tensor y, at::Tensor scale, at::Tensor shift,
 // MFMA GEMM (128x64 tile, 4 waves, LDS +8 padding); bias+ReLU fused in
 // the epilogue.
 
+template <int WR, int WC, int FM, int FN>
 __global__ __launch_bounds__(256)
 void conv_igemm_kernel(const bf16* __restrict__ in,    // [B,H,W,Cin]
                        const bf16* __restrict__ w,     // [Cout,9*Cin] OHWI
@@ -261,7 +262,9 @@ void conv_igemm_kernel(const bf16* __restrict__ in,    // [B,H,W,Cin]
                        bf16* __restrict__ out,         // [B,H,W,Cout]
                        int M, int Hh, int Ww, int Cin, int Cout,
                        int relu) {
-    constexpr int BM = 128, BN = 64, BKc = 64;
+    constexpr int BM = WR * FM * 16;   // 128
+    constexpr int BN = WC * FN * 16;   // 64 or 128
+    constexpr int BKc = 64;
     __shared__ bf16 As[BM * LDS_STRIDE];
     __shared__ bf16 Bs[BN * LDS_STRIDE];
     __shared__ int rbase[BM];
@@ -269,11 +272,12 @@ void conv_igemm_kernel(const bf16* __restrict__ in,    // [B,H,W,Cin]
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
-    const int wave = tid >> 6;           // 4 waves, 32 rows each
+    const int wave = tid >> 6;
+    const int wr = wave / WC;
+    const int wc = wave % WC;
     const int bm = blockIdx.y * BM;
     const int bn = blockIdx.x * BN;
 
-    // per-row pixel coordinates + base offsets (once per block)
     for (int r = tid; r < BM; r += blockDim.x) {
         int m = bm + r;
         if (m < M) {
@@ -285,22 +289,24 @@ void conv_igemm_kernel(const bf16* __restrict__ in,    // [B,H,W,Cin]
             rxs[r] = (short)x;
         } else {
             rbase[r] = 0;
-            rys[r] = -1;     // never valid
+            rys[r] = -1;
             rxs[r] = -1;
         }
     }
     __syncthreads();
 
-    floatx4 acc[2][4];
+    floatx4 acc[FM][FN];
 #pragma unroll
-    for (int i = 0; i < 2; ++i)
+    for (int i = 0; i < FM; ++i)
 #pragma unroll
-        for (int j = 0; j < 4; ++j)
+        for (int j = 0; j < FN; ++j)
             acc[i][j] = floatx4{0.f, 0.f, 0.f, 0.f};
 
     const int lrow = lane & 15;
     const int kgrp = lane >> 4;
     const int KT = 9 * (Cin / BKc);
+    constexpr int A_CHUNKS = BM * BKc / 8 / 256;
+    constexpr int B_CHUNKS = BN * BKc / 8 / 256;
 
     for (int kt = 0; kt < KT; ++kt) {
         const int dxy = kt / (Cin / BKc);
@@ -308,9 +314,8 @@ void conv_igemm_kernel(const bf16* __restrict__ in,    // [B,H,W,Cin]
         const int dy = dxy / 3 - 1, dx = dxy % 3 - 1;
         const int delta = (dy * Ww + dx) * Cin + ci0;
 
-        // stage A: 128 rows x 64 ci = 4 bf16x8 chunks per thread
 #pragma unroll
-        for (int i = 0; i < 4; ++i) {
+        for (int i = 0; i < A_CHUNKS; ++i) {
             int q = tid + 256 * i;
             int row = q >> 3;
             int c8 = (q & 7) * 8;
@@ -320,9 +325,8 @@ void conv_igemm_kernel(const bf16* __restrict__ in,    // [B,H,W,Cin]
                 v = *(const bf16x8*)(in + rbase[row] + delta + c8);
             *(bf16x8*)(As + row * LDS_STRIDE + c8) = v;
         }
-        // stage B: 64 rows (Cout) x 64 k = 2 chunks per thread
 #pragma unroll
-        for (int i = 0; i < 2; ++i) {
+        for (int i = 0; i < B_CHUNKS; ++i) {
             int q = tid + 256 * i;
             int row = q >> 3;
             int c8 = (q & 7) * 8;
@@ -337,20 +341,22 @@ void conv_igemm_kernel(const bf16* __restrict__ in,    // [B,H,W,Cin]
 
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk) {
-            bf16x8 a_frag[2], b_frag[4];
+            bf16x8 a_frag[FM], b_frag[FN];
             const int kof = kk * 32 + kgrp * 8;
 #pragma unroll
-            for (int mi = 0; mi < 2; ++mi)
+            for (int mi = 0; mi < FM; ++mi)
                 a_frag[mi] = *(const bf16x8*)(
-                    As + (wave * 32 + mi * 16 + lrow) * LDS_STRIDE + kof);
+                    As + (wr * FM * 16 + mi * 16 + lrow) * LDS_STRIDE
+                    + kof);
 #pragma unroll
-            for (int ni = 0; ni < 4; ++ni)
+            for (int ni = 0; ni < FN; ++ni)
                 b_frag[ni] = *(const bf16x8*)(
-                    Bs + (ni * 16 + lrow) * LDS_STRIDE + kof);
+                    Bs + (wc * FN * 16 + ni * 16 + lrow) * LDS_STRIDE
+                    + kof);
 #pragma unroll
-            for (int mi = 0; mi < 2; ++mi)
+            for (int mi = 0; mi < FM; ++mi)
 #pragma unroll
-                for (int ni = 0; ni < 4; ++ni)
+                for (int ni = 0; ni < FN; ++ni)
                     acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         a_frag[mi], b_frag[ni], acc[mi][ni], 0, 0, 0);
         }
@@ -358,14 +364,15 @@ void conv_igemm_kernel(const bf16* __restrict__ in,    // [B,H,W,Cin]
     }
 
 #pragma unroll
-    for (int ni = 0; ni < 4; ++ni) {
-        int col = bn + ni * 16 + (lane & 15);
+    for (int ni = 0; ni < FN; ++ni) {
+        int col = bn + wc * FN * 16 + ni * 16 + (lane & 15);
         float bv = (bias != nullptr && col < Cout) ? bf2f(bias[col]) : 0.f;
 #pragma unroll
-        for (int mi = 0; mi < 2; ++mi) {
+        for (int mi = 0; mi < FM; ++mi) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                int row = bm + wave * 32 + mi * 16 + (lane >> 4) * 4 + r;
+                int row = bm + wr * FM * 16 + mi * 16 + (lane >> 4) * 4
+                    + r;
                 if (row < M && col < Cout) {
                     float v = acc[mi][ni][r] + bv;
                     if (relu) v = fmaxf(v, 0.f);
@@ -393,13 +400,24 @@ at::Tensor conv_igemm_fwd(at::Tensor input, at::Tensor w_ohwi,
     if (bias.defined() && bias.numel() > 0)
         bias_ptr = (const bf16*)bias.contiguous().data_ptr();
     int M = B * Hh * Ww;
-    dim3 grid(cdiv(Cout, 64), cdiv(M, 128));
     hipStream_t s = at::cuda::getCurrentCUDAStream();
-    hipLaunchKernelGGL(conv_igemm_kernel, grid, dim3(256), 0, s,
-                       (const bf16*)input.data_ptr(),
-                       (const bf16*)w_ohwi.data_ptr(), bias_ptr,
-                       (bf16*)out.data_ptr(), M, Hh, Ww, Cin, Cout,
-                       relu ? 1 : 0);
+    if (Cout >= 128) {
+        dim3 grid(cdiv(Cout, 128), cdiv(M, 128));
+        hipLaunchKernelGGL((conv_igemm_kernel<2, 2, 4, 4>), grid,
+                           dim3(256), 0, s,
+                           (const bf16*)input.data_ptr(),
+                           (const bf16*)w_ohwi.data_ptr(), bias_ptr,
+                           (bf16*)out.data_ptr(), M, Hh, Ww, Cin, Cout,
+                           relu ? 1 : 0);
+    } else {
+        dim3 grid(cdiv(Cout, 64), cdiv(M, 128));
+        hipLaunchKernelGGL((conv_igemm_kernel<4, 1, 2, 4>), grid,
+                           dim3(256), 0, s,
+                           (const bf16*)input.data_ptr(),
+                           (const bf16*)w_ohwi.data_ptr(), bias_ptr,
+                           (bf16*)out.data_ptr(), M, Hh, Ww, Cin, Cout,
+                           relu ? 1 : 0);
+    }
     HIP_OK(hipGetLastError());
     return out;
 }

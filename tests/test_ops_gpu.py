@@ -320,3 +320,18 @@ def test_maxpool2x2_nhwc():
             x.float(), 2, 2, ceil_mode=True)
         assert y.shape == ref.shape
         assert _rel_err(y, ref) < 1e-3
+
+
+def test_conv_igemm_matches_torch():
+    from sat_amd import _C
+    torch.manual_seed(9)
+    for Cin, Cout, H in [(64, 64, 30), (64, 128, 17), (128, 64, 9)]:
+        x = torch.randn(2, Cin, H, H).to(DEV, torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+        w = (torch.randn(Cout, Cin, 3, 3) * 0.05).to(DEV, torch.bfloat16)
+        b = torch.randn(Cout).to(DEV, torch.bfloat16)
+        w_ohwi = w.permute(0, 2, 3, 1).contiguous().reshape(Cout, -1)
+        y = _C.conv_igemm_fwd(x, w_ohwi, b, True)
+        ref = torch.relu(torch.nn.functional.conv2d(
+            x.float(), w.float(), b.float(), padding=1))
+        assert _rel_err(y, ref) < 2e-2, (Cin, Cout, H)
