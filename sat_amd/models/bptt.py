@@ -71,6 +71,7 @@ class DecoderBPTT(torch.autograd.Function):
         EXPD = torch.empty(T * B, H + D + E, dtype=torch.bfloat16,
                            device=dev)
         HD = torch.empty(T * B, Dd, dtype=torch.bfloat16, device=dev)
+        HID = torch.empty(T * B, Dd, dtype=torch.bfloat16, device=dev)
         ODROP = torch.empty(T * B, H, dtype=torch.bfloat16, device=dev)
         LOGITS = torch.empty(T * B, V, dtype=torch.bfloat16, device=dev)
 
@@ -111,7 +112,7 @@ class DecoderBPTT(torch.autograd.Function):
             out_t, sth_t = _C.expand_fuse(
                 h_raw, pooled, embt, seed, EXPD[sl], od_next,
                 p_lstm, p_fc, s)
-            hid = _C.dense_fwd(EXPD[sl], wd1, bd1, ACT_TANH)
+            hid = _C.dense_fwd_out(EXPD[sl], wd1, bd1, ACT_TANH, HID[sl])
             _C.hash_dropout_out(hid, seed, p_fc, s + 7, HD[sl])
             _C.dense_fwd_out(HD[sl], wd2, bd2, ACT_NONE, LOGITS[sl])
 
@@ -121,7 +122,6 @@ class DecoderBPTT(torch.autograd.Function):
             alphas.append(alpha)
             gates_l.append(gates)
             cprev_l.append(memory)
-            hid_l.append(hid)
 
             memory = c_new
             state_h = sth_t
@@ -138,11 +138,10 @@ class DecoderBPTT(torch.autograd.Function):
 
         ctx_ag.save_for_backward(
             contexts, emb, w1a, b1a, w1b, b1b, v, wl, bl, wd1, bd1,
-            wd2, bd2, seed, XH, EXPD, HD, ODROP, LOGITS, LSE,
+            wd2, bd2, seed, XH, EXPD, HD, HID, ODROP, LOGITS, LSE,
             labels_cat, masks_cat, masks)
         ctx_ag.cdrop = CDROP
-        ctx_ag.saved_lists = (t1s, t2s, tdrops, alphas, gates_l, cprev_l,
-                              hid_l)
+        ctx_ag.saved_lists = (t1s, t2s, tdrops, alphas, gates_l, cprev_l)
         ctx_ag.dims = (B, L, D, T, A, H, E, V, Dd, I)
         ctx_ag.p_fc = p_fc
         ctx_ag.p_lstm = p_lstm
@@ -153,10 +152,10 @@ class DecoderBPTT(torch.autograd.Function):
     @staticmethod
     def backward(ctx_ag, d_ce, d_attn, _d_pred):
         (contexts, emb, w1a, b1a, w1b, b1b, v, wl, bl, wd1, bd1,
-         wd2, bd2, seed, XH, EXPD, HD, ODROP, LOGITS, LSE,
+         wd2, bd2, seed, XH, EXPD, HD, HID, ODROP, LOGITS, LSE,
          labels_cat, masks_cat, masks) = ctx_ag.saved_tensors
-        (t1s, t2s, tdrops, alphas, gates_l, cprev_l,
-         hid_l) = ctx_ag.saved_lists
+        (t1s, t2s, tdrops, alphas, gates_l,
+         cprev_l) = ctx_ag.saved_lists
         B, L, D, T, A, H, E, V, Dd, I = ctx_ag.dims
         p_fc = ctx_ag.p_fc
         p_lstm = ctx_ag.p_lstm
@@ -171,11 +170,8 @@ class DecoderBPTT(torch.autograd.Function):
 
         # ---- batched decode-MLP input grads (no recurrence involved) ----
         DHD = DL.matmul(wd2)                      # [T·B, Dd]
-        DP1 = torch.empty_like(DHD)               # dpre of dec fc_1
-        for t in range(T):
-            sl = slice(t * B, (t + 1) * B)
-            dhid = _drop(DHD[sl], seed, p_fc, t * 16 + 7)
-            DP1[sl] = _C.act_bwd(dhid, hid_l[t], ACT_TANH)
+        DHID = _C.hash_dropout_slabs(DHD, seed, p_fc, 7, 16, T)
+        DP1 = _C.act_bwd(DHID, HID, ACT_TANH)     # dpre of dec fc_1
         DEXPD = DP1.matmul(wd1)                   # [T·B, H+D+E]
 
         # transposed weights for the per-step skinny GEMMs (x @ W forms)

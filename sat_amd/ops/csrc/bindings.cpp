@@ -37,6 +37,9 @@ at::Tensor hash_dropout_steps(at::Tensor x, at::Tensor seed, double p,
                               int64_t salt_base, int64_t salt_stride,
                               int64_t T);
 void act_bwd_out(at::Tensor dy, at::Tensor y, int64_t act, at::Tensor out);
+at::Tensor hash_dropout_slabs(at::Tensor x, at::Tensor seed, double p,
+                              int64_t salt_base, int64_t salt_stride,
+                              int64_t T);
 std::vector<at::Tensor> lstm_pointwise_fwd(at::Tensor gates, at::Tensor c,
                                            double fb);
 std::vector<at::Tensor> lstm_pointwise_bwd(at::Tensor gates, at::Tensor c,
@@ -97,6 +100,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("hash_dropout_out", &hash_dropout_out);
     m.def("hash_dropout_steps", &hash_dropout_steps);
     m.def("act_bwd_out", &act_bwd_out);
+    m.def("hash_dropout_slabs", &hash_dropout_slabs);
     m.def("lstm_pointwise_fwd", &lstm_pointwise_fwd);
     m.def("lstm_pointwise_bwd", &lstm_pointwise_bwd);
     m.def("act_bwd", &act_bwd);

@@ -334,3 +334,47 @@ void act_bwd_out(at::Tensor dy, at::Tensor y, int64_t act, at::Tensor out) {
                        (bf16*)out.data_ptr(), n, (int)act);
     HIP_OK(hipGetLastError());
 }
+
+// ---- dropout over a step-major [T*B, N] buffer where slab t uses salt
+// salt_base + t*salt_stride and slab-local flat indices (matches the
+// per-step hash_dropout calls it replaces) ----
+
+__global__ void hash_dropout_slabs_kernel(const bf16* __restrict__ x,
+                                          const int64_t* __restrict__ seed_p,
+                                          bf16* __restrict__ y,
+                                          int64_t slab_elems, int T,
+                                          float p, int salt_base,
+                                          int salt_stride) {
+    const uint32_t seed = (uint32_t)(*seed_p);
+    int64_t i8 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+    int64_t total = slab_elems * T;
+    if (i8 >= total) return;
+    int t = (int)(i8 / slab_elems);
+    int64_t j = i8 % slab_elems;
+    int salt = salt_base + t * salt_stride;
+    bf16x8 v = *(const bf16x8*)(x + i8);
+    bf16x8 o;
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+        o[e] = f2bf(bf2f(v[e]) * dscale(seed, salt, (uint32_t)(j + e), p));
+    *(bf16x8*)(y + i8) = o;
+}
+
+at::Tensor hash_dropout_slabs(at::Tensor x, at::Tensor seed, double p,
+                              int64_t salt_base, int64_t salt_stride,
+                              int64_t T) {
+    if (p <= 0.0) return x;
+    int64_t n = x.numel();
+    int64_t slab = n / T;
+    TORCH_CHECK(slab % 8 == 0 && n % T == 0);
+    auto y = at::empty_like(x);
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(hash_dropout_slabs_kernel,
+                       dim3(cdiv(n / 8, 256)), dim3(256), 0, s,
+                       (const bf16*)x.data_ptr(),
+                       (const int64_t*)seed.data_ptr(),
+                       (bf16*)y.data_ptr(), slab, (int)T, (float)p,
+                       (int)salt_base, (int)salt_stride);
+    HIP_OK(hipGetLastError());
+    return y;
+}
