@@ -86,6 +86,8 @@ class DecoderBPTT(torch.autograd.Function):
         state_h = init_output
         last_word = torch.zeros(B, dtype=torch.int64, device=dev)
         empty_b = _EMPTY_B(dev)
+        fuse_small = (B <= 32 and (I + H) % 32 == 0
+                      and (H + D + E) % 32 == 0)
 
         # attend input for step 0 (later steps' come fused out of
         # expand_fuse at step t-1)
@@ -105,15 +107,25 @@ class DecoderBPTT(torch.autograd.Function):
             embt = _C.embedding_fwd(last_word, emb)
             _C.lstm_in_fuse(pooled, embt, state_h, seed, p_lstm, s + 3,
                             XH[sl])
-            gates = _C.dense_fwd(XH[sl], wl, bl, ACT_NONE)
-            h_raw, c_new = _C.lstm_pointwise_fwd(gates, memory, 1.0)
+            if fuse_small:
+                gates, h_raw, c_new = _C.dense_lstm_fwd(
+                    XH[sl], wl, bl, memory, 1.0)
+            else:
+                gates = _C.dense_fwd(XH[sl], wl, bl, ACT_NONE)
+                h_raw, c_new = _C.lstm_pointwise_fwd(gates, memory, 1.0)
             od_next = ODROP[(t + 1) * B:(t + 2) * B] if t + 1 < T \
                 else empty_b
             out_t, sth_t = _C.expand_fuse(
                 h_raw, pooled, embt, seed, EXPD[sl], od_next,
                 p_lstm, p_fc, s)
-            hid = _C.dense_fwd_out(EXPD[sl], wd1, bd1, ACT_TANH, HID[sl])
-            _C.hash_dropout_out(hid, seed, p_fc, s + 7, HD[sl])
+            if fuse_small:
+                _C.dense_drop_fwd(EXPD[sl], wd1, bd1, ACT_TANH, seed,
+                                  p_fc, s + 7, HID[sl], HD[sl])
+                hid = HID[sl]
+            else:
+                hid = _C.dense_fwd_out(EXPD[sl], wd1, bd1, ACT_TANH,
+                                       HID[sl])
+                _C.hash_dropout_out(hid, seed, p_fc, s + 7, HD[sl])
             _C.dense_fwd_out(HD[sl], wd2, bd2, ACT_NONE, LOGITS[sl])
 
             t1s.append(t1)

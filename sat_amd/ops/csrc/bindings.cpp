@@ -11,6 +11,12 @@ void scale_bias_act_nhwc(at::Tensor y, at::Tensor scale, at::Tensor shift,
 at::Tensor maxpool2x2_nhwc(at::Tensor input);
 at::Tensor conv_igemm_fwd(at::Tensor input, at::Tensor w_ohwi,
                           at::Tensor bias, bool relu);
+std::vector<at::Tensor> dense_lstm_fwd(at::Tensor xh, at::Tensor wl,
+                                       at::Tensor bl, at::Tensor c_prev,
+                                       double fb);
+void dense_drop_fwd(at::Tensor x, at::Tensor w, at::Tensor b, int64_t act,
+                    at::Tensor seed, double p, int64_t salt,
+                    at::Tensor y, at::Tensor ydrop);
 at::Tensor dense_fwd_out(at::Tensor x, at::Tensor w, at::Tensor bias,
                          int64_t act, at::Tensor out);
 std::vector<at::Tensor> lstm_pointwise_bwd_out(at::Tensor gates,
@@ -89,6 +95,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("scale_bias_act_nhwc", &scale_bias_act_nhwc);
     m.def("maxpool2x2_nhwc", &maxpool2x2_nhwc);
     m.def("conv_igemm_fwd", &conv_igemm_fwd);
+    m.def("dense_lstm_fwd", &dense_lstm_fwd);
+    m.def("dense_drop_fwd", &dense_drop_fwd);
     m.def("conv3_fwd", &conv3_fwd,
           "direct NHWC conv for 3-channel 3x3/s1 (VGG conv1_1)");
     m.def("dense_fwd_out", &dense_fwd_out);

@@ -370,3 +370,133 @@ at::Tensor dense_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
                      int64_t act) {
     return dense_fwd_out(x, w, bias, act, at::Tensor());
 }
+
+// ---- fused skinny epilogues for the BPTT step chain (B <= 32) ----
+
+// gates epilogue + LSTM pointwise: thread per (b,h) sums the 4 gate
+// elements across split-K slabs, applies TF LSTM gate math, writes the
+// bf16 gates (saved for backward) plus h_raw and c_new.
+__global__ void skinny_epi_lstm_kernel(const float* __restrict__ Yf,
+                                       const bf16* __restrict__ bias,
+                                       const bf16* __restrict__ c_prev,
+                                       bf16* __restrict__ gates,
+                                       bf16* __restrict__ h_out,
+                                       bf16* __restrict__ c_out,
+                                       int B, int H, int splitk,
+                                       float fb) {
+    int idx = blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= B * H) return;
+    int b = idx / H, hh = idx % H;
+    int64_t n = (int64_t)B * 4 * H;
+    float g[4];
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+        int64_t off = (int64_t)b * 4 * H + q * H + hh;
+        float acc = 0.f;
+        for (int k = 0; k < splitk; ++k) acc += Yf[k * n + off];
+        if (bias != nullptr) acc += bf2f(bias[q * H + hh]);
+        gates[off] = f2bf(acc);
+        g[q] = acc;
+    }
+    float gi = sigmoidf(g[0]);
+    float gj = tanhf(g[1]);
+    float gf = sigmoidf(g[2] + fb);
+    float go = sigmoidf(g[3]);
+    float cn = bf2f(c_prev[idx]) * gf + gi * gj;
+    h_out[idx] = f2bf(tanhf(cn) * go);
+    c_out[idx] = f2bf(cn);
+}
+
+std::vector<at::Tensor> dense_lstm_fwd(at::Tensor xh, at::Tensor wl,
+                                       at::Tensor bl, at::Tensor c_prev,
+                                       double fb) {
+    // gates = xh @ wl^T + bl ; (h_raw, c_new) = LSTM(gates, c_prev)
+    int64_t M = xh.size(0), K = xh.size(1), N = wl.size(0);
+    int B = c_prev.size(0), H = c_prev.size(1);
+    TORCH_CHECK(M == B && N == 4 * H && M <= 32 && K % 32 == 0);
+    auto gates = at::empty({M, N}, xh.options());
+    auto h_out = at::empty_like(c_prev);
+    auto c_out = at::empty_like(c_prev);
+    int nblocks = cdiv(N, 64);
+    int splitk = 1;
+    while (nblocks * splitk < 192 && splitk < 8 &&
+           (int)(K / 32) >= 2 * splitk)
+        splitk *= 2;
+    auto yf = at::empty({splitk, M, N}, xh.options().dtype(at::kFloat));
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(skinny_gemm_kernel, dim3(nblocks, splitk),
+                       dim3(256), 0, s,
+                       (const bf16*)xh.data_ptr(),
+                       (const bf16*)wl.data_ptr(), nullptr,
+                       nullptr, (float*)yf.data_ptr(), nullptr,
+                       (int)M, (int)N, (int)K, ACT_NONE, splitk);
+    hipLaunchKernelGGL(skinny_epi_lstm_kernel,
+                       dim3(cdiv(B * H, 256)), dim3(256), 0, s,
+                       (const float*)yf.data_ptr(),
+                       (const bf16*)bl.data_ptr(),
+                       (const bf16*)c_prev.data_ptr(),
+                       (bf16*)gates.data_ptr(), (bf16*)h_out.data_ptr(),
+                       (bf16*)c_out.data_ptr(), B, H, splitk, (float)fb);
+    HIP_OK(hipGetLastError());
+    return {gates, h_out, c_out};
+}
+
+// epilogue + activation + hash dropout: writes both y and dropout(y)
+__global__ void skinny_epi_drop_kernel(const float* __restrict__ Yf,
+                                       const bf16* __restrict__ bias,
+                                       const int64_t* __restrict__ seed_p,
+                                       bf16* __restrict__ y,
+                                       bf16* __restrict__ ydrop,
+                                       int64_t n, int N, int act,
+                                       int splitk, float p, int salt) {
+    const uint32_t seed = (uint32_t)(*seed_p);
+    int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= n) return;
+    float v = 0.f;
+    for (int k = 0; k < splitk; ++k) v += Yf[k * n + idx];
+    if (bias != nullptr) v += bf2f(bias[idx % N]);
+    v = apply_act(v, act);
+    y[idx] = f2bf(v);
+    float sc = 1.f;
+    if (p > 0.f) {
+        uint32_t h = (seed * 0x9E3779B1u) ^
+                     ((uint32_t)salt * 0x85EBCA77u) ^
+                     ((uint32_t)idx * 0xC2B2AE3Du);
+        h ^= h >> 16; h *= 0x7FEB352Du;
+        h ^= h >> 15; h *= 0x846CA68Bu;
+        h ^= h >> 16;
+        float u = (h >> 8) * (1.0f / 16777216.0f);
+        sc = u >= p ? 1.0f / (1.0f - p) : 0.0f;
+    }
+    ydrop[idx] = f2bf(v * sc);
+}
+
+void dense_drop_fwd(at::Tensor x, at::Tensor w, at::Tensor b, int64_t act,
+                    at::Tensor seed, double p, int64_t salt,
+                    at::Tensor y, at::Tensor ydrop) {
+    int64_t M = x.size(0), K = x.size(1), N = w.size(0);
+    TORCH_CHECK(M <= 32 && K % 32 == 0);
+    int nblocks = cdiv(N, 64);
+    int splitk = 1;
+    while (nblocks * splitk < 192 && splitk < 8 &&
+           (int)(K / 32) >= 2 * splitk)
+        splitk *= 2;
+    auto yf = at::empty({splitk, M, N}, x.options().dtype(at::kFloat));
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(skinny_gemm_kernel, dim3(nblocks, splitk),
+                       dim3(256), 0, s,
+                       (const bf16*)x.data_ptr(),
+                       (const bf16*)w.data_ptr(), nullptr,
+                       nullptr, (float*)yf.data_ptr(), nullptr,
+                       (int)M, (int)N, (int)K, ACT_NONE, splitk);
+    int64_t n = M * N;
+    const bf16* bias_ptr = (b.defined() && b.numel() > 0)
+        ? (const bf16*)b.data_ptr() : nullptr;
+    hipLaunchKernelGGL(skinny_epi_drop_kernel, dim3(cdiv(n, 256)),
+                       dim3(256), 0, s,
+                       (const float*)yf.data_ptr(), bias_ptr,
+                       (const int64_t*)seed.data_ptr(),
+                       (bf16*)y.data_ptr(), (bf16*)ydrop.data_ptr(),
+                       n, (int)N, (int)act, splitk, (float)p, (int)salt);
+    HIP_OK(hipGetLastError());
+}
