@@ -11,6 +11,10 @@ void scale_bias_act_nhwc(at::Tensor y, at::Tensor scale, at::Tensor shift,
 at::Tensor maxpool2x2_nhwc(at::Tensor input);
 at::Tensor conv_igemm_fwd(at::Tensor input, at::Tensor w_ohwi,
                           at::Tensor bias, bool relu);
+at::Tensor pad1_nhwc(at::Tensor input);
+at::Tensor conv_igemm_glds_fwd(at::Tensor padded, at::Tensor w_ohwi,
+                               at::Tensor bias, int64_t Hh, int64_t Ww,
+                               bool relu);
 std::vector<at::Tensor> dense_lstm_fwd(at::Tensor xh, at::Tensor wl,
                                        at::Tensor bl, at::Tensor c_prev,
                                        double fb);
@@ -95,6 +99,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("scale_bias_act_nhwc", &scale_bias_act_nhwc);
     m.def("maxpool2x2_nhwc", &maxpool2x2_nhwc);
     m.def("conv_igemm_fwd", &conv_igemm_fwd);
+    m.def("pad1_nhwc", &pad1_nhwc);
+    m.def("conv_igemm_glds_fwd", &conv_igemm_glds_fwd);
     m.def("dense_lstm_fwd", &dense_lstm_fwd);
     m.def("dense_drop_fwd", &dense_drop_fwd);
     m.def("conv3_fwd", &conv3_fwd,

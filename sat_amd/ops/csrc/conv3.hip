@@ -421,3 +421,200 @@ at::Tensor conv_igemm_fwd(at::Tensor input, at::Tensor w_ohwi,
     HIP_OK(hipGetLastError());
     return out;
 }
+
+// ---- glds-pipelined implicit-GEMM conv (Cin % 64 == 0, padded input) ----
+//
+// Same math as conv_igemm_kernel but staged with direct-to-LDS DMA
+// (global_load_lds width 16, the gfx950 staging lever) over a 1-pixel
+// zero-padded input, so no edge masks exist in the hot loop; each lane
+// caches its A/B row base addresses once.  128x128 tile, 4 waves 2x2.
+
+__global__ void pad1_nhwc_kernel(const bf16* __restrict__ in,
+                                 bf16* __restrict__ out,
+                                 int B, int H, int W, int C) {
+    // out: [B, H+2, W+2, C] zero-bordered copy of in
+    int Hp = H + 2, Wp = W + 2;
+    int64_t n = (int64_t)B * Hp * Wp * C;
+    int64_t i8 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+    if (i8 >= n) return;
+    int c = (int)(i8 % C);
+    int64_t pix = i8 / C;
+    int x = (int)(pix % Wp);
+    int64_t t = pix / Wp;
+    int y = (int)(t % Hp);
+    int b = (int)(t / Hp);
+    bf16x8 v = {};
+    if (y >= 1 && y <= H && x >= 1 && x <= W)
+        v = *(const bf16x8*)(in
+                             + (((int64_t)b * H + y - 1) * W + x - 1) * C
+                             + c);
+    *(bf16x8*)(out + i8) = v;
+}
+
+at::Tensor pad1_nhwc(at::Tensor input) {
+    CHECK_GPU(input); CHECK_BF16(input);
+    TORCH_CHECK(input.is_contiguous(at::MemoryFormat::ChannelsLast));
+    int B = input.size(0), C = input.size(1), H = input.size(2),
+        W = input.size(3);
+    TORCH_CHECK(C % 8 == 0);
+    auto out = at::empty({B, C, H + 2, W + 2},
+                         input.options()
+                             .memory_format(at::MemoryFormat::ChannelsLast));
+    int64_t n = (int64_t)B * (H + 2) * (W + 2) * C;
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(pad1_nhwc_kernel, dim3(cdiv(n / 8, 256)), dim3(256),
+                       0, s,
+                       (const bf16*)input.data_ptr(),
+                       (bf16*)out.data_ptr(), B, H, W, C);
+    HIP_OK(hipGetLastError());
+    return out;
+}
+
+__global__ __launch_bounds__(256)
+void conv_igemm_glds_kernel(const bf16* __restrict__ inp,  // padded NHWC
+                            const bf16* __restrict__ w,    // [Cout,9*Cin]
+                            const bf16* __restrict__ bias,
+                            bf16* __restrict__ out,
+                            int M, int Hh, int Ww, int Cin, int Cout,
+                            int relu) {
+    // output pixel grid is the UNPADDED HxW; input rows live in the
+    // padded (H+2)x(W+2) image so every (dy,dx) shift is in bounds.
+    constexpr int BM = 128, BN = 128, BKc = 64;
+    __shared__ bf16 lds[2 * BM * BKc];    // A then B, lane-linear
+    bf16* As = lds;
+    bf16* Bs = lds + BM * BKc;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int wr = wave >> 1, wc = wave & 1;
+    const int bm = blockIdx.y * BM;
+    const int bn = blockIdx.x * BN;
+    const int Wp = Ww + 2;
+
+    // per-lane cached row bases: this lane stages rows r = wave*32 + i*8
+    // + (lane>>3) of the A tile and the same rows of the B tile
+    int aBase[4];
+    int rowA[4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+        int r = wave * 32 + i * 8 + (lane >> 3);
+        rowA[i] = r;
+        int m = bm + r;
+        if (m >= M) m = M - 1;           // clamp: padded reads stay legal
+        int b = m / (Hh * Ww);
+        int yx = m % (Hh * Ww);
+        int y = yx / Ww, x = yx % Ww;
+        aBase[i] = ((b * (Hh + 2) + y + 1) * Wp + x + 1) * Cin;
+    }
+    const bf16* wBase[4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+        int cout = bn + wave * 32 + i * 8 + (lane >> 3);
+        if (cout >= Cout) cout = Cout - 1;
+        wBase[i] = w + (int64_t)cout * 9 * Cin;
+    }
+    const int ci8 = (lane & 7) * 8;      // this lane's 16B chunk
+
+    floatx4 acc[4][4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+            acc[i][j] = floatx4{0.f, 0.f, 0.f, 0.f};
+
+    const int lrow = lane & 15;
+    const int kgrp = lane >> 4;
+    const int KT = 9 * (Cin / BKc);
+
+    for (int kt = 0; kt < KT; ++kt) {
+        const int dxy = kt / (Cin / BKc);
+        const int ci0 = (kt % (Cin / BKc)) * BKc;
+        const int dy = dxy / 3 - 1, dx = dxy % 3 - 1;
+        const int delta = (dy * Wp + dx) * Cin + ci0 + ci8;
+
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) uint32_t*)
+                    (inp + aBase[i] + delta),
+                (__attribute__((address_space(3))) uint32_t*)
+                    (As + (wave * 32 + i * 8) * BKc),
+                16, 0, 0);
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) uint32_t*)
+                    (wBase[i] + dxy * Cin + ci0 + ci8),
+                (__attribute__((address_space(3))) uint32_t*)
+                    (Bs + (wave * 32 + i * 8) * BKc),
+                16, 0, 0);
+        }
+        __syncthreads();
+
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk) {
+            bf16x8 a_frag[4], b_frag[4];
+            const int kof = kk * 32 + kgrp * 8;
+#pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+                a_frag[mi] = *(const bf16x8*)(
+                    As + (wr * 64 + mi * 16 + lrow) * BKc + kof);
+#pragma unroll
+            for (int ni = 0; ni < 4; ++ni)
+                b_frag[ni] = *(const bf16x8*)(
+                    Bs + (wc * 64 + ni * 16 + lrow) * BKc + kof);
+#pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+                for (int ni = 0; ni < 4; ++ni)
+                    acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a_frag[mi], b_frag[ni], acc[mi][ni], 0, 0, 0);
+        }
+        __syncthreads();
+    }
+
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+        int col = bn + wc * 64 + ni * 16 + (lane & 15);
+        float bv = (bias != nullptr && col < Cout) ? bf2f(bias[col]) : 0.f;
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = bm + wr * 64 + mi * 16 + (lane >> 4) * 4 + r;
+                if (row < M && col < Cout) {
+                    float vv = acc[mi][ni][r] + bv;
+                    if (relu) vv = fmaxf(vv, 0.f);
+                    out[(int64_t)row * Cout + col] = f2bf(vv);
+                }
+            }
+        }
+    }
+}
+
+at::Tensor conv_igemm_glds_fwd(at::Tensor padded, at::Tensor w_ohwi,
+                               at::Tensor bias, int64_t Hh, int64_t Ww,
+                               bool relu) {
+    CHECK_GPU(padded); CHECK_BF16(padded);
+    CHECK_GPU(w_ohwi); CHECK_CONTIG(w_ohwi); CHECK_BF16(w_ohwi);
+    TORCH_CHECK(padded.is_contiguous(at::MemoryFormat::ChannelsLast));
+    int B = padded.size(0), Cin = padded.size(1);
+    TORCH_CHECK(padded.size(2) == Hh + 2 && padded.size(3) == Ww + 2);
+    int Cout = w_ohwi.size(0);
+    TORCH_CHECK(Cin % 64 == 0 && Cout % 8 == 0);
+    auto out = at::empty({B, Cout, Hh, Ww},
+                         padded.options()
+                             .memory_format(at::MemoryFormat::ChannelsLast));
+    const bf16* bias_ptr = nullptr;
+    if (bias.defined() && bias.numel() > 0)
+        bias_ptr = (const bf16*)bias.contiguous().data_ptr();
+    int M = B * Hh * Ww;
+    dim3 grid(cdiv(Cout, 128), cdiv(M, 128));
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(conv_igemm_glds_kernel, grid, dim3(256), 0, s,
+                       (const bf16*)padded.data_ptr(),
+                       (const bf16*)w_ohwi.data_ptr(), bias_ptr,
+                       (bf16*)out.data_ptr(), (int)M, (int)Hh, (int)Ww,
+                       (int)Cin, (int)Cout, relu ? 1 : 0);
+    HIP_OK(hipGetLastError());
+    return out;
+}
