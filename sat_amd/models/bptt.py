@@ -104,9 +104,8 @@ class DecoderBPTT(torch.autograd.Function):
                 t1, t2, v, seed, p_fc, s + 2, L)
             alpha, pooled = _C.attn_pool_fwd(contexts, att_logits)
 
-            embt = _C.embedding_fwd(last_word, emb)
-            _C.lstm_in_fuse(pooled, embt, state_h, seed, p_lstm, s + 3,
-                            XH[sl])
+            _C.lstm_in_fuse(pooled, emb, last_word, state_h, seed,
+                            p_lstm, s + 3, XH[sl])
             if fuse_small:
                 gates, h_raw, c_new = _C.dense_lstm_fwd(
                     XH[sl], wl, bl, memory, 1.0)
@@ -116,7 +115,7 @@ class DecoderBPTT(torch.autograd.Function):
             od_next = ODROP[(t + 1) * B:(t + 2) * B] if t + 1 < T \
                 else empty_b
             out_t, sth_t = _C.expand_fuse(
-                h_raw, pooled, embt, seed, EXPD[sl], od_next,
+                h_raw, pooled, emb, last_word, seed, EXPD[sl], od_next,
                 p_lstm, p_fc, s)
             if fuse_small:
                 _C.dense_drop_fwd(EXPD[sl], wd1, bd1, ACT_TANH, seed,

@@ -27,10 +27,12 @@ std::vector<at::Tensor> lstm_pointwise_bwd_out(at::Tensor gates,
                                                at::Tensor c, at::Tensor dh,
                                                at::Tensor dc, double fb,
                                                at::Tensor dgates_out);
-void lstm_in_fuse(at::Tensor pooled, at::Tensor emb, at::Tensor sth,
-                  at::Tensor seed, double p, int64_t salt, at::Tensor xh);
+void lstm_in_fuse(at::Tensor pooled, at::Tensor table, at::Tensor ids,
+                  at::Tensor sth, at::Tensor seed, double p, int64_t salt,
+                  at::Tensor xh);
 std::vector<at::Tensor> expand_fuse(at::Tensor h_raw, at::Tensor pooled,
-                                    at::Tensor emb, at::Tensor seed,
+                                    at::Tensor table, at::Tensor ids,
+                                    at::Tensor seed,
                                     at::Tensor expdrop, at::Tensor od_next,
                                     double p_lstm, double p_fc, int64_t s);
 std::vector<at::Tensor> dexp_fuse(at::Tensor dexpd, at::Tensor d_out_carry,

@@ -27,7 +27,8 @@ __device__ __forceinline__ float dscale(uint32_t seed, int salt,
 // ---- forward: xh = [dropout(cat(pooled, emb), p_lstm, salt), state_h] ----
 
 __global__ void lstm_in_fuse_kernel(const bf16* __restrict__ pooled,
-                                    const bf16* __restrict__ emb,
+                                    const bf16* __restrict__ table,
+                                    const int64_t* __restrict__ ids,
                                     const bf16* __restrict__ sth,
                                     const int64_t* __restrict__ seed_p,
                                     bf16* __restrict__ xh,
@@ -42,7 +43,7 @@ __global__ void lstm_in_fuse_kernel(const bf16* __restrict__ pooled,
     float val;
     if (j < I) {
         val = (j < D) ? bf2f(pooled[(int64_t)b * D + j])
-                      : bf2f(emb[(int64_t)b * E + (j - D)]);
+                      : bf2f(table[ids[b] * E + (j - D)]);
         val *= dscale(seed, salt, (uint32_t)(b * I + j), p);
     } else {
         val = bf2f(sth[(int64_t)b * H + (j - I)]);
@@ -50,16 +51,18 @@ __global__ void lstm_in_fuse_kernel(const bf16* __restrict__ pooled,
     xh[idx] = f2bf(val);
 }
 
-void lstm_in_fuse(at::Tensor pooled, at::Tensor emb, at::Tensor sth,
-                  at::Tensor seed, double p, int64_t salt, at::Tensor xh) {
-    int B = pooled.size(0), D = pooled.size(1), E = emb.size(1),
+void lstm_in_fuse(at::Tensor pooled, at::Tensor table, at::Tensor ids,
+                  at::Tensor sth, at::Tensor seed, double p, int64_t salt,
+                  at::Tensor xh) {
+    int B = pooled.size(0), D = pooled.size(1), E = table.size(1),
         H = sth.size(1);
     int64_t n = (int64_t)B * (D + E + H);
     hipStream_t s = at::cuda::getCurrentCUDAStream();
     hipLaunchKernelGGL(lstm_in_fuse_kernel, dim3(cdiv(n, 256)), dim3(256),
                        0, s,
                        (const bf16*)pooled.data_ptr(),
-                       (const bf16*)emb.data_ptr(),
+                       (const bf16*)table.data_ptr(),
+                       (const int64_t*)ids.data_ptr(),
                        (const bf16*)sth.data_ptr(),
                        (const int64_t*)seed.data_ptr(),
                        (bf16*)xh.data_ptr(), B, D, E, H,
@@ -72,7 +75,8 @@ void lstm_in_fuse(at::Tensor pooled, at::Tensor emb, at::Tensor sth,
 
 __global__ void expand_fuse_kernel(const bf16* __restrict__ h_raw,
                                    const bf16* __restrict__ pooled,
-                                   const bf16* __restrict__ emb,
+                                   const bf16* __restrict__ table,
+                                   const int64_t* __restrict__ ids,
                                    const int64_t* __restrict__ seed_p,
                                    bf16* __restrict__ out_t,
                                    bf16* __restrict__ sth_t,
@@ -100,19 +104,20 @@ __global__ void expand_fuse_kernel(const bf16* __restrict__ h_raw,
     } else if (j < H + D) {
         val = bf2f(pooled[(int64_t)b * D + (j - H)]);
     } else {
-        val = bf2f(emb[(int64_t)b * E + (j - H - D)]);
+        val = bf2f(table[ids[b] * E + (j - H - D)]);
     }
     expdrop[idx] = f2bf(val * dscale(seed, s + 6, (uint32_t)(b * W + j),
                                      p_fc));
 }
 
 std::vector<at::Tensor> expand_fuse(at::Tensor h_raw, at::Tensor pooled,
-                                    at::Tensor emb, at::Tensor seed,
+                                    at::Tensor table, at::Tensor ids,
+                                    at::Tensor seed,
                                     at::Tensor expdrop, at::Tensor od_next,
                                     double p_lstm, double p_fc,
                                     int64_t s) {
     int B = h_raw.size(0), H = h_raw.size(1), D = pooled.size(1),
-        E = emb.size(1);
+        E = table.size(1);
     auto out_t = at::empty_like(h_raw);
     auto sth_t = at::empty_like(h_raw);
     bf16* od_ptr = nullptr;
@@ -124,7 +129,8 @@ std::vector<at::Tensor> expand_fuse(at::Tensor h_raw, at::Tensor pooled,
                        0, st,
                        (const bf16*)h_raw.data_ptr(),
                        (const bf16*)pooled.data_ptr(),
-                       (const bf16*)emb.data_ptr(),
+                       (const bf16*)table.data_ptr(),
+                       (const int64_t*)ids.data_ptr(),
                        (const int64_t*)seed.data_ptr(),
                        (bf16*)out_t.data_ptr(), (bf16*)sth_t.data_ptr(),
                        (bf16*)expdrop.data_ptr(), od_ptr,
