@@ -50,6 +50,14 @@ class Config(object):
         self.beta2 = 0.999
         self.epsilon = 1e-6
 
+        # Clusterone environment paths (config.py:46-50) — kept for knob
+        # parity; the MI355X launch path (torchrun/RCCL) does not use them
+        self.path_to_local_logs = './summary'
+        self.root_path_to_local_data = './data'
+        self.local_repo = 'show-attend-and-tell'
+        self.cloud_user_repo = 'Cheng-Lin-Li/show-attend-and-tell'
+        self.cloud_path_to_data = 'data'
+
         # about the saver (config.py:53-55)
         self.save_period = 50
         self.save_dir = './data/models/'
@@ -75,7 +83,7 @@ class Config(object):
         self.eval_caption_file = './data/val/captions_val2014.json'
         self.eval_result_dir = './data/val/results/'
         self.eval_result_file = './data/val/results.json'
-        self.save_eval_result_as_image = False
+        self.save_eval_result_as_image = True  # reference config.py:81
 
         # about the testing (config.py:83-85)
         self.test_image_dir = './data/test/images/'

@@ -134,6 +134,15 @@ class BaseModel(object):
                 if (self.global_step + 1) % config.save_period == 0 \
                         and self.is_chief:
                     self.save()
+                    if writer is not None:
+                        # per-variable mean/std/max/min stats (reference
+                        # model.py:534-542 logs these per step; we log at
+                        # checkpoint cadence)
+                        for name, prm in \
+                                self.model.named_parameters():
+                            if prm.requires_grad:
+                                writer.variable_summary(
+                                    name, prm, self.global_step)
                 if writer is not None:
                     writer.add_scalars(
                         {k: v.item() for k, v in out.items()

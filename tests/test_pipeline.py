@@ -76,3 +76,46 @@ def test_cli_flags_parity():
     assert args.model_file == 'x.npy'
     assert args.cnn_model_file == 'c.npy'
     assert args.beam_size == 5
+
+
+def test_trim_model_cli(tiny_config, tmp_path):
+    """tools/trim_model.py end-to-end (reference trim_model.py parity)."""
+    import subprocess
+    import sys
+    import numpy as np
+    from sat_amd.models.base_model import BaseModel
+    cfg = tiny_config
+    m = BaseModel(cfg)
+    path = m.save()
+    out = str(tmp_path / 'trimmed.npy')
+    r = subprocess.run(
+        [sys.executable, 'tools/trim_model.py', path, out],
+        capture_output=True, text=True,
+        cwd=os.path.dirname(os.path.dirname(
+            os.path.abspath(__file__))))
+    assert r.returncode == 0, r.stderr
+    arrays = np.load(out, allow_pickle=True).item()
+    assert not any('optimizer' in k for k in arrays)
+
+
+def test_eval_sh_exists_and_executable():
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    p = os.path.join(root, 'eval.sh')
+    assert os.path.exists(p)
+    assert os.access(p, os.X_OK)
+
+
+def test_variable_summaries_written(tiny_config):
+    """Per-variable stats land in the event file at checkpoint cadence."""
+    import json as _json
+    from sat_amd.data.dataset import prepare_train_data as prep
+    from sat_amd.models.base_model import BaseModel
+    cfg = tiny_config
+    cfg.save_period = 2
+    data = prep(cfg)
+    m = BaseModel(cfg)
+    m.train(data)
+    path = os.path.join(cfg.summary_dir, 'events.jsonl')
+    recs = [_json.loads(l) for l in open(path)]
+    assert any(any(k.endswith('/mean') for k in r) for r in recs)
+    assert any('total_loss' in r for r in recs)
