@@ -535,15 +535,21 @@ void conv_igemm_glds_kernel(const bf16* __restrict__ inp,  // padded NHWC
 
 #pragma unroll
         for (int i = 0; i < 4; ++i) {
+            // st_16x32 XOR swizzle (guide: bank-spread for ds_read_b128
+            // from 128B rows): physical chunk (lane&7) holds LOGICAL
+            // chunk (lane&7)^2 on rows with bit2 set, so the per-lane
+            // GLOBAL address is pre-swizzled (glds LDS side stays
+            // lane-linear) and fragment reads XOR the same bit.
+            int swz = (rowA[i] & 4) ? 16 : 0;   // elements (=32 B)
             __builtin_amdgcn_global_load_lds(
                 (const __attribute__((address_space(1))) uint32_t*)
-                    (inp + aBase[i] + delta),
+                    (inp + aBase[i] + delta + (ci8 ^ swz) - ci8),
                 (__attribute__((address_space(3))) uint32_t*)
                     (As + (wave * 32 + i * 8) * BKc),
                 16, 0, 0);
             __builtin_amdgcn_global_load_lds(
                 (const __attribute__((address_space(1))) uint32_t*)
-                    (wBase[i] + dxy * Cin + ci0 + ci8),
+                    (wBase[i] + dxy * Cin + ci0 + (ci8 ^ swz)),
                 (__attribute__((address_space(3))) uint32_t*)
                     (Bs + (wave * 32 + i * 8) * BKc),
                 16, 0, 0);
@@ -555,13 +561,17 @@ void conv_igemm_glds_kernel(const bf16* __restrict__ inp,  // padded NHWC
             bf16x8 a_frag[4], b_frag[4];
             const int kof = kk * 32 + kgrp * 8;
 #pragma unroll
-            for (int mi = 0; mi < 4; ++mi)
+            for (int mi = 0; mi < 4; ++mi) {
+                int r = wr * 64 + mi * 16 + lrow;
                 a_frag[mi] = *(const bf16x8*)(
-                    As + (wr * 64 + mi * 16 + lrow) * BKc + kof);
+                    As + r * BKc + (kof ^ ((r & 4) ? 16 : 0)));
+            }
 #pragma unroll
-            for (int ni = 0; ni < 4; ++ni)
+            for (int ni = 0; ni < 4; ++ni) {
+                int r = wc * 64 + ni * 16 + lrow;
                 b_frag[ni] = *(const bf16x8*)(
-                    Bs + (wc * 64 + ni * 16 + lrow) * BKc + kof);
+                    Bs + r * BKc + (kof ^ ((r & 4) ? 16 : 0)));
+            }
 #pragma unroll
             for (int mi = 0; mi < 4; ++mi)
 #pragma unroll

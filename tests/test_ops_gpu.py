@@ -335,3 +335,20 @@ def test_conv_igemm_matches_torch():
         ref = torch.relu(torch.nn.functional.conv2d(
             x.float(), w.float(), b.float(), padding=1))
         assert _rel_err(y, ref) < 2e-2, (Cin, Cout, H)
+
+
+def test_conv_igemm_glds_matches_torch():
+    from sat_amd import _C
+    torch.manual_seed(10)
+    for Cin, Cout, H, W in [(64, 128, 19, 23), (128, 128, 14, 14),
+                            (128, 256, 9, 31)]:
+        x = torch.randn(2, Cin, H, W).to(DEV, torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+        w = (torch.randn(Cout, Cin, 3, 3) * 0.05).to(DEV, torch.bfloat16)
+        b = torch.randn(Cout).to(DEV, torch.bfloat16)
+        w_ohwi = w.permute(0, 2, 3, 1).contiguous().reshape(Cout, -1)
+        xp = _C.pad1_nhwc(x)
+        y = _C.conv_igemm_glds_fwd(xp, w_ohwi, b, H, W, True)
+        ref = torch.relu(torch.nn.functional.conv2d(
+            x.float(), w.float(), b.float(), padding=1))
+        assert _rel_err(y, ref) < 2e-2, (Cin, Cout, H, W)
