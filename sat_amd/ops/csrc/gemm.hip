@@ -156,8 +156,9 @@ void tiled_gemm_kernel(const bf16* __restrict__ A,    // [M,K]
 // otherwise fp32 atomics into a workspace + separate epilogue.
 // ---------------------------------------------------------------------
 
+template <int RF>
 __global__ __launch_bounds__(256)
-void skinny_gemm_kernel(const bf16* __restrict__ A,   // [M,K], M <= 32
+void skinny_gemm_kernel(const bf16* __restrict__ A,   // [M,K], M <= RF*16
                         const bf16* __restrict__ W,   // [N,K]
                         const bf16* __restrict__ bias,
                         bf16* __restrict__ Y,         // [M,N]
@@ -178,9 +179,10 @@ void skinny_gemm_kernel(const bf16* __restrict__ A,   // [M,K], M <= 32
     const int lrow = lane & 15;
     const int kgrp = lane >> 4;
 
-    floatx4 acc[2];
-    acc[0] = floatx4{0.f, 0.f, 0.f, 0.f};
-    acc[1] = floatx4{0.f, 0.f, 0.f, 0.f};
+    floatx4 acc[RF];
+#pragma unroll
+    for (int i = 0; i < RF; ++i)
+        acc[i] = floatx4{0.f, 0.f, 0.f, 0.f};
 
     const bool wfull = (n0 + 16 <= N);
     if (active) {
@@ -197,24 +199,21 @@ void skinny_gemm_kernel(const bf16* __restrict__ A,   // [M,K], M <= 32
                     for (int e = 0; e < 8 && kof + e < K; ++e)
                         b_frag[e] = W[(int64_t)gb * K + kof + e];
             }
-            bf16x8 a0 = {}, a1 = {};
-            if (kof + 8 <= K) {
-                if (lrow < M)
-                    a0 = *(const bf16x8*)(A + (int64_t)lrow * K + kof);
-                if (16 + lrow < M)
-                    a1 = *(const bf16x8*)(A + (int64_t)(16 + lrow) * K + kof);
-            } else if (kof < K) {
-                if (lrow < M)
-                    for (int e = 0; e < 8 && kof + e < K; ++e)
-                        a0[e] = A[(int64_t)lrow * K + kof + e];
-                if (16 + lrow < M)
-                    for (int e = 0; e < 8 && kof + e < K; ++e)
-                        a1[e] = A[(int64_t)(16 + lrow) * K + kof + e];
+#pragma unroll
+            for (int mi = 0; mi < RF; ++mi) {
+                bf16x8 a0 = {};
+                int ar = mi * 16 + lrow;
+                if (kof + 8 <= K) {
+                    if (ar < M)
+                        a0 = *(const bf16x8*)(A + (int64_t)ar * K + kof);
+                } else if (kof < K) {
+                    if (ar < M)
+                        for (int e = 0; e < 8 && kof + e < K; ++e)
+                            a0[e] = A[(int64_t)ar * K + kof + e];
+                }
+                acc[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a0, b_frag, acc[mi], 0, 0, 0);
             }
-            acc[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b_frag,
-                                                             acc[0], 0, 0, 0);
-            acc[1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b_frag,
-                                                             acc[1], 0, 0, 0);
         }
     }
 
@@ -223,7 +222,7 @@ void skinny_gemm_kernel(const bf16* __restrict__ A,   // [M,K], M <= 32
         if (!active || col >= N) return;
         float bv = (bias != nullptr) ? bf2f(bias[col]) : 0.f;
 #pragma unroll
-        for (int mi = 0; mi < 2; ++mi)
+        for (int mi = 0; mi < RF; ++mi)
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 int row = mi * 16 + (lane >> 4) * 4 + r;
@@ -239,7 +238,7 @@ void skinny_gemm_kernel(const bf16* __restrict__ A,   // [M,K], M <= 32
     if (active && col < N) {
         float* slab = Yf + (int64_t)kq * M * N;
 #pragma unroll
-        for (int mi = 0; mi < 2; ++mi)
+        for (int mi = 0; mi < RF; ++mi)
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 int row = mi * 16 + (lane >> 4) * 4 + r;
@@ -315,29 +314,35 @@ at::Tensor dense_fwd_out(at::Tensor x, at::Tensor w, at::Tensor bias,
     }
     hipStream_t stream = at::cuda::getCurrentCUDAStream();
 
-    if (M <= 32 && K % 32 == 0) {
+    if (M <= 64 && K % 32 == 0) {
         // skinny path with split-K chosen to fill the 256-CU chip
         int nblocks = cdiv(N, 64);
         int splitk = 1;
         while (nblocks * splitk < 192 && splitk < 8 &&
                (int)(K / 32) >= 2 * splitk)
             splitk *= 2;
+#define LAUNCH_SKINNY(RF, GRID, BIAS, YPTR, YFPTR, SPLITK) \
+        hipLaunchKernelGGL((skinny_gemm_kernel<RF>), GRID, dim3(256), 0, \
+                           stream, (const bf16*)x.data_ptr(), \
+                           (const bf16*)w.data_ptr(), BIAS, YPTR, YFPTR, \
+                           nullptr, (int)M, (int)N, (int)K, (int)act, \
+                           SPLITK)
         if (splitk == 1) {
-            hipLaunchKernelGGL(skinny_gemm_kernel, dim3(nblocks, 1),
-                               dim3(256), 0, stream,
-                               (const bf16*)x.data_ptr(),
-                               (const bf16*)w.data_ptr(), bias_ptr,
-                               (bf16*)y.data_ptr(), nullptr, nullptr,
-                               (int)M, (int)N, (int)K, (int)act, 1);
+            if (M <= 32)
+                LAUNCH_SKINNY(2, dim3(nblocks, 1), bias_ptr,
+                              (bf16*)y.data_ptr(), nullptr, 1);
+            else
+                LAUNCH_SKINNY(4, dim3(nblocks, 1), bias_ptr,
+                              (bf16*)y.data_ptr(), nullptr, 1);
         } else {
             auto yf = at::empty({splitk, M, N},
                                 x.options().dtype(at::kFloat));
-            hipLaunchKernelGGL(skinny_gemm_kernel, dim3(nblocks, splitk),
-                               dim3(256), 0, stream,
-                               (const bf16*)x.data_ptr(),
-                               (const bf16*)w.data_ptr(), nullptr,
-                               nullptr, (float*)yf.data_ptr(), nullptr,
-                               (int)M, (int)N, (int)K, (int)act, splitk);
+            if (M <= 32)
+                LAUNCH_SKINNY(2, dim3(nblocks, splitk), nullptr, nullptr,
+                              (float*)yf.data_ptr(), splitk);
+            else
+                LAUNCH_SKINNY(4, dim3(nblocks, splitk), nullptr, nullptr,
+                              (float*)yf.data_ptr(), splitk);
             int64_t n = M * N;
             hipLaunchKernelGGL(skinny_epilogue_kernel,
                                dim3(cdiv(n, 256)), dim3(256), 0, stream,
@@ -345,6 +350,7 @@ at::Tensor dense_fwd_out(at::Tensor x, at::Tensor w, at::Tensor bias,
                                (bf16*)y.data_ptr(), n, (int)N, (int)act,
                                splitk);
         }
+#undef LAUNCH_SKINNY
     } else if (N <= 512) {
         dim3 grid(cdiv(N, 64), cdiv(M, 128));
         hipLaunchKernelGGL((tiled_gemm_kernel<4, 1, 2, 4>), grid, dim3(256),
@@ -413,7 +419,7 @@ std::vector<at::Tensor> dense_lstm_fwd(at::Tensor xh, at::Tensor wl,
     // gates = xh @ wl^T + bl ; (h_raw, c_new) = LSTM(gates, c_prev)
     int64_t M = xh.size(0), K = xh.size(1), N = wl.size(0);
     int B = c_prev.size(0), H = c_prev.size(1);
-    TORCH_CHECK(M == B && N == 4 * H && M <= 32 && K % 32 == 0);
+    TORCH_CHECK(M == B && N == 4 * H && M <= 64 && K % 32 == 0);
     auto gates = at::empty({M, N}, xh.options());
     auto h_out = at::empty_like(c_prev);
     auto c_out = at::empty_like(c_prev);
@@ -424,12 +430,20 @@ std::vector<at::Tensor> dense_lstm_fwd(at::Tensor xh, at::Tensor wl,
         splitk *= 2;
     auto yf = at::empty({splitk, M, N}, xh.options().dtype(at::kFloat));
     hipStream_t s = at::cuda::getCurrentCUDAStream();
-    hipLaunchKernelGGL(skinny_gemm_kernel, dim3(nblocks, splitk),
-                       dim3(256), 0, s,
-                       (const bf16*)xh.data_ptr(),
-                       (const bf16*)wl.data_ptr(), nullptr,
-                       nullptr, (float*)yf.data_ptr(), nullptr,
-                       (int)M, (int)N, (int)K, ACT_NONE, splitk);
+    if (M <= 32)
+        hipLaunchKernelGGL((skinny_gemm_kernel<2>), dim3(nblocks, splitk),
+                           dim3(256), 0, s,
+                           (const bf16*)xh.data_ptr(),
+                           (const bf16*)wl.data_ptr(), nullptr,
+                           nullptr, (float*)yf.data_ptr(), nullptr,
+                           (int)M, (int)N, (int)K, ACT_NONE, splitk);
+    else
+        hipLaunchKernelGGL((skinny_gemm_kernel<4>), dim3(nblocks, splitk),
+                           dim3(256), 0, s,
+                           (const bf16*)xh.data_ptr(),
+                           (const bf16*)wl.data_ptr(), nullptr,
+                           nullptr, (float*)yf.data_ptr(), nullptr,
+                           (int)M, (int)N, (int)K, ACT_NONE, splitk);
     hipLaunchKernelGGL(skinny_epi_lstm_kernel,
                        dim3(cdiv(B * H, 256)), dim3(256), 0, s,
                        (const float*)yf.data_ptr(),
@@ -475,7 +489,7 @@ void dense_drop_fwd(at::Tensor x, at::Tensor w, at::Tensor b, int64_t act,
                     at::Tensor seed, double p, int64_t salt,
                     at::Tensor y, at::Tensor ydrop) {
     int64_t M = x.size(0), K = x.size(1), N = w.size(0);
-    TORCH_CHECK(M <= 32 && K % 32 == 0);
+    TORCH_CHECK(M <= 64 && K % 32 == 0);
     int nblocks = cdiv(N, 64);
     int splitk = 1;
     while (nblocks * splitk < 192 && splitk < 8 &&
@@ -483,12 +497,20 @@ void dense_drop_fwd(at::Tensor x, at::Tensor w, at::Tensor b, int64_t act,
         splitk *= 2;
     auto yf = at::empty({splitk, M, N}, x.options().dtype(at::kFloat));
     hipStream_t s = at::cuda::getCurrentCUDAStream();
-    hipLaunchKernelGGL(skinny_gemm_kernel, dim3(nblocks, splitk),
-                       dim3(256), 0, s,
-                       (const bf16*)x.data_ptr(),
-                       (const bf16*)w.data_ptr(), nullptr,
-                       nullptr, (float*)yf.data_ptr(), nullptr,
-                       (int)M, (int)N, (int)K, ACT_NONE, splitk);
+    if (M <= 32)
+        hipLaunchKernelGGL((skinny_gemm_kernel<2>), dim3(nblocks, splitk),
+                           dim3(256), 0, s,
+                           (const bf16*)x.data_ptr(),
+                           (const bf16*)w.data_ptr(), nullptr,
+                           nullptr, (float*)yf.data_ptr(), nullptr,
+                           (int)M, (int)N, (int)K, ACT_NONE, splitk);
+    else
+        hipLaunchKernelGGL((skinny_gemm_kernel<4>), dim3(nblocks, splitk),
+                           dim3(256), 0, s,
+                           (const bf16*)x.data_ptr(),
+                           (const bf16*)w.data_ptr(), nullptr,
+                           nullptr, (float*)yf.data_ptr(), nullptr,
+                           (int)M, (int)N, (int)K, ACT_NONE, splitk);
     int64_t n = M * N;
     const bf16* bias_ptr = (b.defined() && b.numel() > 0)
         ? (const bf16*)b.data_ptr() : nullptr;
