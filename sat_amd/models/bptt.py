@@ -86,7 +86,7 @@ class DecoderBPTT(torch.autograd.Function):
         state_h = init_output
         last_word = torch.zeros(B, dtype=torch.int64, device=dev)
         empty_b = _EMPTY_B(dev)
-        fuse_small = (B <= 64 and (I + H) % 32 == 0
+        fuse_small = (B <= 128 and (I + H) % 32 == 0
                       and (H + D + E) % 32 == 0)
 
         # attend input for step 0 (later steps' come fused out of

@@ -314,7 +314,7 @@ at::Tensor dense_fwd_out(at::Tensor x, at::Tensor w, at::Tensor bias,
     }
     hipStream_t stream = at::cuda::getCurrentCUDAStream();
 
-    if (M <= 64 && K % 32 == 0) {
+    if (M <= 128 && K % 32 == 0) {
         // skinny path with split-K chosen to fill the 256-CU chip
         int nblocks = cdiv(N, 64);
         int splitk = 1;
@@ -331,8 +331,11 @@ at::Tensor dense_fwd_out(at::Tensor x, at::Tensor w, at::Tensor bias,
             if (M <= 32)
                 LAUNCH_SKINNY(2, dim3(nblocks, 1), bias_ptr,
                               (bf16*)y.data_ptr(), nullptr, 1);
-            else
+            else if (M <= 64)
                 LAUNCH_SKINNY(4, dim3(nblocks, 1), bias_ptr,
+                              (bf16*)y.data_ptr(), nullptr, 1);
+            else
+                LAUNCH_SKINNY(8, dim3(nblocks, 1), bias_ptr,
                               (bf16*)y.data_ptr(), nullptr, 1);
         } else {
             auto yf = at::empty({splitk, M, N},
@@ -340,8 +343,11 @@ at::Tensor dense_fwd_out(at::Tensor x, at::Tensor w, at::Tensor bias,
             if (M <= 32)
                 LAUNCH_SKINNY(2, dim3(nblocks, splitk), nullptr, nullptr,
                               (float*)yf.data_ptr(), splitk);
-            else
+            else if (M <= 64)
                 LAUNCH_SKINNY(4, dim3(nblocks, splitk), nullptr, nullptr,
+                              (float*)yf.data_ptr(), splitk);
+            else
+                LAUNCH_SKINNY(8, dim3(nblocks, splitk), nullptr, nullptr,
                               (float*)yf.data_ptr(), splitk);
             int64_t n = M * N;
             hipLaunchKernelGGL(skinny_epilogue_kernel,
@@ -419,7 +425,7 @@ std::vector<at::Tensor> dense_lstm_fwd(at::Tensor xh, at::Tensor wl,
     // gates = xh @ wl^T + bl ; (h_raw, c_new) = LSTM(gates, c_prev)
     int64_t M = xh.size(0), K = xh.size(1), N = wl.size(0);
     int B = c_prev.size(0), H = c_prev.size(1);
-    TORCH_CHECK(M == B && N == 4 * H && M <= 64 && K % 32 == 0);
+    TORCH_CHECK(M == B && N == 4 * H && M <= 128 && K % 32 == 0);
     auto gates = at::empty({M, N}, xh.options());
     auto h_out = at::empty_like(c_prev);
     auto c_out = at::empty_like(c_prev);
@@ -430,20 +436,16 @@ std::vector<at::Tensor> dense_lstm_fwd(at::Tensor xh, at::Tensor wl,
         splitk *= 2;
     auto yf = at::empty({splitk, M, N}, xh.options().dtype(at::kFloat));
     hipStream_t s = at::cuda::getCurrentCUDAStream();
-    if (M <= 32)
-        hipLaunchKernelGGL((skinny_gemm_kernel<2>), dim3(nblocks, splitk),
-                           dim3(256), 0, s,
-                           (const bf16*)xh.data_ptr(),
-                           (const bf16*)wl.data_ptr(), nullptr,
-                           nullptr, (float*)yf.data_ptr(), nullptr,
-                           (int)M, (int)N, (int)K, ACT_NONE, splitk);
-    else
-        hipLaunchKernelGGL((skinny_gemm_kernel<4>), dim3(nblocks, splitk),
-                           dim3(256), 0, s,
-                           (const bf16*)xh.data_ptr(),
-                           (const bf16*)wl.data_ptr(), nullptr,
-                           nullptr, (float*)yf.data_ptr(), nullptr,
-                           (int)M, (int)N, (int)K, ACT_NONE, splitk);
+#define LAUNCH_SLSTM(RF) \
+    hipLaunchKernelGGL((skinny_gemm_kernel<RF>), dim3(nblocks, splitk), \
+                       dim3(256), 0, s, (const bf16*)xh.data_ptr(), \
+                       (const bf16*)wl.data_ptr(), nullptr, nullptr, \
+                       (float*)yf.data_ptr(), nullptr, (int)M, (int)N, \
+                       (int)K, ACT_NONE, splitk)
+    if (M <= 32) LAUNCH_SLSTM(2);
+    else if (M <= 64) LAUNCH_SLSTM(4);
+    else LAUNCH_SLSTM(8);
+#undef LAUNCH_SLSTM
     hipLaunchKernelGGL(skinny_epi_lstm_kernel,
                        dim3(cdiv(B * H, 256)), dim3(256), 0, s,
                        (const float*)yf.data_ptr(),
@@ -489,7 +491,7 @@ void dense_drop_fwd(at::Tensor x, at::Tensor w, at::Tensor b, int64_t act,
                     at::Tensor seed, double p, int64_t salt,
                     at::Tensor y, at::Tensor ydrop) {
     int64_t M = x.size(0), K = x.size(1), N = w.size(0);
-    TORCH_CHECK(M <= 64 && K % 32 == 0);
+    TORCH_CHECK(M <= 128 && K % 32 == 0);
     int nblocks = cdiv(N, 64);
     int splitk = 1;
     while (nblocks * splitk < 192 && splitk < 8 &&
@@ -497,20 +499,16 @@ void dense_drop_fwd(at::Tensor x, at::Tensor w, at::Tensor b, int64_t act,
         splitk *= 2;
     auto yf = at::empty({splitk, M, N}, x.options().dtype(at::kFloat));
     hipStream_t s = at::cuda::getCurrentCUDAStream();
-    if (M <= 32)
-        hipLaunchKernelGGL((skinny_gemm_kernel<2>), dim3(nblocks, splitk),
-                           dim3(256), 0, s,
-                           (const bf16*)x.data_ptr(),
-                           (const bf16*)w.data_ptr(), nullptr,
-                           nullptr, (float*)yf.data_ptr(), nullptr,
-                           (int)M, (int)N, (int)K, ACT_NONE, splitk);
-    else
-        hipLaunchKernelGGL((skinny_gemm_kernel<4>), dim3(nblocks, splitk),
-                           dim3(256), 0, s,
-                           (const bf16*)x.data_ptr(),
-                           (const bf16*)w.data_ptr(), nullptr,
-                           nullptr, (float*)yf.data_ptr(), nullptr,
-                           (int)M, (int)N, (int)K, ACT_NONE, splitk);
+#define LAUNCH_SDROP(RF) \
+    hipLaunchKernelGGL((skinny_gemm_kernel<RF>), dim3(nblocks, splitk), \
+                       dim3(256), 0, s, (const bf16*)x.data_ptr(), \
+                       (const bf16*)w.data_ptr(), nullptr, nullptr, \
+                       (float*)yf.data_ptr(), nullptr, (int)M, (int)N, \
+                       (int)K, ACT_NONE, splitk)
+    if (M <= 32) LAUNCH_SDROP(2);
+    else if (M <= 64) LAUNCH_SDROP(4);
+    else LAUNCH_SDROP(8);
+#undef LAUNCH_SDROP
     int64_t n = M * N;
     const bf16* bias_ptr = (b.defined() && b.numel() > 0)
         ? (const bf16*)b.data_ptr() : nullptr;
