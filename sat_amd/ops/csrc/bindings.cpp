@@ -60,7 +60,6 @@ std::vector<at::Tensor> lstm_pointwise_bwd(at::Tensor gates, at::Tensor c,
 at::Tensor act_bwd(at::Tensor dy, at::Tensor y, int64_t act);
 at::Tensor hash_dropout(at::Tensor x, at::Tensor seed, double p,
                         int64_t salt);
-at::Tensor attn_score_fwd(at::Tensor temp, at::Tensor v);
 std::vector<at::Tensor> attn_pool_fwd(at::Tensor ctx, at::Tensor logits);
 std::vector<at::Tensor> attn_scores_fused(at::Tensor t1, at::Tensor t2,
                                           at::Tensor v, at::Tensor seed,
@@ -121,7 +120,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("lstm_pointwise_bwd", &lstm_pointwise_bwd);
     m.def("act_bwd", &act_bwd);
     m.def("hash_dropout", &hash_dropout);
-    m.def("attn_score_fwd", &attn_score_fwd);
     m.def("attn_pool_fwd", &attn_pool_fwd);
     m.def("attn_scores_fused", &attn_scores_fused);
     m.def("attn_pool_bwd", &attn_pool_bwd);

@@ -144,41 +144,6 @@ at::Tensor act_bwd(at::Tensor dy, at::Tensor y, int64_t act) {
 // LDS-staged softmax-over-L + weighted context sum (model.py:435,263-264).
 // ======================================================================
 
-__global__ void attn_score_kernel(const bf16* __restrict__ temp,
-                                  const bf16* __restrict__ v,
-                                  float* __restrict__ logits,
-                                  int M, int A) {
-    // 4 waves per block, one row per wave
-    int row = blockIdx.x * 4 + (threadIdx.x >> 6);
-    if (row >= M) return;
-    int lane = threadIdx.x & 63;
-    const bf16* t = temp + (int64_t)row * A;
-    float acc = 0.f;
-    // A % 8 == 0: 8-element chunks tile A exactly; chunk c = lane + 64k
-    for (int a0 = lane * 8; a0 + 8 <= A; a0 += 64 * 8) {
-        bf16x8 tv = *(const bf16x8*)(t + a0);
-        bf16x8 vv = *(const bf16x8*)(v + a0);
-#pragma unroll
-        for (int e = 0; e < 8; ++e) acc += bf2f(tv[e]) * bf2f(vv[e]);
-    }
-    acc = wave_sum(acc);
-    if (lane == 0) logits[row] = acc;
-}
-
-at::Tensor attn_score_fwd(at::Tensor temp, at::Tensor v) {
-    CHECK_GPU(temp); CHECK_CONTIG(temp); CHECK_BF16(temp);
-    CHECK_GPU(v); CHECK_CONTIG(v); CHECK_BF16(v);
-    int M = temp.size(0), A = temp.size(1);
-    TORCH_CHECK(A % 8 == 0, "A must be a multiple of 8 (got ", A, ")");
-    auto logits = at::empty({M}, temp.options().dtype(at::kFloat));
-    hipStream_t s = at::cuda::getCurrentCUDAStream();
-    hipLaunchKernelGGL(attn_score_kernel, dim3(cdiv(M, 4)), dim3(256), 0, s,
-                       (const bf16*)temp.data_ptr(), (const bf16*)v.data_ptr(),
-                       (float*)logits.data_ptr(), M, A);
-    HIP_OK(hipGetLastError());
-    return logits;
-}
-
 #define MAX_L 1024
 
 // counter-based dropout hash: deterministic in (seed, salt, index), so the

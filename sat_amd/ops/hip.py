@@ -5,7 +5,6 @@ sat_amd/ops/csrc/*.hip, gfx950-only) provides:
 
   dense_fwd                       MFMA GEMM + fused bias/activation (bf16)
   lstm_pointwise_fwd / _bwd       fused LSTM gate math (i,j,f,o; recompute bwd)
-  attn_score_fwd                  scores GEMV temp[M,A]·v[A] (no N=1 GEMM)
   attn_pool_fwd                   LDS-staged softmax over L + ctx-weighted sum
   embedding_fwd / embedding_bwd   gather / scatter-add
   ce_fwd / ce_bwd                 fused masked softmax cross-entropy
