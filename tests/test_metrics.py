@@ -91,3 +91,21 @@ def test_full_eval_driver(tmp_path):
     assert scorer.eval['Bleu_1'] > 0.9  # echoing a GT caption scores high
     assert set(scorer.eval) == {'Bleu_1', 'Bleu_2', 'Bleu_3', 'Bleu_4',
                                 'METEOR', 'ROUGE_L', 'CIDEr'}
+
+
+def test_metrics_tolerate_empty_candidate():
+    gts = {1: ['a man riding a horse']}
+    res = {1: ['']}
+    scores, _ = Bleu(4).compute_score(gts, res)
+    assert all(0.0 <= s < 1e-6 for s in scores)
+    s, _ = Rouge().compute_score(gts, res)
+    assert s == 0.0
+    s, _ = Meteor().compute_score(gts, res)
+    assert s == 0.0
+    s, _ = Cider().compute_score(gts, res)
+    assert s == 0.0
+
+
+def test_bleu_single_word():
+    scores, _ = Bleu(4).compute_score({1: ['dog']}, {1: ['dog']})
+    assert scores[0] > 0.9  # unigram perfect; higher n-grams degenerate
