@@ -101,3 +101,12 @@ class Config(object):
         self.synthetic_data = False      # synthetic COCO-shaped data (no files)
         self.synthetic_num_images = 640
         self.seed = 1234
+
+        # Optional JSON overrides from the environment (testing / ops
+        # hook): SAT_CONFIG_OVERRIDES='{"batch_size": 2, ...}'
+        import json as _json
+        import os as _os
+        overrides = _os.environ.get('SAT_CONFIG_OVERRIDES')
+        if overrides:
+            for k, v in _json.loads(overrides).items():
+                setattr(self, k, v)
