@@ -64,6 +64,17 @@ class COCO(object):
     def getImgIds(self):
         return list(self.imgs.keys())
 
+    def getCatIds(self):
+        # caption datasets carry no categories (parity stub: the
+        # reference's vendored copy exposes these, coco.py:158-261)
+        return [c['id'] for c in self.dataset.get('categories', [])]
+
+    def loadCats(self, ids):
+        if not isinstance(ids, (list, tuple)):
+            ids = [ids]
+        cats = {c['id']: c for c in self.dataset.get('categories', [])}
+        return [cats[i] for i in ids]
+
     def loadAnns(self, ids):
         if not isinstance(ids, (list, tuple)):
             ids = [ids]

@@ -130,14 +130,17 @@ class Optimizer(object):
                 vhat = s['v'] / (1 - cfg.beta2 ** self.step_count)
                 upd = mhat / (vhat.sqrt() + cfg.epsilon)
             elif self.kind == 'RMSProp':
+                # TF RMSPropOptimizer: mom = m*mom + lr*g/sqrt(ms[-mg^2]+eps)
+                # and var -= mom (epsilon inside the sqrt)
                 s['ms'].mul_(cfg.decay).addcmul_(g, g, value=1 - cfg.decay)
                 denom = s['ms']
                 if cfg.centered:
                     s['mg'].mul_(cfg.decay).add_(g, alpha=1 - cfg.decay)
                     denom = denom - s['mg'] ** 2
                 s['mom'].mul_(cfg.momentum).add_(
-                    g / (denom.sqrt() + 1e-10))
-                upd = s['mom']
+                    lr * g / (denom + 1e-10).sqrt())
+                p.data.add_(-s['mom'].to(p.dtype))
+                continue
             elif self.kind == 'Momentum':
                 s['mom'].mul_(cfg.momentum).add_(g)
                 if cfg.use_nesterov:

@@ -79,6 +79,7 @@ def load_cnn(model, data_path, verbose=True):
     (or a flat {name: array} dict) into model.cnn, matching by scope name."""
     print('Loading the CNN from %s...' % data_path)
     data = np.load(data_path, allow_pickle=True, encoding='latin1').item()
+    data = {_translate_scope(k): v for k, v in data.items()}
     cnn_state = dict(model.cnn.named_parameters())
     cnn_state.update(dict(model.cnn.named_buffers()))
     count = 0
@@ -104,6 +105,26 @@ def load_cnn(model, data_path, verbose=True):
                     count += 1
     print('%d tensors loaded.' % count)
     return count
+
+
+_BRANCH = {'branch2a': 'conv_a', 'branch2b': 'conv_b',
+           'branch2c': 'conv_c', 'branch1': 'shortcut'}
+
+
+def _translate_scope(scope):
+    """Map Caffe-style ResNet scope names (res2a_branch2a, bn2a_branch1,
+    ...) onto this repo's module paths (res2a.conv_a, res2a.shortcut_bn).
+    VGG scopes (conv1_1, ...) pass through unchanged."""
+    for pre, kind in (('res', ''), ('bn', 'bn')):
+        if scope.startswith(pre) and '_branch' in scope:
+            block, branch = scope[len(pre):].split('_', 1)
+            if branch in _BRANCH:
+                leaf = _BRANCH[branch]
+                if kind == 'bn':
+                    leaf = ('shortcut_bn' if leaf == 'shortcut'
+                            else 'bn' + leaf[-2:])
+                return 'res%s.%s' % (block, leaf)
+    return scope
 
 
 def trim(model_file, out_file=None):
