@@ -119,3 +119,41 @@ def test_variable_summaries_written(tiny_config):
     recs = [_json.loads(l) for l in open(path)]
     assert any(any(k.endswith('/mean') for k in r) for r in recs)
     assert any('total_loss' in r for r in recs)
+
+
+def test_config_env_overrides(monkeypatch):
+    import json as _json
+    from config import Config
+    monkeypatch.setenv('SAT_CONFIG_OVERRIDES',
+                       _json.dumps({'batch_size': 7, 'cnn': 'resnet50'}))
+    cfg = Config()
+    assert cfg.batch_size == 7 and cfg.cnn == 'resnet50'
+
+
+def test_config_knob_names_match_reference():
+    """Every reference config.py knob must exist under the same name."""
+    from config import Config
+    cfg = Config()
+    knobs = [
+        'cnn', 'max_caption_length', 'dim_embedding', 'num_lstm_units',
+        'num_initalize_layers', 'dim_initalize_layer', 'num_attend_layers',
+        'dim_attend_layer', 'num_decode_layers', 'dim_decode_layer',
+        'fc_kernel_initializer_scale', 'fc_kernel_regularizer_scale',
+        'fc_activity_regularizer_scale', 'conv_kernel_regularizer_scale',
+        'conv_activity_regularizer_scale', 'fc_drop_rate',
+        'lstm_drop_rate', 'attention_loss_factor', 'num_epochs',
+        'batch_size', 'optimizer', 'initial_learning_rate',
+        'learning_rate_decay_factor', 'num_steps_per_decay',
+        'clip_gradients', 'momentum', 'use_nesterov', 'decay', 'centered',
+        'beta1', 'beta2', 'epsilon', 'path_to_local_logs',
+        'root_path_to_local_data', 'local_repo', 'cloud_user_repo',
+        'cloud_path_to_data', 'save_period', 'save_dir', 'summary_dir',
+        'max_train_ann_num', 'max_eval_ann_num', 'vocabulary_file',
+        'vocabulary_size', 'train_image_dir', 'train_caption_file',
+        'temp_annotation_file', 'temp_data_file', 'eval_image_dir',
+        'eval_caption_file', 'eval_result_dir', 'eval_result_file',
+        'save_eval_result_as_image', 'test_image_dir', 'test_result_dir',
+        'test_result_file',
+    ]
+    missing = [k for k in knobs if not hasattr(cfg, k)]
+    assert not missing, missing
