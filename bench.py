@@ -108,7 +108,6 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
-    n_gpus = world if world > 1 else (args.gpus if have_gpu else 0) or 1
     total_images = args.steps * cfg.batch_size * (world if world > 1 else 1)
     value = total_images / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
