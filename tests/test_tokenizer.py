@@ -26,3 +26,16 @@ def test_ptb_tokenizer_multiple_captions():
     tok = PTBTokenizer()
     out = tok.tokenize({1: ['a dog.', 'a cat!'], 2: ['a bus.']})
     assert out[1] == ['a dog', 'a cat'] and out[2] == ['a bus']
+
+
+def test_word_tokenize_numbers_and_unicode():
+    assert word_tokenize("2 dogs and 3.5 cats") == \
+        ['2', 'dogs', 'and', '3.5', 'cats']
+    # non-ASCII letters are dropped rather than crashing
+    out = word_tokenize("café naïve dog")
+    assert 'dog' in out
+
+
+def test_word_tokenize_empty():
+    assert word_tokenize("") == []
+    assert word_tokenize("   ") == []
