@@ -98,6 +98,7 @@ class Config(object):
         self.use_hip_kernels = True      # hand-written CDNA4 kernels on GPU
         self.use_hip_graph = True        # capture the train step in a hipGraph
         self.allreduce_bucket_mb = 16    # DP gradient bucket size (xGMI-tuned)
+        self.use_glds_conv = False       # glds conv routing (see ROADMAP #0)
         self.synthetic_data = False      # synthetic COCO-shaped data (no files)
         self.synthetic_num_images = 640
         self.seed = 1234

@@ -112,6 +112,8 @@ class Conv2d(tnn.Module):
         self.stride = stride
         self.kernel_size = kernel_size
         self.activation = activation
+        self._glds_conv = getattr(nn_policy.config, 'use_glds_conv',
+                                  False)
         self.weight = tnn.Parameter(
             torch.empty(out_ch, in_ch, kernel_size, kernel_size))
         nn_policy.init_conv_(self.weight)
@@ -141,12 +143,20 @@ class Conv2d(tnn.Module):
                     else torch.empty(0, dtype=x.dtype, device=x.device),
                     self.activation == 'relu')
                 return y
-        # frozen GPU path for Cin=64 3x3/s1 layers: our implicit-GEMM
-        # MFMA conv (measured faster than MIOpen there: 337 vs 407 us on
-        # conv1_2; MIOpen's asm igemm wins at Cin>=128 and keeps those)
+        # frozen GPU implicit-GEMM paths (per-shape winners,
+        # profiles/r01_conv_shapes.txt):
+        #  * Cin=64: register-staged MFMA conv (337 vs 407 us on conv1_2)
+        #  * Cin<=128 & Cout>=128: the XOR-swizzled glds variant also beats
+        #    MIOpen (120 vs 169 us on conv2_1) but landed after this
+        #    round's GPU validation budget — gated off by default; flip
+        #    `use_glds_conv` after one GPU pass (docs/ROADMAP.md #0)
         if (x.is_cuda and x.dtype == torch.bfloat16
                 and not torch.is_grad_enabled()
-                and k == 3 and st == 1 and w.shape[1] == 64
+                and k == 3 and st == 1
+                and (w.shape[1] == 64
+                     or (self._glds_conv and w.shape[1] == 128
+                         and w.shape[0] >= 128 and w.shape[1] % 64 == 0))
+                and w.shape[0] % 8 == 0
                 and x.is_contiguous(memory_format=torch.channels_last)):
             from ..ops import hip
             if hip.available():
@@ -156,11 +166,17 @@ class Conv2d(tnn.Module):
                     self._w_ohwi_ver = self.weight._version
                     self._w_ohwi = w.permute(0, 2, 3, 1).contiguous() \
                         .reshape(w.shape[0], -1)
-                return _C.conv_igemm_fwd(
-                    x, self._w_ohwi,
-                    b if b is not None else
-                    torch.empty(0, dtype=x.dtype, device=x.device),
-                    self.activation == 'relu')
+                eb = (b if b is not None else
+                      torch.empty(0, dtype=x.dtype, device=x.device))
+                relu = self.activation == 'relu'
+                if (self._glds_conv and w.shape[0] >= 128
+                        and w.shape[1] % 64 == 0):
+                    xp = _C.pad1_nhwc(x)
+                    return _C.conv_igemm_glds_fwd(
+                        xp, self._w_ohwi, eb, x.shape[2], x.shape[3],
+                        relu)
+                if w.shape[1] == 64:
+                    return _C.conv_igemm_fwd(x, self._w_ohwi, eb, relu)
         # frozen GPU path: fused NHWC bias+ReLU kernel after the MIOpen
         # conv instead of two separate eager elementwise passes
         fuse_epi = (x.is_cuda and x.dtype == torch.bfloat16
