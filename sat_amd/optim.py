@@ -99,10 +99,14 @@ class Optimizer(object):
                     # dense storage in any layout (channels_last conv
                     # grads included) is fine: p/g/m/v share the layout
                     # and the kernel iterates flat storage order
-                    assert g.dtype == torch.float32
-                    assert (g.is_contiguous()
+                    if g.dtype != torch.float32 or not (
+                            g.is_contiguous()
                             or g.is_contiguous(
-                                memory_format=torch.channels_last))
+                                memory_format=torch.channels_last)):
+                        raise RuntimeError(
+                            'fused Adam needs dense fp32 grads; got '
+                            '%s for a %s param' % (g.dtype,
+                                                   tuple(p0.shape)))
             gsq = _C.sq_norm_mt(self._mt_desc, self._mt_cum,
                                 len(self.params), self._mt_total)
             _C.adam_step_mt(
