@@ -97,8 +97,17 @@ class Config(object):
         self.compute_dtype = 'bf16'      # 'bf16' | 'fp32' — CNN/decoder compute
         self.use_hip_kernels = True      # hand-written CDNA4 kernels on GPU
         self.use_hip_graph = True        # capture the train step in a hipGraph
+        self.use_bptt = True             # fused hand-written BPTT over the
+        #                                  T-step decoder (GPU bf16, default
+        #                                  architecture only; False = per-op
+        #                                  autograd loop, same numerics)
         self.allreduce_bucket_mb = 16    # DP gradient bucket size (xGMI-tuned)
-        self.use_glds_conv = False       # glds conv routing (see ROADMAP #0)
+        self.overlap_allreduce = True    # overlap DP grad all-reduce with the
+        #                                  next step's frozen-CNN forward
+        self.use_glds_conv = True        # LDS-staged (glds) implicit-GEMM conv
+        #                                  kernels for Cin<=128/Cout>=128 3x3
+        #                                  layers (beats MIOpen per
+        #                                  profiles/r01_conv_shapes.txt)
         self.synthetic_data = False      # synthetic COCO-shaped data (no files)
         self.synthetic_num_images = 640
         self.seed = 1234

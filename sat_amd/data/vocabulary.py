@@ -69,16 +69,29 @@ class Vocabulary(object):
                 if w in self.word2idx]
 
     def get_sentence(self, idxs):
-        """Translate a vector of indices back into a sentence string."""
-        words = [self.words[int(i)] for i in idxs]
-        if not words or words[-1] != '.':
-            words.append('.')
-        length = int(np.argmax(np.array(words) == '.')) + 1
-        words = words[:length]
-        sentence = "".join(
-            [" " + w if not w.startswith("'") and w not in string.punctuation
-             else w for w in words]).strip()
-        return sentence
+        """Translate a vector of indices back into a sentence string.
+
+        Semantics (parity contract with reference vocabulary.py:53-63,
+        verified by the reference-CSV round-trip test): the caption stops
+        at the first '.' token inclusive — one is supplied if the indices
+        never produce it — and tokens are joined with a single space,
+        except that punctuation and apostrophe-led clitics ("'s", "n't")
+        attach directly to the preceding token.
+        """
+        pieces = []
+        terminated = False
+        for i in idxs:
+            w = self.words[int(i)]
+            attaches = w.startswith("'") or w in string.punctuation
+            if pieces and not attaches:
+                pieces.append(' ')
+            pieces.append(w)
+            if w == '.':
+                terminated = True
+                break
+        if not terminated:
+            pieces.append('.')
+        return ''.join(pieces)
 
     def save(self, save_file):
         """Save the vocabulary as a CSV (word, index, frequency)."""

@@ -78,9 +78,6 @@ class GraphedTrainStep(object):
 
     def _capture(self, images, sentences, masks):
         self.static_in = (images.clone(), sentences.clone(), masks.clone())
-        # warmup/capture run real optimizer steps; snapshot + restore so
-        # training numerics are unaffected by graph initialization.
-        snap = self._snapshot()
         side = torch.cuda.Stream()
         side.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(side):
@@ -93,19 +90,46 @@ class GraphedTrainStep(object):
         with torch.cuda.graph(self.graph):
             self.static_out = self._inner(*self.static_in)
         torch.cuda.synchronize()
-        self._restore(snap)
+
+    def _all_ranks_ok(self, ok):
+        """Agree across ranks on the capture outcome so no rank replays a
+        graph while another runs eager (their per-step collective counts
+        would then diverge and the job would hang)."""
+        import torch.distributed as dist
+        if not (dist.is_available() and dist.is_initialized()
+                and dist.get_world_size() > 1):
+            return ok
+        dev = 'cpu' if dist.get_backend() == 'gloo' else 'cuda'
+        flag = torch.tensor([1.0 if ok else 0.0], device=dev)
+        dist.all_reduce(flag, op=dist.ReduceOp.MIN)
+        return bool(flag.item() >= 0.5)
 
     def step(self, images, sentences, masks):
         if self.failed:
             return self._inner(images, sentences, masks)
         if self.graph is None:
+            # warmup/capture run real optimizer steps on the first batch;
+            # snapshot HERE (not inside _capture) so a partial capture
+            # failure still restores params/Adam state/step count/RNG
+            # before the eager fallback — otherwise the warmup steps'
+            # duplicate updates would be silently kept.
+            snap = self._snapshot()
+            ok = True
             try:
                 self._capture(images, sentences, masks)
             except Exception as e:
                 print('[sat_amd] hipGraph capture failed (%r); '
                       'falling back to eager steps' % (e,))
+                ok = False
+            if not self._all_ranks_ok(ok):
+                if ok:
+                    print('[sat_amd] hipGraph capture failed on a peer '
+                          'rank; all ranks falling back to eager steps')
                 self.failed = True
-                torch.cuda.synchronize()
+                self.graph = None
+            torch.cuda.synchronize()
+            self._restore(snap)
+            if self.failed:
                 return self._inner(images, sentences, masks)
         si, ss, sm = self.static_in
         si.copy_(images, non_blocking=True)

@@ -65,6 +65,10 @@ class CaptionGenerator(tnn.Module):
                 and contexts.dtype == torch.bfloat16
                 and getattr(cfg, 'use_hip_kernels', True)
                 and getattr(cfg, 'use_bptt', True)
+                # activity regularizers need per-layer activations in the
+                # autograd graph — the fused BPTT doesn't expose them
+                and getattr(cfg, 'fc_activity_regularizer_scale', 0.0)
+                == 0.0
                 and cfg.num_attend_layers == 2
                 and cfg.num_decode_layers == 2
                 # kernel shape contracts (fall back to the per-op loop

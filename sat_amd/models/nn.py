@@ -37,8 +37,19 @@ class NN(object):
             if self.train_cnn else 0.0
         self.fc_drop_rate = config.fc_drop_rate
         self.lstm_drop_rate = config.lstm_drop_rate
+        # L1 activity regularizers (reference nn.py:23-27,39-43): applied
+        # to the output of every activation-carrying fc/conv layer, gated
+        # on is_train (fc) / train_cnn (conv) exactly like the kernels.
+        self.fc_act_reg = getattr(
+            config, 'fc_activity_regularizer_scale', 0.0) \
+            if self.is_train else 0.0
+        self.conv_act_reg = getattr(
+            config, 'conv_activity_regularizer_scale', 0.0) \
+            if self.train_cnn else 0.0
         # (param, scale) pairs for reg_loss; filled as layers are built
         self._regularized = []
+        # per-forward activity-loss terms (cleared by reg_loss)
+        self._act_losses = []
 
     def init_fc_(self, t):
         tnn.init.uniform_(t, -self.fc_scale, self.fc_scale)
@@ -54,12 +65,24 @@ class NN(object):
         if self.conv_reg > 0:
             self._regularized.append((p, self.conv_reg))
 
+    def add_activity_loss(self, y, scale):
+        """Record scale · Σ|y| for one layer output (TF l1_regularizer
+        as activity_regularizer, reference nn.py:23-27,39-43)."""
+        if scale > 0 and torch.is_grad_enabled():
+            self._act_losses.append(scale * y.float().abs().sum())
+
     def reg_loss(self):
-        """Σ scale · l2_loss(w) with l2_loss = sum(w²)/2 (TF semantics)."""
-        if not self._regularized:
-            return torch.zeros(())
-        return sum(s * 0.5 * (p.float() ** 2).sum()
-                   for p, s in self._regularized)
+        """Σ scale · l2_loss(w) (l2_loss = sum(w²)/2, TF semantics) plus
+        any activity-regularizer terms recorded this forward."""
+        total = torch.zeros(())
+        if self._regularized:
+            total = sum(s * 0.5 * (p.float() ** 2).sum()
+                        for p, s in self._regularized)
+        if self._act_losses:
+            act = sum(self._act_losses)
+            self._act_losses = []
+            total = total.to(act.device) + act
+        return total
 
     def dropout(self, x):
         return ops.dropout(x, self.fc_drop_rate, self.is_train)
@@ -73,6 +96,7 @@ class Dense(tnn.Module):
                  use_bias=True):
         super().__init__()
         self.activation = activation
+        self._nn = nn_policy
         self.weight = tnn.Parameter(torch.empty(units, in_dim))
         nn_policy.init_fc_(self.weight)
         nn_policy.register_fc_kernel(self.weight)
@@ -96,10 +120,16 @@ class Dense(tnn.Module):
 
     def forward(self, x):
         if self._wc is not None and self._wc.dtype == x.dtype:
-            return ops.dense(x, self._wc, self._bc, self.activation)
-        w = self.weight.to(x.dtype)
-        b = self.bias.to(x.dtype) if self.bias is not None else None
-        return ops.dense(x, w, b, self.activation)
+            y = ops.dense(x, self._wc, self._bc, self.activation)
+        else:
+            w = self.weight.to(x.dtype)
+            b = self.bias.to(x.dtype) if self.bias is not None else None
+            y = ops.dense(x, w, b, self.activation)
+        # activity regularizer only on activation-carrying layers
+        # (reference nn.py:92-95)
+        if self.activation is not None and self._nn.fc_act_reg > 0:
+            self._nn.add_activity_loss(y, self._nn.fc_act_reg)
+        return y
 
 
 class Conv2d(tnn.Module):
@@ -112,8 +142,9 @@ class Conv2d(tnn.Module):
         self.stride = stride
         self.kernel_size = kernel_size
         self.activation = activation
+        self._nn = nn_policy
         self._glds_conv = getattr(nn_policy.config, 'use_glds_conv',
-                                  False)
+                                  True)
         self.weight = tnn.Parameter(
             torch.empty(out_ch, in_ch, kernel_size, kernel_size))
         nn_policy.init_conv_(self.weight)
@@ -205,6 +236,11 @@ class Conv2d(tnn.Module):
             y = y + b.reshape(1, -1, 1, 1)
         if self.activation == 'relu':
             y = torch.relu(y)
+        # conv activity regularizer (reference nn.py:54-57): only on
+        # activation-carrying layers, only when the CNN trains — which is
+        # exactly when this autograd path (not the inference kernels) runs
+        if self.activation is not None and self._nn.conv_act_reg > 0:
+            self._nn.add_activity_loss(y, self._nn.conv_act_reg)
         return y
 
 
@@ -271,8 +307,13 @@ class BatchNorm(tnn.Module):
             from ..ops import hip
             if hip.available():
                 from sat_amd import _C
+                # all four stat/affine tensors key the cache: a partial
+                # load_cnn/checkpoint match may in-place update only
+                # running_mean or bias, which must invalidate too
                 ver = (self.bn.running_var._version,
-                       self.bn.weight._version)
+                       self.bn.running_mean._version,
+                       self.bn.weight._version,
+                       self.bn.bias._version)
                 if (self._ss is None or self._ss[0].device != x.device
                         or getattr(self, '_ss_ver', None) != ver):
                     self._ss_ver = ver

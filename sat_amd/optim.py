@@ -142,7 +142,7 @@ class Optimizer(object):
                     s['mg'].mul_(cfg.decay).add_(g, alpha=1 - cfg.decay)
                     denom = denom - s['mg'] ** 2
                 s['mom'].mul_(cfg.momentum).add_(
-                    lr * g / (denom + 1e-10).sqrt())
+                    lr * g / (denom + cfg.epsilon).sqrt())
                 p.data.add_(-s['mom'].to(p.dtype))
                 continue
             elif self.kind == 'Momentum':

@@ -21,7 +21,8 @@ import torch.distributed as dist
 
 
 class _Bucket:
-    __slots__ = ('params', 'numel', 'flat', 'offsets', 'pending', 'work')
+    __slots__ = ('params', 'numel', 'flat', 'offsets', 'pending', 'work',
+                 'unfilled')
 
     def __init__(self):
         self.params = []
@@ -30,6 +31,7 @@ class _Bucket:
         self.offsets = {}
         self.pending = 0
         self.work = None
+        self.unfilled = set()
 
 
 class DataParallelGrads(object):
@@ -66,6 +68,7 @@ class DataParallelGrads(object):
         for b in self.buckets:
             b.pending = len(b.params)
             b.work = None
+            b.unfilled = set(b.params)
 
     def _hook(self, p):
         b = self._param_bucket[p]
@@ -75,6 +78,7 @@ class DataParallelGrads(object):
         off = b.offsets[p]
         b.flat[off:off + p.numel()].copy_(
             p.grad.detach().reshape(-1).float())
+        b.unfilled.discard(p)
         b.pending -= 1
         if b.pending == 0:
             b.flat.div_(self.world)
@@ -85,12 +89,19 @@ class DataParallelGrads(object):
         """Wait for in-flight buckets and write averaged grads back."""
         for b in self.buckets:
             if b.pending != 0:
-                # gradient never materialized for some params (e.g. unused);
-                # reduce what we have so ranks stay in lockstep.
+                # Gradients never materialized for some params this step
+                # (e.g. conditionally-unused layers); reduce the bucket
+                # anyway so ranks stay in collective lockstep — but ZERO
+                # the unfilled slots first: `flat` still holds the
+                # previous step's reduced values for them (ghost grads).
                 if b.flat is None:
                     b.flat = torch.zeros(
                         b.numel, dtype=torch.float32,
                         device=next(iter(b.offsets)).device)
+                else:
+                    for p in b.unfilled:
+                        off = b.offsets[p]
+                        b.flat[off:off + p.numel()].zero_()
                 b.flat.div_(self.world)
                 b.work = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
                                          group=self.group, async_op=True)
@@ -98,6 +109,14 @@ class DataParallelGrads(object):
             if b.work is not None:
                 b.work.wait()
             for p in b.params:
+                # A param whose hook never fired gets the cross-rank
+                # average written back only if it had a grad to begin
+                # with (another rank may have used the layer); params
+                # with no grad anywhere stay grad-less so the optimizer
+                # sees them as untouched instead of decaying momentum
+                # with fabricated zeros.
+                if p in b.unfilled and p.grad is None:
+                    continue
                 off = b.offsets[p]
                 g = b.flat[off:off + p.numel()].reshape(p.shape)
                 if p.grad is None:

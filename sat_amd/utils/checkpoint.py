@@ -71,6 +71,19 @@ def load(model, optimizer, config, model_file=None):
         optimizer.load_state_arrays(arrays)
     global_step = int(float(arrays.get('global_step', 0)))
     print('%d tensors loaded.' % count)
+    # Name-matching silently skips misses (reference base_model.py:272-278
+    # semantics) — but a checkpoint from a DIFFERENT naming scheme (e.g. a
+    # reference TF .npy with names like 'word_embedding:0') matches zero
+    # tensors and would otherwise be indistinguishable from a good load.
+    # Such files need a name-translation pass first (tools/, docs/).
+    if count == 0 and len(state) > 0:
+        raise RuntimeError(
+            'checkpoint %s matched 0 of %d model tensors — wrong file or '
+            'incompatible variable naming (reference TF checkpoints need '
+            'name translation before loading)' % (path, len(state)))
+    if count < len(state) // 2:
+        print('WARNING: only %d of %d model tensors matched %s — partial '
+              'restore' % (count, len(state), path))
     return global_step
 
 

@@ -190,3 +190,59 @@ def test_single_layer_mlp_variants(tiny_config):
     out['total_loss'].backward()
     assert all(torch.isfinite(p.grad).all()
                for p in model.decoder.parameters() if p.grad is not None)
+
+
+def test_activity_regularizers_apply():
+    """fc/conv activity regularizers (reference nn.py:23-27,39-43) are
+    L1-of-activations terms in reg_loss — not silent dead knobs."""
+    import torch
+    from config import Config
+    from sat_amd.models.nn import NN, Dense
+
+    cfg = Config()
+    cfg.phase = 'train'
+    cfg.train_cnn = False
+    cfg.fc_activity_regularizer_scale = 0.01
+    pol = NN(cfg)
+    d = Dense(pol, 4, 3, 'tanh')
+    x = torch.randn(2, 4)
+    y = d(x)
+    expected = cfg.fc_kernel_regularizer_scale * 0.5 \
+        * float((d.weight ** 2).sum()) \
+        + 0.01 * float(y.abs().sum())
+    got = float(pol.reg_loss())
+    assert abs(got - expected) < 1e-5
+    # cleared after reg_loss
+    assert pol._act_losses == []
+    # activation-free layers carry no activity term (reference nn.py:92-95)
+    d2 = Dense(pol, 4, 3, None)
+    d2(x)
+    assert pol._act_losses == []
+
+
+def test_activity_regularizer_reaches_total_loss():
+    import torch
+    from config import Config
+    from sat_amd.models.caption_generator import CaptionGenerator
+
+    def tiny(scale):
+        cfg = Config()
+        cfg.phase = 'train'
+        cfg.train_cnn = False
+        cfg.fc_activity_regularizer_scale = scale
+        cfg.vocabulary_size = 30
+        cfg.dim_embedding = 16
+        cfg.num_lstm_units = 16
+        cfg.dim_initalize_layer = 16
+        cfg.dim_attend_layer = 16
+        cfg.dim_decode_layer = 16
+        cfg.max_caption_length = 4
+        torch.manual_seed(0)
+        m = CaptionGenerator(cfg)
+        torch.manual_seed(1)
+        img = torch.randn(1, 3, 224, 224)
+        s = torch.randint(0, 30, (1, 4))
+        mk = torch.ones(1, 4)
+        return float(m(img, s, mk)['reg_loss'])
+
+    assert tiny(0.05) > tiny(0.0)

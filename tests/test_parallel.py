@@ -111,3 +111,52 @@ def test_shard_dataset():
     assert a.count == 5 and b.count == 5
     assert set(a.image_ids) | set(b.image_ids) == set(ids)
     assert set(a.image_ids).isdisjoint(set(b.image_ids))
+
+
+def _worker_stale_bucket(rank, world, tmpfile, q):
+    dist.init_process_group(
+        'gloo', init_method='file://%s' % tmpfile,
+        rank=rank, world_size=world)
+    try:
+        class Two(torch.nn.Module):
+            def __init__(self):
+                super().__init__()
+                self.a = torch.nn.Linear(8, 8)
+                self.b = torch.nn.Linear(8, 8)
+
+            def forward(self, x, use_b):
+                y = self.a(x)
+                return self.b(y) if use_b else y
+
+        torch.manual_seed(7)
+        model = Two()
+        from sat_amd.parallel.ddp import DataParallelGrads
+        ddp = DataParallelGrads(model, bucket_mb=64)  # one bucket
+
+        x = torch.randn(4, 8)
+        # step 1: both layers used -> flat holds b's grads afterwards
+        model(x, True).pow(2).mean().backward()
+        ddp.finish_backward()
+        # step 2: engine-style in-place zero, then b is NOT used
+        for p in model.parameters():
+            p.grad.zero_()
+        model(x, False).pow(2).mean().backward()
+        ddp.finish_backward()
+        # the stale-flat bug would resurrect step 1's b-grads here
+        ghost = float(model.b.weight.grad.abs().max())
+        q.put(('ok', rank, ghost == 0.0))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_partial_bucket_has_no_ghost_gradients(tmp_path):
+    """A bucket whose params were used last step but not this step must
+    all-reduce zeros for those slots, not the previous step's values
+    (VERDICT round-1 Weak #3)."""
+    world = 2
+    q = mp.get_context('spawn').Queue()
+    f = str(tmp_path / 'init3')
+    mp.spawn(_worker_stale_bucket, args=(world, f, q), nprocs=world,
+             join=True)
+    results = [q.get() for _ in range(world)]
+    assert all(r[0] == 'ok' and r[2] for r in results)
