@@ -102,8 +102,6 @@ class Config(object):
         #                                  architecture only; False = per-op
         #                                  autograd loop, same numerics)
         self.allreduce_bucket_mb = 16    # DP gradient bucket size (xGMI-tuned)
-        self.overlap_allreduce = True    # overlap DP grad all-reduce with the
-        #                                  next step's frozen-CNN forward
         self.use_glds_conv = True        # LDS-staged (glds) implicit-GEMM conv
         #                                  kernels for Cin<=128/Cout>=128 3x3
         #                                  layers (beats MIOpen per

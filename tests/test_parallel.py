@@ -160,3 +160,71 @@ def test_partial_bucket_has_no_ghost_gradients(tmp_path):
              join=True)
     results = [q.get() for _ in range(world)]
     assert all(r[0] == 'ok' and r[2] for r in results)
+
+
+def test_split_function_grads_fire_before_upstream_backward():
+    """The BPTT split (sat_amd/models/bptt.py) relies on AccumulateGrad's
+    top autograd priority: a downstream Function that takes leaf params
+    directly gets their grads accumulated (and DDP hooks fired) BEFORE
+    the upstream Function's backward runs — that window is where the
+    bucket all-reduce overlaps the recurrent reverse loop."""
+    order = []
+
+    class Core(torch.autograd.Function):
+        @staticmethod
+        def forward(ctx, x):
+            return x * 2
+
+        @staticmethod
+        def backward(ctx, g):
+            order.append('core-bwd')
+            return g * 2
+
+    class Head(torch.autograd.Function):
+        @staticmethod
+        def forward(ctx, x, w):
+            ctx.save_for_backward(w)
+            return x * w.sum()
+
+        @staticmethod
+        def backward(ctx, g):
+            order.append('head-bwd')
+            (w,) = ctx.saved_tensors
+            return g * w.sum(), g.sum() * torch.ones_like(w)
+
+    x = torch.randn(3, requires_grad=True)
+    w = torch.randn(4, requires_grad=True)
+    w.register_post_accumulate_grad_hook(
+        lambda p: order.append('w-hook'))
+    Head.apply(Core.apply(x), w).sum().backward()
+    assert order == ['head-bwd', 'w-hook', 'core-bwd'], order
+
+
+def test_bucket_phase_segregation():
+    """Decode-head params must never share a bucket with recurrent-core
+    params, so the head buckets launch without waiting for core grads."""
+    import torch.nn as tnn
+
+    class M(tnn.Module):
+        def __init__(self):
+            super().__init__()
+            self.embedding = tnn.Linear(4, 4)
+            self.dec_fc_1 = tnn.Linear(4, 4)
+            self.dec_fc_2 = tnn.Linear(4, 4)
+            self.att_fc_1a = tnn.Linear(4, 4)
+
+    if not dist.is_initialized():
+        dist.init_process_group(
+            'gloo', init_method='tcp://127.0.0.1:29511',
+            rank=0, world_size=1)
+    try:
+        from sat_amd.parallel.ddp import DataParallelGrads
+        m = M()
+        ddp = DataParallelGrads(m, bucket_mb=1024)  # size never splits
+        names = {p: n for n, p in m.named_parameters()}
+        for b in ddp.buckets:
+            kinds = {('head' if 'dec_fc' in names[p] else 'core')
+                     for p in b.params}
+            assert len(kinds) == 1, [names[p] for p in b.params]
+    finally:
+        dist.destroy_process_group()
