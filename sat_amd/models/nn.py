@@ -175,38 +175,49 @@ class Conv2d(tnn.Module):
                     self.activation == 'relu')
                 return y
         # frozen GPU implicit-GEMM paths (per-shape winners,
-        # profiles/r01_conv_shapes.txt):
+        # profiles/r01_conv_shapes.txt + r02 8-phase kernel):
+        #  * Cout%256==0 with enough output pixels to fill the chip at a
+        #    256^2 tile: the 8-phase deep-pipelined igemm (conv8p.hip) —
+        #    the Cin>=256 layers MIOpen used to keep;
         #  * Cin=64: register-staged MFMA conv (337 vs 407 us on conv1_2)
-        #  * Cin<=128 & Cout>=128: the XOR-swizzled glds variant also beats
-        #    MIOpen (120 vs 169 us on conv2_1) but landed after this
-        #    round's GPU validation budget — gated off by default; flip
-        #    `use_glds_conv` after one GPU pass (docs/ROADMAP.md #0)
+        #  * Cin<=128 & Cout>=128: the XOR-swizzled glds variant
+        #    (120 vs 169 us on conv2_1); `use_glds_conv` gates both
+        #    LDS-staged variants.
         if (x.is_cuda and x.dtype == torch.bfloat16
                 and not torch.is_grad_enabled()
                 and k == 3 and st == 1
-                and (w.shape[1] == 64
-                     or (self._glds_conv and w.shape[1] == 128
-                         and w.shape[0] >= 128 and w.shape[1] % 64 == 0))
-                and w.shape[0] % 8 == 0
+                and w.shape[0] % 8 == 0 and w.shape[1] % 8 == 0
                 and x.is_contiguous(memory_format=torch.channels_last)):
             from ..ops import hip
             if hip.available():
                 from sat_amd import _C
-                if getattr(self, '_w_ohwi', None) is None or \
-                        self._w_ohwi_ver != self.weight._version:
-                    self._w_ohwi_ver = self.weight._version
-                    self._w_ohwi = w.permute(0, 2, 3, 1).contiguous() \
-                        .reshape(w.shape[0], -1)
-                eb = (b if b is not None else
-                      torch.empty(0, dtype=x.dtype, device=x.device))
+                Cout, Cin = w.shape[0], w.shape[1]
+                M = x.shape[0] * x.shape[2] * x.shape[3]
                 relu = self.activation == 'relu'
-                if (self._glds_conv and w.shape[0] >= 128
-                        and w.shape[1] % 64 == 0):
-                    xp = _C.pad1_nhwc(x)
-                    return _C.conv_igemm_glds_fwd(
-                        xp, self._w_ohwi, eb, x.shape[2], x.shape[3],
-                        relu)
-                if w.shape[1] == 64:
+                use_8p = (self._glds_conv and Cout % 256 == 0
+                          and Cin % 64 == 0 and M >= 12544)
+                use_glds = (self._glds_conv and not use_8p
+                            and Cout >= 128 and Cout % 8 == 0
+                            and Cin % 64 == 0 and Cin <= 128)
+                use_ig64 = (not use_8p and not use_glds and Cin == 64)
+                if use_8p or use_glds or use_ig64:
+                    if getattr(self, '_w_ohwi', None) is None or \
+                            self._w_ohwi_ver != self.weight._version:
+                        self._w_ohwi_ver = self.weight._version
+                        self._w_ohwi = w.permute(0, 2, 3, 1).contiguous() \
+                            .reshape(w.shape[0], -1)
+                    eb = (b if b is not None else
+                          torch.empty(0, dtype=x.dtype, device=x.device))
+                    if use_8p:
+                        xp = _C.pad1_nhwc(x)
+                        return _C.conv_igemm_8p_fwd(
+                            xp, self._w_ohwi, eb, x.shape[2], x.shape[3],
+                            relu)
+                    if use_glds:
+                        xp = _C.pad1_nhwc(x)
+                        return _C.conv_igemm_glds_fwd(
+                            xp, self._w_ohwi, eb, x.shape[2], x.shape[3],
+                            relu)
                     return _C.conv_igemm_fwd(x, self._w_ohwi, eb, relu)
         # frozen GPU path: fused NHWC bias+ReLU kernel after the MIOpen
         # conv instead of two separate eager elementwise passes

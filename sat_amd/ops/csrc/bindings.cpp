@@ -12,6 +12,9 @@ at::Tensor maxpool2x2_nhwc(at::Tensor input);
 at::Tensor conv_igemm_fwd(at::Tensor input, at::Tensor w_ohwi,
                           at::Tensor bias, bool relu);
 at::Tensor pad1_nhwc(at::Tensor input);
+at::Tensor conv_igemm_8p_fwd(at::Tensor padded, at::Tensor w_ohwi,
+                             at::Tensor bias, int64_t Hh, int64_t Ww,
+                             bool relu);
 at::Tensor conv_igemm_glds_fwd(at::Tensor padded, at::Tensor w_ohwi,
                                at::Tensor bias, int64_t Hh, int64_t Ww,
                                bool relu);
@@ -102,6 +105,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("conv_igemm_fwd", &conv_igemm_fwd);
     m.def("pad1_nhwc", &pad1_nhwc);
     m.def("conv_igemm_glds_fwd", &conv_igemm_glds_fwd);
+    m.def("conv_igemm_8p_fwd", &conv_igemm_8p_fwd,
+          "8-phase deep-pipelined implicit-GEMM conv (Cout%256==0)");
     m.def("dense_lstm_fwd", &dense_lstm_fwd);
     m.def("dense_drop_fwd", &dense_drop_fwd);
     m.def("conv3_fwd", &conv3_fwd,

@@ -31,6 +31,7 @@ setup(
                 os.path.join(CSRC, 'kernels.hip'),
                 os.path.join(CSRC, 'bptt_fuse.hip'),
                 os.path.join(CSRC, 'conv3.hip'),
+                os.path.join(CSRC, 'conv8p.hip'),
             ],
             extra_compile_args={
                 'cxx': ['-O3', '-std=c++17'],

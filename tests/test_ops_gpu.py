@@ -352,3 +352,43 @@ def test_conv_igemm_glds_matches_torch():
         ref = torch.relu(torch.nn.functional.conv2d(
             x.float(), w.float(), b.float(), padding=1))
         assert _rel_err(y, ref) < 2e-2, (Cin, Cout, H, W)
+
+
+def test_conv_igemm_8p_matches_torch():
+    """8-phase pipelined igemm (conv8p.hip) vs fp32 torch conv.  Shapes
+    cover: M % 256 != 0 (edge-clamp rows), Cout 256 and 512, Cin 64..512,
+    and non-square H/W."""
+    from sat_amd import _C
+    torch.manual_seed(11)
+    for Cin, Cout, B, H, W in [(64, 256, 2, 17, 19), (128, 256, 3, 16, 16),
+                               (256, 512, 2, 14, 14), (512, 512, 1, 28, 28)]:
+        x = torch.randn(B, Cin, H, W).to(DEV, torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+        w = (torch.randn(Cout, Cin, 3, 3) * 0.05).to(DEV, torch.bfloat16)
+        b = torch.randn(Cout).to(DEV, torch.bfloat16)
+        w_ohwi = w.permute(0, 2, 3, 1).contiguous().reshape(Cout, -1)
+        xp = _C.pad1_nhwc(x)
+        y = _C.conv_igemm_8p_fwd(xp, w_ohwi, b, H, W, True)
+        ref = torch.relu(torch.nn.functional.conv2d(
+            x.float(), w.float(), b.float(), padding=1))
+        assert _rel_err(y, ref) < 2e-2, (Cin, Cout, B, H, W)
+
+
+def test_conv_igemm_8p_race_screen():
+    """Sync-structure discipline (CDNA4 guide §5.4): a new pipelined
+    schedule needs a multi-run race screen — same inputs, repeated
+    launches, every run must agree with the fp32 reference."""
+    from sat_amd import _C
+    torch.manual_seed(12)
+    Cin, Cout, B, H, W = (256, 256, 2, 23, 29)
+    x = torch.randn(B, Cin, H, W).to(DEV, torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    w = (torch.randn(Cout, Cin, 3, 3) * 0.05).to(DEV, torch.bfloat16)
+    b = torch.randn(Cout).to(DEV, torch.bfloat16)
+    w_ohwi = w.permute(0, 2, 3, 1).contiguous().reshape(Cout, -1)
+    ref = torch.relu(torch.nn.functional.conv2d(
+        x.float(), w.float(), b.float(), padding=1))
+    xp = _C.pad1_nhwc(x)
+    for run in range(6):
+        y = _C.conv_igemm_8p_fwd(xp, w_ohwi, b, H, W, True)
+        assert _rel_err(y, ref) < 2e-2, 'race screen run %d' % run
