@@ -57,6 +57,8 @@ def main():
     p.add_argument('--batch', type=int, default=32)
     p.add_argument('--cnn', default='vgg16',
                    choices=['vgg16', 'resnet50'])
+    p.add_argument('--train-cnn', action='store_true',
+                   help='BASELINE config #4 shape: end-to-end CNN+RNN')
     p.add_argument('--no-hip-graph', action='store_true')
     args = p.parse_args()
 
@@ -69,7 +71,7 @@ def main():
 
     cfg = Config()
     cfg.phase = 'train'
-    cfg.train_cnn = False
+    cfg.train_cnn = args.train_cnn
     cfg.cnn = args.cnn
     cfg.synthetic_data = True
     cfg.batch_size = args.batch
@@ -127,7 +129,9 @@ def main():
             "dtype": "bf16" if have_gpu else "fp32",
             "data": "synthetic",
             "config": {
-                "model": "%s_frozen+attention_lstm512" % cfg.cnn,
+                "model": "%s_%s+attention_lstm512"
+                % (cfg.cnn,
+                   'train_cnn' if cfg.train_cnn else 'frozen'),
                 "global_batch": cfg.batch_size *
                 (world if world > 1 else 1),
                 "seq_len": cfg.max_caption_length,

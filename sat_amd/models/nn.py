@@ -158,6 +158,20 @@ class Conv2d(tnn.Module):
         w = self.weight.to(x.dtype)
         b = self.bias.to(x.dtype) if self.bias is not None else None
         k, st = self.kernel_size, self.stride
+        # --train_cnn: hand-written forward/dgrad/wgrad autograd triple
+        # (reference model.py:505-511 backward surface on in-tree kernels)
+        if (torch.is_grad_enabled()
+                and (x.requires_grad or w.requires_grad)
+                and self._glds_conv and k == 3 and st == 1
+                and self.activation in ('relu', None)):
+            from ..ops.convgrad import Conv3x3Train, conv3x3_train_ok
+            if conv3x3_train_ok(x, w):
+                y = Conv3x3Train.apply(x, w, b,
+                                       self.activation == 'relu')
+                if self.activation is not None \
+                        and self._nn.conv_act_reg > 0:
+                    self._nn.add_activity_loss(y, self._nn.conv_act_reg)
+                return y
         if (x.is_cuda and x.dtype == torch.bfloat16
                 and not torch.is_grad_enabled()
                 and k == 3 and st == 1 and w.shape[1] == 3

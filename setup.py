@@ -32,6 +32,7 @@ setup(
                 os.path.join(CSRC, 'bptt_fuse.hip'),
                 os.path.join(CSRC, 'conv3.hip'),
                 os.path.join(CSRC, 'conv8p.hip'),
+                os.path.join(CSRC, 'conv_bwd.hip'),
             ],
             extra_compile_args={
                 'cxx': ['-O3', '-std=c++17'],

@@ -392,3 +392,62 @@ def test_conv_igemm_8p_race_screen():
     for run in range(6):
         y = _C.conv_igemm_8p_fwd(xp, w_ohwi, b, H, W, True)
         assert _rel_err(y, ref) < 2e-2, 'race screen run %d' % run
+
+
+def test_conv3x3_wgrad_matches_torch():
+    from sat_amd import _C
+    torch.manual_seed(13)
+    for Cin, Cout, B, H, W in [(64, 64, 2, 14, 14), (128, 256, 2, 9, 11),
+                               (64, 128, 3, 16, 16)]:
+        x = torch.randn(B, Cin, H, W).to(DEV, torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+        dy = torch.randn(B, Cout, H, W).to(DEV, torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+        xpad = _C.pad1_nhwc(x)
+        dy_rows = dy.permute(0, 2, 3, 1).reshape(B * H * W, Cout)
+        dwf = _C.conv3x3_wgrad(xpad, dy_rows, H, W)
+        got = dwf.reshape(Cout, 3, 3, Cin).permute(0, 3, 1, 2)
+        xr = x.float().detach().requires_grad_(True)
+        wr = torch.zeros(Cout, Cin, 3, 3, device=DEV,
+                         requires_grad=True)
+        torch.nn.functional.conv2d(xr, wr, padding=1) \
+            .backward(dy.float())
+        assert _rel_err(got, wr.grad) < 2e-2, (Cin, Cout, B, H, W)
+
+
+def test_conv3x3_train_function_grads():
+    """Full autograd triple (fwd + dgrad + wgrad + dbias + fused ReLU)
+    vs fp32 torch reference."""
+    from sat_amd.ops.convgrad import Conv3x3Train
+    torch.manual_seed(14)
+    for Cin, Cout, B, H, W, relu in [(64, 128, 2, 14, 14, True),
+                                     (128, 256, 2, 16, 16, True),
+                                     (256, 512, 1, 14, 14, False)]:
+        x0 = torch.randn(B, Cin, H, W) * 0.5
+        w0 = torch.randn(Cout, Cin, 3, 3) * 0.05
+        b0 = torch.randn(Cout) * 0.1
+        dy0 = torch.randn(B, Cout, H, W)
+
+        x = x0.to(DEV, torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last) \
+            .requires_grad_(True)
+        w = w0.to(DEV, torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last) \
+            .requires_grad_(True)
+        b = b0.to(DEV, torch.bfloat16).requires_grad_(True)
+        y = Conv3x3Train.apply(x, w, b, relu)
+        y.backward(dy0.to(DEV, torch.bfloat16)
+                   .contiguous(memory_format=torch.channels_last))
+
+        xr = x0.to(DEV).requires_grad_(True)
+        wr = w0.to(DEV).requires_grad_(True)
+        br = b0.to(DEV).requires_grad_(True)
+        yr = torch.nn.functional.conv2d(xr, wr, br, padding=1)
+        if relu:
+            yr = torch.relu(yr)
+        yr.backward(dy0.to(DEV))
+
+        assert _rel_err(y, yr) < 2e-2
+        assert _rel_err(x.grad, xr.grad) < 3e-2, (Cin, Cout, relu)
+        assert _rel_err(w.grad, wr.grad) < 3e-2, (Cin, Cout, relu)
+        assert _rel_err(b.grad, br.grad) < 3e-2, (Cin, Cout, relu)
