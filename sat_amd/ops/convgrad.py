@@ -66,13 +66,15 @@ class Conv3x3Train(torch.autograd.Function):
         M = B * H * W
 
         dy = dy.contiguous(memory_format=torch.channels_last)
-        if ctx.relu:
-            dy = _C.act_bwd(dy, y, ACT_RELU)
-
-        # NHWC storage viewed as [M, Cout] rows
+        # NHWC storage viewed as [M, Cout] rows (contiguous, zero-copy)
         dy_rows = dy.permute(0, 2, 3, 1).reshape(M, Cout)
         if not dy_rows.is_contiguous():
             dy_rows = dy_rows.contiguous()
+        if ctx.relu:
+            y_rows = y.permute(0, 2, 3, 1).reshape(M, Cout)
+            dy_rows = _C.act_bwd(dy_rows, y_rows, ACT_RELU)
+            # back to a channels_last 4D view for dgrad
+            dy = dy_rows.reshape(B, H, W, Cout).permute(0, 3, 1, 2)
 
         dbias = dy_rows.float().sum(0).to(dy.dtype) \
             if ctx.has_bias else None
