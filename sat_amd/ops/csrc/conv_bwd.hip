@@ -5,11 +5,14 @@
 // viewed as ONE GEMM  C[Cout][9*Cin] = dy^T [Cout][M] @ Xcol [M][9*Cin]
 // whose output IS the OHWI weight layout the forward kernels consume.
 // Both operands are K(=m)-major in memory, so each 64-m chunk is staged
-// TRANSPOSED into LDS ([c][m] image, +8 element row padding to spread
-// the scalar-write banks) and the MFMA fragments then read contiguous
-// k runs.  Split-K over m-chunks with an fp32 atomicAdd epilogue fills
-// the chip on small-tile layers (conv1/2) where (9Cin/64)x(Cout/64)
-// blocks alone would not.
+// TRANSPOSED into LDS: each thread loads 8 consecutive-m rows of one
+// 8-channel slice and repacks them in registers so every LDS store is a
+// full ds_write_b128 (8 m-values of one channel) — no scalar-write
+// transpose tax.  Waves 0-1 stage dy, waves 2-3 stage x (each k-chunk =
+// 8 loads + 8 b128 writes per thread).  128x128 output tile, 4 waves,
+// +8-element LDS row padding for conflict-free transposed reads.
+// Split-K over m-chunks with an fp32 atomicAdd epilogue (plain stores
+// when the grid needs no split) fills the chip on small-tile layers.
 //
 // dgrad needs no new kernel: it is the same 3x3/s1 conv as forward with
 // the weight flipped and io-transposed (sat_amd/ops/convgrad.py routes
@@ -19,7 +22,7 @@
 
 namespace {
 constexpr int PAD = 8;                 // LDS row pad (bf16 elements)
-constexpr int LROW = 64 + PAD;         // LDS row stride
+constexpr int LROW = 64 + PAD;         // LDS row stride (m-extent 64)
 }
 
 __global__ __launch_bounds__(256)
@@ -27,64 +30,123 @@ void conv3x3_wgrad_kernel(const bf16* __restrict__ xpad, // [B,H+2,W+2,Ci]
                           const bf16* __restrict__ dy,   // [M, Cout] NHWC
                           float* __restrict__ dw,        // [Cout, 9*Cin]
                           int M, int Hh, int Ww, int Cin, int Cout,
-                          int chunks_per, int nchunks) {
-    __shared__ bf16 lds[2 * 64 * LROW];
-    bf16* A = lds;                     // [64 co][LROW m]
-    bf16* Bx = lds + 64 * LROW;        // [64 n ][LROW m]
+                          int chunks_per, int nchunks, int use_atomic) {
+    __shared__ bf16 lds[2 * 128 * LROW];                 // 36 KiB
+    bf16* A = lds;                      // [128 co][LROW m]
+    bf16* Bx = lds + 128 * LROW;        // [128 n ][LROW m]
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
     const int wave = tid >> 6;
     const int wr = wave >> 1, wc = wave & 1;
-    const int n0 = blockIdx.x * 64;    // within 9*Cin
-    const int co0 = blockIdx.y * 64;
-    const int dxy = n0 / Cin;          // this block's (ky,kx) plane
-    const int ci0 = n0 % Cin;          // 64-aligned since Cin % 64 == 0
-    const int dyy = dxy / 3 - 1, dxx = dxy % 3 - 1;
+    const int n0 = blockIdx.x * 128;    // within 9*Cin
+    const int co0 = blockIdx.y * 128;
     const int Wp = Ww + 2;
     const int HW = Hh * Ww;
+    const int N9 = 9 * Cin;
 
-    floatx4 acc[2][2];
+    // ---- staging role of this thread (constant): waves 0-1 stage dy,
+    // waves 2-3 stage x; each stages 8 m-rows x 8 channels per chunk ----
+    const int st = tid & 127;           // 0..127 within the role group
+    const int sc8 = (st & 15) * 8;      // channel slice [sc8, sc8+8)
+    const int sm8 = (st >> 4) * 8;      // m slice [sm8, sm8+8) of chunk
+    const bool stage_dy = wave < 2;
+    // clamped global channel base (epilogue guards discard duplicates)
+    int dyco = co0 + sc8;
+    if (dyco > Cout - 8) dyco = Cout - 8;
+    int nidx = n0 + sc8;
+    if (nidx > N9 - 8) nidx = N9 - 8;
+    const int sdxy = nidx / Cin;        // 8-chunk never straddles a dxy
+    const int sci = nidx % Cin;
+    const int sdy = sdxy / 3 - 1, sdx = sdxy % 3 - 1;
+
+    floatx4 acc[4][4];
 #pragma unroll
-    for (int i = 0; i < 2; ++i)
+    for (int i = 0; i < 4; ++i)
 #pragma unroll
-        for (int j = 0; j < 2; ++j)
+        for (int j = 0; j < 4; ++j)
             acc[i][j] = floatx4{0.f, 0.f, 0.f, 0.f};
 
     const int lrow = lane & 15;
     const int kgrp = lane >> 4;
 
-    // this thread's staging slice: rows mrow, mrow+32 of the 64-m chunk,
-    // 16-B chunk c8 of the 64-channel tile
-    const int mrow = tid >> 3;         // 0..31
-    const int c8 = (tid & 7) * 8;
-
     const int c_begin = blockIdx.z * chunks_per;
     const int c_end = min(c_begin + chunks_per, nchunks);
 
     for (int ch = c_begin; ch < c_end; ++ch) {
-        const int m0 = ch * 64;
+        const int m0 = ch * 64 + sm8;   // this thread's first m
+        const bool full = (m0 + 7 < M);  // hoisted guard: per-element
+        bf16x8 v[8];                     // load-selects de-pipeline
+        const bf16x8 vz = {};            // (guide §5 trap c)
+        if (stage_dy) {
+            const bf16* p = dy + (int64_t)m0 * Cout + dyco;
+            if (full) {
 #pragma unroll
-        for (int half = 0; half < 2; ++half) {
-            const int mi = mrow + half * 32;
-            const int m = m0 + mi;
-            bf16x8 vdy = {};
-            bf16x8 vx = {};
-            if (m < M) {
-                vdy = *(const bf16x8*)(dy + (int64_t)m * Cout + co0 + c8);
-                const int b = m / HW;
-                const int yx = m % HW;
-                const int y = yx / Ww, x = yx % Ww;
-                const int64_t base =
-                    (((int64_t)b * (Hh + 2) + y + 1 + dyy) * Wp
-                     + x + 1 + dxx) * Cin;
-                vx = *(const bf16x8*)(xpad + base + ci0 + c8);
+                for (int e = 0; e < 8; ++e) {
+                    v[e] = *(const bf16x8*)p;
+                    p += Cout;
+                }
+            } else {
+                for (int e = 0; e < 8; ++e) {
+                    v[e] = (m0 + e < M) ? *(const bf16x8*)p : vz;
+                    p += Cout;
+                }
             }
-            // transposed scalar writes: LDS[c][m]
+        } else {
+            // incremental padded-image base over 8 consecutive pixels
+            int b = m0 / HW, yx = m0 % HW;
+            int y = yx / Ww, x = yx % Ww;
+            const bf16* p = xpad
+                + (((int64_t)b * (Hh + 2) + y + 1 + sdy) * Wp
+                   + x + 1 + sdx) * Cin + sci;
+            if (full) {
+#pragma unroll
+                for (int e = 0; e < 8; ++e) {
+                    v[e] = *(const bf16x8*)p;
+                    ++x;
+                    if (x == Ww) {      // wrap to next image row
+                        x = 0; ++y;
+                        if (y == Hh) {  // wrap to next batch image
+                            y = 0;
+                            p += (int64_t)3 * Wp * Cin
+                                - (Ww - 1) * Cin;
+                        } else {
+                            p += (int64_t)3 * Cin;  // 2 pad cols + 1
+                        }
+                    } else {
+                        p += Cin;
+                    }
+                }
+            } else {
+                for (int e = 0; e < 8; ++e) {
+                    v[e] = (m0 + e < M) ? *(const bf16x8*)p : vz;
+                    ++x;
+                    if (x == Ww) {
+                        x = 0; ++y;
+                        if (y == Hh) {
+                            y = 0;
+                            p += (int64_t)3 * Wp * Cin
+                                - (Ww - 1) * Cin;
+                        } else {
+                            p += (int64_t)3 * Cin;
+                        }
+                    } else {
+                        p += Cin;
+                    }
+                }
+            }
+        }
+        __syncthreads();                // previous MFMA done before write
+        {
+            bf16* dst = (stage_dy ? A : Bx) + (int64_t)sc8 * LROW + sm8;
+            // repack: row e of LDS gets channel e across the 8 m's
 #pragma unroll
             for (int e = 0; e < 8; ++e) {
-                A[(c8 + e) * LROW + mi] = vdy[e];
-                Bx[(c8 + e) * LROW + mi] = vx[e];
+                bf16x8 o;
+#pragma unroll
+                for (int mth = 0; mth < 8; ++mth)
+                    o[mth] = v[mth][e];
+                *(bf16x8*)(dst + e * LROW) = o;
             }
         }
         __syncthreads();
@@ -92,37 +154,39 @@ void conv3x3_wgrad_kernel(const bf16* __restrict__ xpad, // [B,H+2,W+2,Ci]
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk) {
             const int kof = kk * 32 + kgrp * 8;
-            bf16x8 af[2], bf[2];
+            bf16x8 af[4], bf[4];
 #pragma unroll
-            for (int i = 0; i < 2; ++i) {
+            for (int i = 0; i < 4; ++i) {
                 af[i] = *(const bf16x8*)(
-                    A + (wr * 32 + i * 16 + lrow) * LROW + kof);
+                    A + (wr * 64 + i * 16 + lrow) * LROW + kof);
                 bf[i] = *(const bf16x8*)(
-                    Bx + (wc * 32 + i * 16 + lrow) * LROW + kof);
+                    Bx + (wc * 64 + i * 16 + lrow) * LROW + kof);
             }
 #pragma unroll
-            for (int i = 0; i < 2; ++i)
+            for (int i = 0; i < 4; ++i)
 #pragma unroll
-                for (int j = 0; j < 2; ++j)
+                for (int j = 0; j < 4; ++j)
                     acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         af[i], bf[j], acc[i][j], 0, 0, 0);
         }
-        __syncthreads();
     }
 
-    // fp32 atomic epilogue (dw zero-initialized by the host wrapper)
+    // epilogue: atomics only when K was split (dw pre-zeroed by host)
 #pragma unroll
-    for (int j = 0; j < 2; ++j) {
-        const int col = n0 + wc * 32 + j * 16 + (lane & 15);
+    for (int j = 0; j < 4; ++j) {
+        const int col = n0 + wc * 64 + j * 16 + (lane & 15);
+        if (col >= N9) continue;
 #pragma unroll
-        for (int i = 0; i < 2; ++i) {
+        for (int i = 0; i < 4; ++i) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                const int row = co0 + wr * 32 + i * 16 + (lane >> 4) * 4
+                const int row = co0 + wr * 64 + i * 16 + (lane >> 4) * 4
                     + r;
-                if (row < Cout)
-                    atomicAdd(dw + (int64_t)row * 9 * Cin + col,
-                              acc[i][j][r]);
+                if (row < Cout) {
+                    float* d = dw + (int64_t)row * N9 + col;
+                    if (use_atomic) atomicAdd(d, acc[i][j][r]);
+                    else *d = acc[i][j][r];
+                }
             }
         }
     }
@@ -142,21 +206,22 @@ at::Tensor conv3x3_wgrad(at::Tensor xpad, at::Tensor dy_rows,
     TORCH_CHECK(M64 < (1LL << 31));
     int M = (int)M64;
 
-    auto dw = at::zeros({Cout, 9 * Cin},
-                        xpad.options().dtype(at::kFloat));
     int nchunks = cdiv(M, 64);
-    int tiles = (9 * Cin / 64) * (Cout / 64);
-    // fill ~2 blocks/CU (the kernel is 2-wave-tile, high occupancy)
+    int tiles = cdiv(9 * Cin, 128) * cdiv(Cout, 128);
     int splitk = std::max(1, std::min(32, 512 / tiles));
     int chunks_per = cdiv(nchunks, splitk);
     splitk = cdiv(nchunks, chunks_per);
-    dim3 grid(9 * Cin / 64, Cout / 64, splitk);
+    auto dw = (splitk > 1)
+        ? at::zeros({Cout, 9 * Cin}, xpad.options().dtype(at::kFloat))
+        : at::empty({Cout, 9 * Cin}, xpad.options().dtype(at::kFloat));
+    dim3 grid(cdiv(9 * Cin, 128), cdiv(Cout, 128), splitk);
     hipStream_t s = at::cuda::getCurrentCUDAStream();
     hipLaunchKernelGGL(conv3x3_wgrad_kernel, grid, dim3(256), 0, s,
                        (const bf16*)xpad.data_ptr(),
                        (const bf16*)dy_rows.data_ptr(),
                        (float*)dw.data_ptr(), M, (int)Hh, (int)Ww,
-                       (int)Cin, (int)Cout, chunks_per, nchunks);
+                       (int)Cin, (int)Cout, chunks_per, nchunks,
+                       splitk > 1 ? 1 : 0);
     HIP_OK(hipGetLastError());
     return dw;
 }

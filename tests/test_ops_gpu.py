@@ -443,11 +443,18 @@ def test_conv3x3_train_function_grads():
         wr = w0.to(DEV).requires_grad_(True)
         br = b0.to(DEV).requires_grad_(True)
         yr = torch.nn.functional.conv2d(xr, wr, br, padding=1)
+        dy_ref = dy0.to(DEV)
         if relu:
-            yr = torch.relu(yr)
-        yr.backward(dy0.to(DEV))
+            # use OUR kernel's bf16 ReLU mask in the fp32 reference:
+            # elements with y ~ 0 flip sides of the threshold under bf16
+            # rounding, which is a representation artifact, not a grad
+            # bug — a mask-consistent reference isolates the real math
+            dy_ref = dy_ref * (y.float() > 0)
+            assert _rel_err(y, torch.relu(yr)) < 2e-2
+        else:
+            assert _rel_err(y, yr) < 2e-2
+        yr.backward(dy_ref)
 
-        assert _rel_err(y, yr) < 2e-2
         assert _rel_err(x.grad, xr.grad) < 3e-2, (Cin, Cout, relu)
         assert _rel_err(w.grad, wr.grad) < 3e-2, (Cin, Cout, relu)
         assert _rel_err(b.grad, br.grad) < 3e-2, (Cin, Cout, relu)
