@@ -209,7 +209,7 @@ class Conv2d(tnn.Module):
                 M = x.shape[0] * x.shape[2] * x.shape[3]
                 relu = self.activation == 'relu'
                 use_8p = (self._glds_conv and Cout % 256 == 0
-                          and Cin % 64 == 0 and M >= 12544)
+                          and Cin % 64 == 0 and M >= 3136)
                 use_glds = (self._glds_conv and not use_8p
                             and Cout >= 128 and Cout % 8 == 0
                             and Cin % 64 == 0 and Cin <= 128)

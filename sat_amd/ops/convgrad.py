@@ -31,7 +31,7 @@ def _route_fwd(x, w_ohwi, eb, relu):
     Cin = x.shape[1]
     H, W = x.shape[2], x.shape[3]
     M = x.shape[0] * H * W
-    if Cout % 256 == 0 and Cin % 64 == 0 and M >= 12544:
+    if Cout % 256 == 0 and Cin % 64 == 0 and M >= 3136:
         return _C.conv_igemm_8p_fwd(_C.pad1_nhwc(x), w_ohwi, eb, H, W,
                                     relu)
     if Cout >= 128 and Cout % 8 == 0 and Cin % 64 == 0 and Cin <= 128:

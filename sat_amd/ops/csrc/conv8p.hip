@@ -53,13 +53,19 @@ __device__ __forceinline__ int swz16(int row) {
 
 }  // namespace
 
+// SPLIT=0: one block owns a tile's whole K; bias/ReLU epilogue to bf16.
+// SPLIT=1: grid.z slices the K-tile range; fp32 atomicAdd into scratch
+// (zeroed by the host), bias/ReLU applied by f32_bias_act_nhwc — fills
+// the chip on low-M shapes (conv5 at batch 32: 50 blocks -> 200).
+template <int SPLIT>
 __global__ __launch_bounds__(512)
 void conv_igemm_8p_kernel(const bf16* __restrict__ inp,  // padded NHWC
                           const bf16* __restrict__ w,    // [Cout,9*Cin]
                           const bf16* __restrict__ bias,
                           bf16* __restrict__ out,
+                          float* __restrict__ outf,
                           int M, int Hh, int Ww, int Cin, int Cout,
-                          int relu) {
+                          int relu, int kt_per) {
     __shared__ bf16 lds[8 * SLOT];            // ONE shared object (guide
                                               // §5 trap 4a) = 128 KiB
 
@@ -103,6 +109,8 @@ void conv_igemm_8p_kernel(const bf16* __restrict__ inp,  // padded NHWC
     bf16* dstJ1 = lds + (64 + wave * 8) * 64;
 
     const int KT = 9 * (Cin / 64);
+    const int kt0 = SPLIT ? blockIdx.z * kt_per : 0;
+    const int kt1 = SPLIT ? min(kt0 + kt_per, KT) : KT;
 
     // per-K-tile source deltas: tile kt has (dxy = kt / (Cin/64),
     // ci0 = (kt % (Cin/64)) * 64)
@@ -188,9 +196,9 @@ void conv_igemm_8p_kernel(const bf16* __restrict__ inp,  // padded NHWC
 
     bf16x8 afr[4][2], bfr[2][2];
 
-    // ---- prologue: stage K-tile 0 (4 half-tiles, 8 glds/thread),
-    // then retire A0+B0 before any wave reads them ----
-    STAGE(0, 0); STAGE(0, 1); STAGE(0, 2); STAGE(0, 3);
+    // ---- prologue: stage the first K-tile (4 half-tiles, 8 glds per
+    // thread), then retire A0+B0 before any wave reads them ----
+    STAGE(kt0, 0); STAGE(kt0, 1); STAGE(kt0, 2); STAGE(kt0, 3);
     VMW(4); BAR();
 
     // ---- steady loop: compute tile t, stage tile t+1 ----
@@ -199,7 +207,7 @@ void conv_igemm_8p_kernel(const bf16* __restrict__ inp,  // padded NHWC
     // counted-wait own stagings, raw barrier (cross-wave publication of
     // the wait), drain lgkm, MFMA the quadrant, barrier (WAR fence for
     // the slot restaged 8 phases after its staging).
-    for (int t = 0; t + 1 < KT; ++t) {
+    for (int t = kt0; t + 1 < kt1; ++t) {
         // phase 4t: quadrant (A0,B0)
         READ_A(afr, t, 0); READ_B(bfr, t, 0);
         STAGE(t + 1, 0);
@@ -229,7 +237,7 @@ void conv_igemm_8p_kernel(const bf16* __restrict__ inp,  // padded NHWC
     // ---- tail tile: everything staged; drain once, then plain phases ----
     VMW(0); BAR();
     {
-        const int t = KT - 1;
+        const int t = kt1 - 1;
         READ_A(afr, t, 0); READ_B(bfr, t, 0); LGKM0();
         MFMA16(0, 0);
         READ_A(afr, t, 1); LGKM0();
@@ -250,13 +258,13 @@ void conv_igemm_8p_kernel(const bf16* __restrict__ inp,  // padded NHWC
 #undef A_DELTA
 #undef B_DELTA
 
-    // ---- epilogue: bias + ReLU, guarded store ----
+    // ---- epilogue: bias + ReLU + guarded store (or split-K atomics) ----
 #pragma unroll
     for (int bH = 0; bH < 2; ++bH) {
 #pragma unroll
         for (int ni = 0; ni < 2; ++ni) {
             int col = bn + bH * 128 + wc * 32 + ni * 16 + (lane & 15);
-            float bv = (bias != nullptr && col < Cout)
+            float bv = (!SPLIT && bias != nullptr && col < Cout)
                 ? bf2f(bias[col]) : 0.f;
 #pragma unroll
             for (int aH = 0; aH < 2; ++aH) {
@@ -267,15 +275,39 @@ void conv_igemm_8p_kernel(const bf16* __restrict__ inp,  // padded NHWC
                         int row = bm + aH * 128 + wr * 64 + mi * 16
                             + (lane >> 4) * 4 + r;
                         if (row < M && col < Cout) {
-                            float v =
-                                acc[aH * 4 + mi][bH * 2 + ni][r] + bv;
-                            if (relu) v = fmaxf(v, 0.f);
-                            out[(int64_t)row * Cout + col] = f2bf(v);
+                            if (SPLIT) {
+                                atomicAdd(
+                                    outf + (int64_t)row * Cout + col,
+                                    acc[aH * 4 + mi][bH * 2 + ni][r]);
+                            } else {
+                                float v = acc[aH * 4 + mi]
+                                    [bH * 2 + ni][r] + bv;
+                                if (relu) v = fmaxf(v, 0.f);
+                                out[(int64_t)row * Cout + col]
+                                    = f2bf(v);
+                            }
                         }
                     }
                 }
             }
         }
+    }
+}
+
+// fp32 split-K scratch -> bias + act -> bf16 NHWC
+__global__ void f32_bias_act_nhwc_kernel(const float* __restrict__ yf,
+                                         const bf16* __restrict__ bias,
+                                         bf16* __restrict__ out,
+                                         int64_t n, int C, int relu) {
+    int64_t i8 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+    if (i8 >= n) return;
+    int c0 = (int)(i8 % C);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+        float t = yf[i8 + e];
+        if (bias != nullptr) t += bf2f(bias[c0 + e]);
+        if (relu) t = fmaxf(t, 0.f);
+        out[i8 + e] = f2bf(t);
     }
 }
 
@@ -300,13 +332,39 @@ at::Tensor conv_igemm_8p_fwd(at::Tensor padded, at::Tensor w_ohwi,
     int64_t M64 = (int64_t)B * Hh * Ww;
     TORCH_CHECK(M64 < (1LL << 31), "conv_igemm_8p: M too large");
     int M = (int)M64;
-    dim3 grid(cdiv(Cout, 256), cdiv(M, 256));
+    int blocks = cdiv(Cout, 256) * cdiv(M, 256);
+    int KT = 9 * (Cin / 64);
+    int splitz = std::max(1, std::min({8, 224 / blocks, KT / 2}));
     hipStream_t s = at::cuda::getCurrentCUDAStream();
-    hipLaunchKernelGGL(conv_igemm_8p_kernel, grid, dim3(512), 0, s,
-                       (const bf16*)padded.data_ptr(),
-                       (const bf16*)w_ohwi.data_ptr(), bias_ptr,
-                       (bf16*)out.data_ptr(), M, (int)Hh, (int)Ww,
-                       (int)Cin, (int)Cout, relu ? 1 : 0);
+    if (splitz > 1) {
+        auto scratch = at::zeros({M, Cout},
+                                 padded.options().dtype(at::kFloat));
+        int kt_per = cdiv(KT, splitz);
+        splitz = cdiv(KT, kt_per);
+        dim3 grid(cdiv(Cout, 256), cdiv(M, 256), splitz);
+        hipLaunchKernelGGL((conv_igemm_8p_kernel<1>), grid, dim3(512),
+                           0, s,
+                           (const bf16*)padded.data_ptr(),
+                           (const bf16*)w_ohwi.data_ptr(), bias_ptr,
+                           nullptr, (float*)scratch.data_ptr(),
+                           M, (int)Hh, (int)Ww,
+                           (int)Cin, (int)Cout, relu ? 1 : 0, kt_per);
+        int64_t n = (int64_t)M * Cout;
+        hipLaunchKernelGGL(f32_bias_act_nhwc_kernel,
+                           dim3(cdiv(n / 8, 256)), dim3(256), 0, s,
+                           (const float*)scratch.data_ptr(), bias_ptr,
+                           (bf16*)out.data_ptr(), n, (int)Cout,
+                           relu ? 1 : 0);
+    } else {
+        dim3 grid(cdiv(Cout, 256), cdiv(M, 256));
+        hipLaunchKernelGGL((conv_igemm_8p_kernel<0>), grid, dim3(512),
+                           0, s,
+                           (const bf16*)padded.data_ptr(),
+                           (const bf16*)w_ohwi.data_ptr(), bias_ptr,
+                           (bf16*)out.data_ptr(), nullptr,
+                           M, (int)Hh, (int)Ww,
+                           (int)Cin, (int)Cout, relu ? 1 : 0, KT);
+    }
     HIP_OK(hipGetLastError());
     return out;
 }
