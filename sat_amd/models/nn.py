@@ -208,8 +208,11 @@ class Conv2d(tnn.Module):
                 Cout, Cin = w.shape[0], w.shape[1]
                 M = x.shape[0] * x.shape[2] * x.shape[3]
                 relu = self.activation == 'relu'
+                # 8p wins every Cout%256 shape with M>=12544; at
+                # conv5-batch-32 (M=6272) MIOpen gk still wins 77 vs
+                # 99us (profiles/r02_conv_shapes.txt) — evidence-routed
                 use_8p = (self._glds_conv and Cout % 256 == 0
-                          and Cin % 64 == 0 and M >= 3136)
+                          and Cin % 64 == 0 and M >= 12544)
                 use_glds = (self._glds_conv and not use_8p
                             and Cout >= 128 and Cout % 8 == 0
                             and Cin % 64 == 0 and Cin <= 128)
