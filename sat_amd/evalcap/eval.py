@@ -48,7 +48,13 @@ class COCOEvalCap(object):
             else:
                 self.setEval(score, method)
                 self.setImgToEvalImgs(scores, img_ids, method)
-                print("%s: %0.3f" % (method, score))
+                label = method
+                if method == "METEOR":
+                    # exact+stem beam alignment only — no synonym /
+                    # paraphrase stages (missing data files); NOT
+                    # comparable to published METEOR numbers
+                    label = "METEOR (approx: exact+stem stages)"
+                print("%s: %0.3f" % (label, score))
         self.setEvalImgs()
 
     def setEval(self, score, method):

@@ -1,15 +1,32 @@
 """METEOR for caption evaluation (pure Python, no Java).
 
 The reference shells out to the METEOR-1.5 Java jar through a persistent
-subprocess pipe (`utils/coco/pycocoevalcap/meteor/meteor.py:15-58`); the jar
-is a missing git-LFS blob in this environment.  This module re-implements the
-METEOR scoring *formula* in Python with the exact-match and simple-stem
-matcher stages (the paraphrase-table stage needs the missing
-`paraphrase-en.gz` data file and is omitted — documented approximation).
+subprocess pipe (`utils/coco/pycocoevalcap/meteor/meteor.py:15-58`); the
+jar is a missing git-LFS blob in this environment.  This module
+re-implements METEOR 1.5's scoring pipeline:
 
-Parameters are METEOR 1.5's English defaults: alpha=0.85, beta=0.2,
-gamma=0.6 (fragmentation penalty on chunk count).  Interface matches the
-reference: `compute_score(gts, res)` -> (mean score, per-image scores).
+  * candidate matches from the EXACT (weight 1.0) and STEM (weight 0.6)
+    matcher stages;
+  * alignment resolution as METEOR does it — a BEAM SEARCH over match
+    permutations selecting the alignment with (1) the most matched
+    words, then (2) the fewest chunks, then (3) the highest stage
+    weight — not a greedy left-to-right pass;
+  * weighted P/R fmean (alpha=0.85) and the fragmentation penalty
+    gamma * (chunks/matches)^beta with METEOR 1.5's English defaults
+    (beta=0.2, gamma=0.6);
+  * per-segment best-reference selection.
+
+Documented approximations vs METEOR 1.5 proper (both need data files
+that are missing LFS blobs in the reference checkout too):
+  * no SYNONYM stage (needs WordNet) and no PARAPHRASE stage (needs
+    paraphrase-en.gz);
+  * no function-word discounting (delta) — without a function-word list
+    every token is a content word, which is the delta-neutral case.
+Scores are therefore NOT comparable to published METEOR numbers; eval
+output labels them "METEOR (approx)".
+
+Interface matches the reference: `compute_score(gts, res)` ->
+(mean score, per-image scores).
 """
 
 import numpy as np
@@ -17,64 +34,84 @@ import numpy as np
 _ALPHA = 0.85
 _BETA = 0.2
 _GAMMA = 0.6
+_W_EXACT = 1.0
+_W_STEM = 0.6
+_BEAM = 128
 
 
 def _stem(w):
-    """Tiny Porter-ish suffix stripper — approximates METEOR's stem module."""
+    """Tiny Porter-ish suffix stripper — approximates METEOR's stemmer."""
     for suf in ('ing', 'edly', 'ed', 'es', 's', 'ly'):
         if w.endswith(suf) and len(w) - len(suf) >= 3:
             return w[: len(w) - len(suf)]
     return w
 
 
-def _align(hyp, ref):
-    """Greedy left-to-right alignment: exact first, then stem matches.
-
-    Returns (num_matches, num_chunks) where chunks are maximal runs of
-    matches that are contiguous and order-preserving in both strings.
-    """
-    m = len(hyp)
-    used_ref = [False] * len(ref)
-    match_of = [None] * m  # hyp position -> ref position
-
-    # stage 1: exact
-    for i, w in enumerate(hyp):
+def _candidates(hyp, ref):
+    """Per hyp position: list of (ref position, stage weight) options,
+    exact matches preferred over stem matches for the same pair."""
+    rs = [_stem(r) for r in ref]
+    out = []
+    for w in hyp:
+        ws = _stem(w)
+        opts = []
         for j, r in enumerate(ref):
-            if not used_ref[j] and match_of[i] is None and w == r:
-                match_of[i] = j
-                used_ref[j] = True
-                break
-    # stage 2: stem
-    hs = [_stem(w) for w in hyp]
-    rs = [_stem(w) for w in ref]
-    for i in range(m):
-        if match_of[i] is not None:
-            continue
-        for j in range(len(ref)):
-            if not used_ref[j] and hs[i] == rs[j]:
-                match_of[i] = j
-                used_ref[j] = True
-                break
+            if w == r:
+                opts.append((j, _W_EXACT))
+            elif ws == rs[j]:
+                opts.append((j, _W_STEM))
+        out.append(opts)
+    return out
 
-    pairs = [(i, j) for i, j in enumerate(match_of) if j is not None]
-    matches = len(pairs)
-    if matches == 0:
-        return 0, 0
-    chunks = 1
-    for (i0, j0), (i1, j1) in zip(pairs, pairs[1:]):
-        if not (i1 == i0 + 1 and j1 == j0 + 1):
-            chunks += 1
-    return matches, chunks
+
+def _align(hyp, ref):
+    """METEOR-style alignment via beam search.
+
+    State per partial alignment (hyp prefix processed):
+      (used-ref bitmask, matches, chunks, weight, last matched (i, j)).
+    Returns (matches, chunks, total stage weight).
+    """
+    cands = _candidates(hyp, ref)
+    # beam entries: key = used mask; value tuple
+    # (matches, -chunks, weight) is the comparison order METEOR uses
+    beam = [(0, 0, 0, 0.0, None)]  # mask, matches, chunks, weight, last
+    for i, opts in enumerate(cands):
+        nxt = []
+        for mask, m, ch, wsum, last in beam:
+            # skip this hyp word
+            nxt.append((mask, m, ch, wsum, last))
+            for j, w in opts:
+                bit = 1 << j
+                if mask & bit:
+                    continue
+                contig = (last is not None
+                          and last[0] == i - 1 and last[1] == j - 1)
+                nxt.append((mask | bit, m + 1,
+                            ch + (0 if contig else 1),
+                            wsum + w, (i, j)))
+        # prune: best state per (comparable) rank
+        nxt.sort(key=lambda s: (-s[1], s[2], -s[3]))
+        seen = set()
+        beam = []
+        for s in nxt:
+            if s[0] in seen:
+                continue
+            seen.add(s[0])
+            beam.append(s)
+            if len(beam) >= _BEAM:
+                break
+    best = max(beam, key=lambda s: (s[1], -s[2], s[3]))
+    return best[1], best[2], best[3]
 
 
 def _score_pair(hyp_toks, ref_toks):
     if not hyp_toks or not ref_toks:
         return 0.0
-    m, ch = _align(hyp_toks, ref_toks)
+    m, ch, wsum = _align(hyp_toks, ref_toks)
     if m == 0:
         return 0.0
-    p = m / len(hyp_toks)
-    r = m / len(ref_toks)
+    p = wsum / len(hyp_toks)
+    r = wsum / len(ref_toks)
     fmean = (p * r) / (_ALPHA * p + (1 - _ALPHA) * r)
     frag = ch / m
     penalty = _GAMMA * (frag ** _BETA)
@@ -83,7 +120,7 @@ def _score_pair(hyp_toks, ref_toks):
 
 class Meteor(object):
     def method(self):
-        return "METEOR"
+        return "METEOR (approx)"
 
     def compute_score(self, gts, res):
         scores = []

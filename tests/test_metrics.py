@@ -109,3 +109,34 @@ def test_metrics_tolerate_empty_candidate():
 def test_bleu_single_word():
     scores, _ = Bleu(4).compute_score({1: ['dog']}, {1: ['dog']})
     assert scores[0] > 0.9  # unigram perfect; higher n-grams degenerate
+
+
+def test_meteor_beam_alignment_hand_computed():
+    """Hand-checkable METEOR alignments: the beam aligner must find the
+    chunk-minimizing permutation a greedy left-to-right matcher misses."""
+    from sat_amd.evalcap.meteor import _align, _score_pair
+
+    # identical sentences: all matched, one chunk
+    m, ch, w = _align(['a', 'b', 'c'], ['a', 'b', 'c'])
+    assert (m, ch, w) == (3, 1, 3.0)
+
+    # duplicate word: a greedy left-to-right matcher binds hyp 'the' to
+    # ref[0], splitting the alignment into 3 chunks.  The optimal
+    # permutation (dog->1, and->2, the->3, cat->4) is contiguous in
+    # both sentences: ONE chunk.
+    hyp = ['dog', 'and', 'the', 'cat']
+    ref = ['the', 'dog', 'and', 'the', 'cat']
+    m, ch, w = _align(hyp, ref)
+    assert m == 4 and w == 4.0
+    assert ch == 1, 'expected chunk-minimizing alignment, got %d' % ch
+
+    # stem-stage match carries weight 0.6
+    m, ch, w = _align(['running'], ['runs'])
+    assert m == 1 and abs(w - 0.6) < 1e-9
+
+    # no matches
+    assert _score_pair(['x'], ['y']) == 0.0
+
+    # perfect match scores fmean * (1 - gamma * 1^beta)
+    s = _score_pair(['a', 'b'], ['a', 'b'])
+    assert abs(s - (1.0 * (1 - 0.6))) < 1e-9
