@@ -137,6 +137,7 @@ def test_meteor_beam_alignment_hand_computed():
     # no matches
     assert _score_pair(['x'], ['y']) == 0.0
 
-    # perfect match scores fmean * (1 - gamma * 1^beta)
+    # perfect 2-word match: fmean=1, frag = 1 chunk / 2 matches = 0.5,
+    # penalty = gamma * 0.5^beta
     s = _score_pair(['a', 'b'], ['a', 'b'])
-    assert abs(s - (1.0 * (1 - 0.6))) < 1e-9
+    assert abs(s - (1.0 - 0.6 * 0.5 ** 0.2)) < 1e-9
