@@ -102,6 +102,9 @@ class Config(object):
         #                                  architecture only; False = per-op
         #                                  autograd loop, same numerics)
         self.allreduce_bucket_mb = 16    # DP gradient bucket size (xGMI-tuned)
+        self.use_device_beam = True      # device-resident beam search (one
+        #                                  host sync per batch; False = the
+        #                                  host-heap reference path)
         self.use_glds_conv = True        # LDS-staged (glds) implicit-GEMM conv
         #                                  kernels for Cin<=128/Cout>=128 3x3
         #                                  layers (beats MIOpen per

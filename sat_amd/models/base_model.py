@@ -248,7 +248,137 @@ class BaseModel(object):
     @torch.no_grad()
     def beam_search(self, image_files, vocabulary):
         """Beam-search captions for a batch of images
-        (semantics of reference base_model.py:163-240)."""
+        (semantics of reference base_model.py:163-240).
+
+        Dispatches to the device-resident scorer (one host sync per
+        batch) unless config.use_device_beam is False; the host-heap
+        path below is the semantics reference."""
+        if getattr(self.config, 'use_device_beam', True):
+            return self.beam_search_device(image_files, vocabulary)
+        return self.beam_search_host(image_files, vocabulary)
+
+    @torch.no_grad()
+    def beam_search_device(self, image_files, vocabulary):
+        """Device-resident beam search: per-step candidate scoring,
+        top-k selection, '.'-termination and the bounded completed set
+        all run on-GPU; the only host sync is the final result readout.
+
+        Hypothesis semantics are the reference's (base_model.py:184-240):
+        beam_size+1 expansion per live hypothesis, probability-PRODUCT
+        scores (fp64, matching host float math), completed set bounded
+        at beam_size, partial fallback when nothing completed."""
+        config = self.config
+        self.model.eval()
+        W = beam_size = getattr(config, 'beam_size', 3)
+        B = len(image_files)
+        T = config.max_caption_length
+        dev = self.device
+
+        images = self._images_to_device(
+            self.image_loader.load_images(image_files))
+        contexts, init_memory, init_output = self.model.encode(images)
+        H = init_memory.shape[1]
+
+        try:
+            period = vocabulary.words.index('.')
+        except ValueError:
+            period = -1
+
+        # fixed W slots per image; dead slots carry score -inf
+        NEG = float('-inf')
+        scores = torch.full((B, W), NEG, dtype=torch.float64, device=dev)
+        scores[:, 0] = 0.0  # log-space accumulator? no: product below
+        scores = scores.exp()                     # [B,W]: 1, 0, 0
+        seqs = torch.zeros(B, W, T, dtype=torch.int64, device=dev)
+        lens = torch.zeros(B, W, dtype=torch.int64, device=dev)
+        memory = init_memory.unsqueeze(1).repeat(1, W, 1).clone()
+        output = init_output.unsqueeze(1).repeat(1, W, 1).clone()
+        last_word = torch.zeros(B, W, dtype=torch.int64, device=dev)
+
+        c_scores = torch.zeros(B, W, dtype=torch.float64, device=dev)
+        c_seqs = torch.zeros(B, W, T, dtype=torch.int64, device=dev)
+        c_lens = torch.zeros(B, W, dtype=torch.int64, device=dev)
+
+        ctx_rep = contexts.repeat_interleave(W, dim=0)
+
+        for t in range(T):
+            mem2, out2, probs = self.model.decode_step(
+                ctx_rep, last_word.reshape(-1),
+                memory.reshape(B * W, H), output.reshape(B * W, H))
+            top_p, top_w = probs.topk(beam_size + 1, dim=1)
+            # candidate score = parent score * p(word)  [B, W*(K)]
+            cand = (scores.unsqueeze(2)
+                    * top_p.double().reshape(B, W, -1)).reshape(B, -1)
+            cand_w = top_w.reshape(B, -1)
+            K = beam_size + 1
+
+            is_period = cand_w == period
+            # ---- completed: merge period-candidates into the bounded
+            # completed set (reference TopN(beam_size) push) ----
+            comp_cand = torch.where(is_period, cand,
+                                    torch.zeros_like(cand))
+            merged_s = torch.cat([c_scores, comp_cand], dim=1)
+            parent = torch.arange(W * K, device=dev).reshape(1, -1) \
+                .expand(B, -1) // K
+            new_seq = seqs.gather(
+                1, parent.unsqueeze(2).expand(B, W * K, T)).clone()
+            new_len = lens.gather(1, parent)
+            step_seq = new_seq.scatter(
+                2, new_len.clamp(max=T - 1).reshape(B, W * K, 1),
+                cand_w.reshape(B, W * K, 1))
+            step_len = (new_len + 1).clamp(max=T)
+            merged_seq = torch.cat([c_seqs, step_seq], dim=1)
+            merged_len = torch.cat([c_lens, step_len], dim=1)
+            keep_s, keep_i = merged_s.topk(W, dim=1)
+            c_scores = keep_s
+            c_seqs = merged_seq.gather(
+                1, keep_i.unsqueeze(2).expand(B, W, T))
+            c_lens = merged_len.gather(1, keep_i)
+
+            # ---- partial: best W non-period candidates ----
+            part = torch.where(is_period,
+                               torch.full_like(cand, NEG), cand)
+            part = torch.nan_to_num(part, nan=NEG, neginf=NEG)
+            sel_s, sel_i = part.topk(W, dim=1)
+            scores = sel_s.clamp_min(0.0)
+            sel_parent = sel_i // K
+            seqs = step_seq.gather(
+                1, sel_i.unsqueeze(2).expand(B, W, T))
+            lens = step_len.gather(1, sel_i)
+            last_word = cand_w.gather(1, sel_i)
+            mem3 = mem2.reshape(B, W, H).gather(
+                1, sel_parent.unsqueeze(2).expand(B, W, H))
+            out3 = out2.reshape(B, W, H).gather(
+                1, sel_parent.unsqueeze(2).expand(B, W, H))
+            memory, output = mem3, out3
+
+        # ---- readout (single host sync) ----
+        c_scores_h = c_scores.cpu().numpy()
+        c_seqs_h = c_seqs.cpu().numpy()
+        c_lens_h = c_lens.cpu().numpy()
+        p_scores_h = scores.cpu().numpy()
+        p_seqs_h = seqs.cpu().numpy()
+        p_lens_h = lens.cpu().numpy()
+
+        results = []
+        for k in range(B):
+            use_partial = float(c_scores_h[k].max()) <= 0.0
+            ss = p_scores_h[k] if use_partial else c_scores_h[k]
+            qq = p_seqs_h[k] if use_partial else c_seqs_h[k]
+            ll = p_lens_h[k] if use_partial else c_lens_h[k]
+            order = ss.argsort()[::-1][:beam_size]
+            caps = [CaptionData([int(v) for v in qq[i][:ll[i]]],
+                                None, None, float(ss[i]))
+                    for i in order if ss[i] > 0.0 or use_partial]
+            if not caps:
+                caps = [CaptionData([], None, None, 0.0)]
+            results.append(caps)
+        return results
+
+    @torch.no_grad()
+    def beam_search_host(self, image_files, vocabulary):
+        """Host-heap beam search (the semantics reference; one topk
+        sync per step)."""
         config = self.config
         self.model.eval()
         beam_size = getattr(config, 'beam_size', 3)
