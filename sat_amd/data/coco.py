@@ -134,7 +134,46 @@ class COCO(object):
         res.createIndex()
         return res
 
-    def download(self, target_dir=None, img_ids=None):
-        """No-op: this environment has no network. Kept for interface parity
-        with reference coco.py:292-314."""
-        return
+    def download(self, target_dir=None, img_ids=None, retries=2):
+        """Download missing images by their 'coco_url' (reference
+        coco.py:292-314 semantics: skip files already on disk, fetch the
+        rest).  Adds bounded retries — the reference has none — and
+        keeps going past individual failures so an offline run degrades
+        to the reference's behavior (missing files surface at load
+        time) instead of crashing data prep.
+
+        Returns (downloaded, failed) counts.  file:// URLs work, which
+        is how the offline tests exercise the path."""
+        import time
+        import urllib.request
+
+        if target_dir is None:
+            print('Please specify target directory')
+            return -1
+        imgs = (list(self.imgs.values()) if not img_ids
+                else self.loadImgs(img_ids))
+        os.makedirs(target_dir, exist_ok=True)
+        done = failed = 0
+        t0 = time.time()
+        for img in imgs:
+            url = img.get('coco_url') or img.get('url')
+            if not url:
+                continue
+            fname = os.path.join(target_dir, img['file_name'])
+            if os.path.exists(fname):
+                continue
+            ok = False
+            for _ in range(retries + 1):
+                try:
+                    urllib.request.urlretrieve(url, fname)
+                    ok = True
+                    break
+                except Exception:
+                    if os.path.exists(fname):
+                        os.remove(fname)
+            done += ok
+            failed += not ok
+        if done or failed:
+            print('downloaded %d images, %d failed (t=%.1fs)'
+                  % (done, failed, time.time() - t0))
+        return done, failed

@@ -55,3 +55,47 @@ def test_load_res(tmp_path):
     rf.write_text(json.dumps(results))
     res = c.loadRes(str(rf))
     assert len(res.imgToAnns) == 2
+
+
+def test_download_fetches_missing_images(tmp_path):
+    """download() (reference coco.py:292-314 parity): fetches by
+    coco_url, skips files already present, survives failures.  Tested
+    offline via file:// URLs."""
+    import json
+    from sat_amd.data.coco import COCO
+
+    src = tmp_path / 'src'
+    src.mkdir()
+    (src / 'img1.jpg').write_bytes(b'JPG1')
+    (src / 'img2.jpg').write_bytes(b'JPG2')
+
+    ann = {
+        'images': [
+            {'id': 1, 'file_name': 'img1.jpg',
+             'coco_url': 'file://%s' % (src / 'img1.jpg')},
+            {'id': 2, 'file_name': 'img2.jpg',
+             'coco_url': 'file://%s' % (src / 'img2.jpg')},
+            {'id': 3, 'file_name': 'img3.jpg',
+             'coco_url': 'file://%s/does_not_exist.jpg' % src},
+        ],
+        'annotations': [
+            {'id': 10, 'image_id': 1, 'caption': 'a cat .'},
+            {'id': 11, 'image_id': 2, 'caption': 'a dog .'},
+            {'id': 12, 'image_id': 3, 'caption': 'a bird .'},
+        ],
+        'type': 'captions',
+    }
+    af = tmp_path / 'anns.json'
+    af.write_text(json.dumps(ann))
+
+    tgt = tmp_path / 'imgs'
+    tgt.mkdir()
+    (tgt / 'img2.jpg').write_bytes(b'ALREADY_HERE')
+
+    coco = COCO(str(af))
+    done, failed = coco.download(str(tgt), retries=0)
+    assert done == 1 and failed == 1
+    assert (tgt / 'img1.jpg').read_bytes() == b'JPG1'
+    # pre-existing file untouched (reference skips existing)
+    assert (tgt / 'img2.jpg').read_bytes() == b'ALREADY_HERE'
+    assert not (tgt / 'img3.jpg').exists()
