@@ -143,6 +143,14 @@ def prepare_train_data(config):
     print("Captions processed.")
     print("Number of captions = %d" % len(captions))
 
+    # fetch any images the annotation set references but the disk lacks
+    # (reference dataset.py:157 -> coco.download); offline this degrades
+    # to a counted-failure no-op
+    missing = [int(i) for i, f in zip(image_ids, image_files)
+               if not os.path.exists(f)]
+    if missing:
+        coco.download(config.train_image_dir, list(set(missing)))
+
     dataset = DataSet(image_ids, image_files, config.batch_size,
                       word_idxs, masks, True, True)
     return dataset
