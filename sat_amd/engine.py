@@ -35,9 +35,14 @@ class GraphedTrainStep(object):
 
     def _inner(self, images, sentences, masks):
         out = self.model(images, sentences, masks)
-        for p in self.optimizer.params:
-            if p.grad is not None:
-                p.grad.zero_()
+        opt = self.optimizer
+        fused = (opt.kind == 'Adam' and len(opt.params) > 0
+                 and opt.params[0].is_cuda)
+        if not fused:
+            # the fused Adam zeroes grads inside its own update kernel
+            for p in opt.params:
+                if p.grad is not None:
+                    p.grad.zero_()
         out['total_loss'].backward()
         if self.ddp is not None:
             self.ddp.finish_backward()

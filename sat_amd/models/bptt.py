@@ -205,9 +205,9 @@ class DecoderCoreBPTT(torch.autograd.Function):
             dt1, dt2f, _dv = _C.attn_scores_bwd_acc(
                 tdrops[t], v, dlog_att, seed, p_fc, s + 2, L, dv_acc)
 
-            dpre1b = _C.act_bwd(dt2f.to(torch.bfloat16), t2s[t], ACT_TANH)
-            DPRE1B[sl] = dpre1b
-            dodrop = _C.dense_fwd(dpre1b, w1b_t, _EMPTY_B(dev), ACT_NONE)
+            _C.act_bwd_f32_out(dt2f, t2s[t], ACT_TANH, DPRE1B[sl])
+            dodrop = _C.dense_fwd(DPRE1B[sl], w1b_t, _EMPTY_B(dev),
+                                  ACT_NONE)
             d_out_carry = _drop(dodrop, seed, p_fc, s + 1)
 
             sl_a = slice(t * B * L, (t + 1) * B * L)

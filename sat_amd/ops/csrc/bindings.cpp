@@ -63,6 +63,8 @@ std::vector<at::Tensor> lstm_pointwise_bwd(at::Tensor gates, at::Tensor c,
                                            at::Tensor dh, at::Tensor dc,
                                            double fb);
 at::Tensor act_bwd(at::Tensor dy, at::Tensor y, int64_t act);
+void act_bwd_f32_out(at::Tensor dy, at::Tensor y, int64_t act,
+                     at::Tensor out);
 at::Tensor hash_dropout(at::Tensor x, at::Tensor seed, double p,
                         int64_t salt);
 std::vector<at::Tensor> attn_pool_fwd(at::Tensor ctx, at::Tensor logits);
@@ -92,7 +94,8 @@ at::Tensor sq_norm_mt(at::Tensor desc, at::Tensor cum, int64_t n_tensors,
 void adam_step_mt(at::Tensor desc, at::Tensor cum, int64_t n_tensors,
                   int64_t total, at::Tensor step_dev, double lr0,
                   double decay_factor, double steps_per_decay, double b1,
-                  double b2, double eps, double clip, at::Tensor gsq);
+                  double b2, double eps, double clip, at::Tensor gsq,
+                  bool zero_grads);
 void adam_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
                std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
                at::Tensor step_dev, double lr0, double decay_factor,
@@ -128,6 +131,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("lstm_pointwise_fwd", &lstm_pointwise_fwd);
     m.def("lstm_pointwise_bwd", &lstm_pointwise_bwd);
     m.def("act_bwd", &act_bwd);
+    m.def("act_bwd_f32_out", &act_bwd_f32_out);
     m.def("hash_dropout", &hash_dropout);
     m.def("attn_pool_fwd", &attn_pool_fwd);
     m.def("attn_scores_fused", &attn_scores_fused);

@@ -341,6 +341,34 @@ void act_bwd_out(at::Tensor dy, at::Tensor y, int64_t act, at::Tensor out) {
     HIP_OK(hipGetLastError());
 }
 
+// fp32-dy variant writing into a preallocated bf16 slab: replaces the
+// per-step  dt2f.to(bf16) -> act_bwd -> DPRE1B[sl].copy_  three-kernel
+// chain in the core BPTT backward with ONE launch
+__global__ void act_bwd_f32_out_kernel(const float* __restrict__ dy,
+                                       const bf16* __restrict__ y,
+                                       bf16* __restrict__ dpre,
+                                       int64_t n, int act) {
+    int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= n) return;
+    float g = dy[idx];
+    float yv = bf2f(y[idx]);
+    if (act == 1) g *= (1.f - yv * yv);
+    else if (act == 2) g *= (yv > 0.f ? 1.f : 0.f);
+    dpre[idx] = f2bf(g);
+}
+
+void act_bwd_f32_out(at::Tensor dy, at::Tensor y, int64_t act,
+                     at::Tensor out) {
+    int64_t n = dy.numel();
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(act_bwd_f32_out_kernel, dim3(cdiv(n, 256)),
+                       dim3(256), 0, s,
+                       (const float*)dy.data_ptr(),
+                       (const bf16*)y.data_ptr(),
+                       (bf16*)out.data_ptr(), n, (int)act);
+    HIP_OK(hipGetLastError());
+}
+
 // ---- dropout over a step-major [T*B, N] buffer where slab t uses salt
 // salt_base + t*salt_stride and slab-local flat indices (matches the
 // per-step hash_dropout calls it replaces) ----

@@ -911,7 +911,7 @@ __global__ void adam_mt_kernel(const int64_t* __restrict__ desc,
                                const float* __restrict__ step_dev,
                                float lr0, float decay_factor,
                                float steps_per_decay, float b1, float b2,
-                               float eps, float clip) {
+                               float eps, float clip, int zero_g) {
     extern __shared__ int64_t scum[];
     for (int i = threadIdx.x; i <= n_tensors; i += blockDim.x)
         scum[i] = cum[i];
@@ -937,7 +937,7 @@ __global__ void adam_mt_kernel(const int64_t* __restrict__ desc,
         }
         int64_t off = i - scum[lo];
         float* p = (float*)desc[lo * 4 + 0];
-        const float* g = (const float*)desc[lo * 4 + 1];
+        float* g = (float*)desc[lo * 4 + 1];
         float* m = (float*)desc[lo * 4 + 2];
         float* v = (float*)desc[lo * 4 + 3];
         float gi = g[off] * scale;
@@ -946,13 +946,19 @@ __global__ void adam_mt_kernel(const int64_t* __restrict__ desc,
         m[off] = mi;
         v[off] = vi;
         p[off] -= lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+        // fold next step's grad zeroing into the update pass: removes
+        // one fill launch per parameter tensor per step (autograd then
+        // accumulates into the zeroed buffer at fixed addresses, which
+        // also keeps hipGraph capture stable)
+        if (zero_g) g[off] = 0.f;
     }
 }
 
 void adam_step_mt(at::Tensor desc, at::Tensor cum, int64_t n_tensors,
                   int64_t total, at::Tensor step_dev, double lr0,
                   double decay_factor, double steps_per_decay, double b1,
-                  double b2, double eps, double clip, at::Tensor gsq) {
+                  double b2, double eps, double clip, at::Tensor gsq,
+                  bool zero_grads) {
     hipStream_t s = at::cuda::getCurrentCUDAStream();
     int blocks = (int)std::min<int64_t>(cdiv(total, 256 * 4), 4096);
     size_t shm = (n_tensors + 1) * sizeof(int64_t);
@@ -964,6 +970,6 @@ void adam_step_mt(at::Tensor desc, at::Tensor cum, int64_t n_tensors,
                        (const float*)step_dev.data_ptr(),
                        (float)lr0, (float)decay_factor,
                        (float)steps_per_decay, (float)b1, (float)b2,
-                       (float)eps, (float)clip);
+                       (float)eps, (float)clip, zero_grads ? 1 : 0);
     HIP_OK(hipGetLastError());
 }

@@ -109,13 +109,16 @@ class Optimizer(object):
                                                    tuple(p0.shape)))
             gsq = _C.sq_norm_mt(self._mt_desc, self._mt_cum,
                                 len(self.params), self._mt_total)
+            # zero_grads=True: the update kernel clears p.grad in the
+            # same pass, so the engine skips its per-tensor fill
+            # launches and autograd accumulates into stable addresses
             _C.adam_step_mt(
                 self._mt_desc, self._mt_cum, len(self.params),
                 self._mt_total, self.step_dev,
                 cfg.initial_learning_rate,
                 cfg.learning_rate_decay_factor, cfg.num_steps_per_decay,
                 cfg.beta1, cfg.beta2, cfg.epsilon,
-                cfg.clip_gradients, gsq)
+                cfg.clip_gradients, gsq, True)
             return
 
         # ---- eager path (CPU, or non-Adam optimizers) ----
