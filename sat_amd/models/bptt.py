@@ -206,9 +206,10 @@ class DecoderCoreBPTT(torch.autograd.Function):
                 tdrops[t], v, dlog_att, seed, p_fc, s + 2, L, dv_acc)
 
             _C.act_bwd_f32_out(dt2f, t2s[t], ACT_TANH, DPRE1B[sl])
-            dodrop = _C.dense_fwd(DPRE1B[sl], w1b_t, _EMPTY_B(dev),
-                                  ACT_NONE)
-            d_out_carry = _drop(dodrop, seed, p_fc, s + 1)
+            # dodrop GEMM with the ODROP mask (salt s+1) regenerated in
+            # the split-K epilogue — one launch instead of GEMM+dropout
+            d_out_carry = _C.dense_fwd_drop(DPRE1B[sl], w1b_t, seed,
+                                            p_fc, s + 1)
 
             sl_a = slice(t * B * L, (t + 1) * B * L)
             _C.act_bwd_out(dt1, t1s[t], ACT_TANH, DPRE1A[sl_a])

@@ -3,6 +3,8 @@
 #include <vector>
 
 at::Tensor dense_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, int64_t act);
+at::Tensor dense_fwd_drop(at::Tensor x, at::Tensor w, at::Tensor seed,
+                          double p, int64_t salt);
 at::Tensor conv3_fwd(at::Tensor input, at::Tensor weight, at::Tensor bias,
                      bool relu);
 void bias_act_nhwc(at::Tensor y, at::Tensor bias, bool relu);
@@ -104,6 +106,8 @@ void adam_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("dense_fwd", &dense_fwd, "MFMA GEMM + bias/act (bf16)");
+    m.def("dense_fwd_drop", &dense_fwd_drop,
+          "skinny GEMM with hash-dropout fused into the split-K epilogue");
     m.def("bias_act_nhwc", &bias_act_nhwc);
     m.def("scale_bias_act_nhwc", &scale_bias_act_nhwc);
     m.def("maxpool2x2_nhwc", &maxpool2x2_nhwc);

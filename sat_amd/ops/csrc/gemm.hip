@@ -383,6 +383,78 @@ at::Tensor dense_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
     return dense_fwd_out(x, w, bias, act, at::Tensor());
 }
 
+// ---- GEMM + counter-hash dropout fused into the split-K epilogue ----
+// (same hash as hash_dropout_kernel — backward/forward mask pairs must
+// regenerate identically)
+
+__device__ __forceinline__ uint32_t mix3g(uint32_t a, uint32_t b,
+                                          uint32_t c) {
+    uint32_t h = a * 0x9E3779B1u ^ b * 0x85EBCA77u ^ c * 0xC2B2AE3Du;
+    h ^= h >> 16; h *= 0x7FEB352Du;
+    h ^= h >> 15; h *= 0x846CA68Bu;
+    h ^= h >> 16;
+    return h;
+}
+
+__device__ __forceinline__ float dscaleg(uint32_t seed, int salt,
+                                         uint32_t idx, float p) {
+    if (p <= 0.f) return 1.f;
+    uint32_t h = mix3g(seed, (uint32_t)salt, idx);
+    float u = (h >> 8) * (1.0f / 16777216.0f);
+    return u >= p ? 1.0f / (1.0f - p) : 0.0f;
+}
+
+__global__ void skinny_epilogue_drop_kernel(
+        const float* __restrict__ Yf, bf16* __restrict__ Y,
+        const int64_t* __restrict__ seed_p, int64_t n, int splitk,
+        float p, int salt) {
+    int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= n) return;
+    float v = 0.f;
+    for (int k = 0; k < splitk; ++k) v += Yf[k * n + idx];
+    v *= dscaleg((uint32_t)(*seed_p), salt, (uint32_t)idx, p);
+    Y[idx] = f2bf(v);
+}
+
+at::Tensor dense_fwd_drop(at::Tensor x, at::Tensor w, at::Tensor seed,
+                          double p, int64_t salt) {
+    // bias-free, activation-free skinny GEMM whose split-K epilogue
+    // applies hash dropout (salt/index convention of hash_dropout) —
+    // removes one elementwise launch per decoder reverse step
+    CHECK_GPU(x); CHECK_CONTIG(x); CHECK_BF16(x);
+    CHECK_GPU(w); CHECK_CONTIG(w); CHECK_BF16(w);
+    int64_t M = x.size(0), K = x.size(1), N = w.size(0);
+    hipStream_t stream = at::cuda::getCurrentCUDAStream();
+    if (p <= 0.0 || M > 128 || K % 32 != 0) {
+        // non-skinny shapes: plain GEMM + the separate dropout pass
+        // (defined in kernels.hip; same hash)
+        extern at::Tensor hash_dropout(at::Tensor, at::Tensor, double,
+                                       int64_t);
+        auto y = dense_fwd_out(x, w, at::Tensor(), 0, at::Tensor());
+        return p > 0.0 ? hash_dropout(y, seed, p, salt) : y;
+    }
+    auto y = at::empty({M, N}, x.options());
+    int nblocks = cdiv(N, 64);
+    int splitk = 1;
+    while (nblocks * splitk < 192 && splitk < 8 &&
+           (int)(K / 32) >= 2 * splitk)
+        splitk *= 2;
+    auto yf = at::empty({splitk, M, N}, x.options().dtype(at::kFloat));
+#define LAUNCH_SK(RF)     hipLaunchKernelGGL((skinny_gemm_kernel<RF>), dim3(nblocks, splitk),                        dim3(256), 0, stream, (const bf16*)x.data_ptr(),                        (const bf16*)w.data_ptr(), nullptr, nullptr,                        (float*)yf.data_ptr(), nullptr, (int)M, (int)N,                        (int)K, 0, splitk)
+    if (M <= 32) LAUNCH_SK(2);
+    else if (M <= 64) LAUNCH_SK(4);
+    else LAUNCH_SK(8);
+#undef LAUNCH_SK
+    int64_t n = M * N;
+    hipLaunchKernelGGL(skinny_epilogue_drop_kernel,
+                       dim3(cdiv(n, 256)), dim3(256), 0, stream,
+                       (const float*)yf.data_ptr(), (bf16*)y.data_ptr(),
+                       (const int64_t*)seed.data_ptr(), n, splitk,
+                       (float)p, (int)salt);
+    HIP_OK(hipGetLastError());
+    return y;
+}
+
 // ---- fused skinny epilogues for the BPTT step chain (B <= 32) ----
 
 // gates epilogue + LSTM pointwise: thread per (b,h) sums the 4 gate

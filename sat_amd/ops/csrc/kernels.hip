@@ -367,6 +367,67 @@ __global__ void attn_pool_bwd_p2_kernel(
     }
 }
 
+// both phases in one block per image: per-row dots s_l (4 waves strided
+// over L) land in LDS, then the softmax-Jacobian combine — removes the
+// p2 launch and the sbuf HBM round trip from the reverse-loop chain
+__global__ void attn_pool_bwd_fused_kernel(
+        const bf16* __restrict__ ctx, const float* __restrict__ alpha,
+        const float* __restrict__ dalpha,
+        const bf16* __restrict__ dpooled, bf16* __restrict__ dctx,
+        float* __restrict__ dlogits, int L, int D) {
+    __shared__ float sL[MAX_L];
+    __shared__ float red[4];
+    int b = blockIdx.x;
+    int tid = threadIdx.x;
+    int wid = tid >> 6, lane = tid & 63;
+    const bf16* cb = ctx + (int64_t)b * L * D;
+    const bf16* dp = dpooled + (int64_t)b * D;
+
+    for (int l = wid; l < L; l += 4) {
+        float acc = 0.f;
+        float a = alpha[(int64_t)b * L + l];
+        for (int d0 = lane * 8; d0 + 8 <= D; d0 += 64 * 8) {
+            bf16x8 cv = *(const bf16x8*)(cb + (int64_t)l * D + d0);
+            bf16x8 dv = *(const bf16x8*)(dp + d0);
+            if (dctx != nullptr) {
+                bf16x8 out;
+#pragma unroll
+                for (int e = 0; e < 8; ++e) {
+                    float dpe = bf2f(dv[e]);
+                    acc += bf2f(cv[e]) * dpe;
+                    out[e] = f2bf(a * dpe);
+                }
+                *(bf16x8*)(dctx + (int64_t)b * L * D + (int64_t)l * D
+                           + d0) = out;
+            } else {
+#pragma unroll
+                for (int e = 0; e < 8; ++e)
+                    acc += bf2f(cv[e]) * bf2f(dv[e]);
+            }
+        }
+        acc = wave_sum(acc);
+        if (lane == 0) sL[l] = acc;
+    }
+    __syncthreads();
+
+    float part = 0.f;
+    for (int l = tid; l < L; l += blockDim.x) {
+        float da = sL[l];
+        if (dalpha != nullptr) da += dalpha[(int64_t)b * L + l];
+        part += alpha[(int64_t)b * L + l] * da;
+    }
+    part = wave_sum(part);
+    if ((tid & 63) == 0) red[tid >> 6] = part;
+    __syncthreads();
+    float dot = red[0] + red[1] + red[2] + red[3];
+    for (int l = tid; l < L; l += blockDim.x) {
+        float da = sL[l];
+        if (dalpha != nullptr) da += dalpha[(int64_t)b * L + l];
+        dlogits[(int64_t)b * L + l] =
+            alpha[(int64_t)b * L + l] * (da - dot);
+    }
+}
+
 std::vector<at::Tensor> attn_pool_bwd(at::Tensor ctx, at::Tensor alpha,
                                       at::Tensor dalpha, at::Tensor dpooled,
                                       bool need_dctx) {
@@ -375,7 +436,6 @@ std::vector<at::Tensor> attn_pool_bwd(at::Tensor ctx, at::Tensor alpha,
     int B = ctx.size(0), L = ctx.size(1), D = ctx.size(2);
     TORCH_CHECK(L <= MAX_L && D % 8 == 0);
     auto dlogits = at::empty({B, L}, alpha.options());
-    auto sbuf = at::empty({B, L}, alpha.options());
     at::Tensor dctx;
     bf16* dctx_ptr = nullptr;
     if (need_dctx) {
@@ -388,15 +448,12 @@ std::vector<at::Tensor> attn_pool_bwd(at::Tensor ctx, at::Tensor alpha,
     if (dalpha.defined() && dalpha.numel() > 0)
         dalpha_ptr = (const float*)dalpha.data_ptr();
     hipStream_t s = at::cuda::getCurrentCUDAStream();
-    hipLaunchKernelGGL(attn_pool_bwd_p1_kernel, dim3(B, 4), dim3(256), 0, s,
+    hipLaunchKernelGGL(attn_pool_bwd_fused_kernel, dim3(B), dim3(256),
+                       0, s,
                        (const bf16*)ctx.data_ptr(),
-                       (const float*)alpha.data_ptr(),
-                       (const bf16*)dpooled.data_ptr(),
-                       (float*)sbuf.data_ptr(), dctx_ptr, L, D);
-    hipLaunchKernelGGL(attn_pool_bwd_p2_kernel, dim3(B), dim3(256), 0, s,
                        (const float*)alpha.data_ptr(), dalpha_ptr,
-                       (const float*)sbuf.data_ptr(),
-                       (float*)dlogits.data_ptr(), L);
+                       (const bf16*)dpooled.data_ptr(), dctx_ptr,
+                       (float*)dlogits.data_ptr(), L, D);
     HIP_OK(hipGetLastError());
     return {dlogits, dctx};
 }
@@ -591,6 +648,82 @@ __global__ void attn_pool_sum_kernel(const bf16* __restrict__ ctx,
     }
 }
 
+// fused softmax + weighted pool: each (b, d-chunk) block recomputes the
+// L-wide softmax in LDS (cheap: L <= 1024 floats) and pools its 128
+// columns — one launch instead of softmax + pool, no alpha round-trip
+// on the forward critical path (alpha is still written once, by the
+// chunk-0 block, for backward/attention-loss use).
+__global__ void attn_pool_fused_kernel(const bf16* __restrict__ ctx,
+                                       const float* __restrict__ logits,
+                                       float* __restrict__ alpha,
+                                       bf16* __restrict__ pooled,
+                                       int L, int D) {
+    __shared__ float sa[MAX_L];
+    __shared__ float red[8];
+    __shared__ float part[16][128 + 4];
+    int b = blockIdx.x;
+    int d0 = blockIdx.y * 128;
+    int tid = threadIdx.x;
+
+    // softmax over L into sa (same math as attn_softmax_kernel)
+    float lmax = -1e30f;
+    for (int l = tid; l < L; l += blockDim.x) {
+        float x = logits[(int64_t)b * L + l];
+        sa[l] = x;
+        lmax = fmaxf(lmax, x);
+    }
+    lmax = wave_max(lmax);
+    if ((tid & 63) == 0) red[tid >> 6] = lmax;
+    __syncthreads();
+    float m = -1e30f;
+    for (int w = 0; w < (int)(blockDim.x >> 6); ++w)
+        m = fmaxf(m, red[w]);
+    __syncthreads();
+    float lsum = 0.f;
+    for (int l = tid; l < L; l += blockDim.x) {
+        float e = __expf(sa[l] - m);
+        sa[l] = e;
+        lsum += e;
+    }
+    lsum = wave_sum(lsum);
+    if ((tid & 63) == 0) red[tid >> 6] = lsum;
+    __syncthreads();
+    float z = 0.f;
+    for (int w = 0; w < (int)(blockDim.x >> 6); ++w) z += red[w];
+    float inv = 1.0f / z;
+    __syncthreads();
+    for (int l = tid; l < L; l += blockDim.x) {
+        sa[l] *= inv;
+        if (blockIdx.y == 0)
+            alpha[(int64_t)b * L + l] = sa[l];
+    }
+    __syncthreads();
+
+    // pool this block's 128 columns
+    int dg = tid & 15;
+    int lg = tid >> 4;
+    int d = d0 + dg * 8;
+    float acc[8] = {};
+    if (d < D) {
+        const bf16* cb = ctx + (int64_t)b * L * D;
+        for (int l = lg; l < L; l += 16) {
+            bf16x8 cv = *(const bf16x8*)(cb + (int64_t)l * D + d);
+            float a = sa[l];
+#pragma unroll
+            for (int e = 0; e < 8; ++e) acc[e] += a * bf2f(cv[e]);
+        }
+    }
+#pragma unroll
+    for (int e = 0; e < 8; ++e) part[lg][dg * 8 + e] = acc[e];
+    __syncthreads();
+    if (tid < 128 && d0 + tid < D) {
+        float ssum = 0.f;
+#pragma unroll
+        for (int g = 0; g < 16; ++g) ssum += part[g][tid];
+        pooled[(int64_t)b * D + d0 + tid] = f2bf(ssum);
+    }
+}
+
 std::vector<at::Tensor> attn_pool_fwd(at::Tensor ctx, at::Tensor logits) {
     CHECK_GPU(ctx); CHECK_CONTIG(ctx); CHECK_BF16(ctx);
     CHECK_GPU(logits); CHECK_CONTIG(logits); CHECK_F32(logits);
@@ -599,13 +732,11 @@ std::vector<at::Tensor> attn_pool_fwd(at::Tensor ctx, at::Tensor logits) {
     auto alpha = at::empty({B, L}, ctx.options().dtype(at::kFloat));
     auto pooled = at::empty({B, D}, ctx.options());
     hipStream_t s = at::cuda::getCurrentCUDAStream();
-    hipLaunchKernelGGL(attn_softmax_kernel, dim3(B), dim3(256), 0, s,
-                       (const float*)logits.data_ptr(),
-                       (float*)alpha.data_ptr(), L);
-    hipLaunchKernelGGL(attn_pool_sum_kernel, dim3(B, cdiv(D, 128)),
+    hipLaunchKernelGGL(attn_pool_fused_kernel, dim3(B, cdiv(D, 128)),
                        dim3(256), 0, s,
                        (const bf16*)ctx.data_ptr(),
-                       (const float*)alpha.data_ptr(),
+                       (const float*)logits.data_ptr(),
+                       (float*)alpha.data_ptr(),
                        (bf16*)pooled.data_ptr(), L, D);
     HIP_OK(hipGetLastError());
     return {alpha, pooled};
