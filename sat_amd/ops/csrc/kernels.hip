@@ -436,6 +436,7 @@ std::vector<at::Tensor> attn_pool_bwd(at::Tensor ctx, at::Tensor alpha,
     int B = ctx.size(0), L = ctx.size(1), D = ctx.size(2);
     TORCH_CHECK(L <= MAX_L && D % 8 == 0);
     auto dlogits = at::empty({B, L}, alpha.options());
+    auto sbuf = at::empty({B, L}, alpha.options());
     at::Tensor dctx;
     bf16* dctx_ptr = nullptr;
     if (need_dctx) {
@@ -448,12 +449,19 @@ std::vector<at::Tensor> attn_pool_bwd(at::Tensor ctx, at::Tensor alpha,
     if (dalpha.defined() && dalpha.numel() > 0)
         dalpha_ptr = (const float*)dalpha.data_ptr();
     hipStream_t s = at::cuda::getCurrentCUDAStream();
-    hipLaunchKernelGGL(attn_pool_bwd_fused_kernel, dim3(B), dim3(256),
-                       0, s,
+    // NOTE: a one-block-per-image fused p1+p2 variant
+    // (attn_pool_bwd_fused_kernel above) measured SLOWER at batch 32 —
+    // it cuts the p1 grid from B*4 to B blocks on a latency-bound
+    // chain; the two-launch form stays (r02 A/B evidence).
+    hipLaunchKernelGGL(attn_pool_bwd_p1_kernel, dim3(B, 4), dim3(256), 0, s,
                        (const bf16*)ctx.data_ptr(),
+                       (const float*)alpha.data_ptr(),
+                       (const bf16*)dpooled.data_ptr(),
+                       (float*)sbuf.data_ptr(), dctx_ptr, L, D);
+    hipLaunchKernelGGL(attn_pool_bwd_p2_kernel, dim3(B), dim3(256), 0, s,
                        (const float*)alpha.data_ptr(), dalpha_ptr,
-                       (const bf16*)dpooled.data_ptr(), dctx_ptr,
-                       (float*)dlogits.data_ptr(), L, D);
+                       (const float*)sbuf.data_ptr(),
+                       (float*)dlogits.data_ptr(), L);
     HIP_OK(hipGetLastError());
     return {dlogits, dctx};
 }
