@@ -17,6 +17,9 @@ at::Tensor pad1_nhwc(at::Tensor input);
 at::Tensor conv_igemm_8p_fwd(at::Tensor padded, at::Tensor w_ohwi,
                              at::Tensor bias, int64_t Hh, int64_t Ww,
                              bool relu);
+at::Tensor conv_igemm_glds64_fwd(at::Tensor padded, at::Tensor w_ohwi,
+                                 at::Tensor bias, int64_t Hh, int64_t Ww,
+                                 bool relu);
 at::Tensor conv3x3_wgrad(at::Tensor xpad, at::Tensor dy_rows,
                          int64_t Hh, int64_t Ww);
 at::Tensor conv_igemm_glds_fwd(at::Tensor padded, at::Tensor w_ohwi,
@@ -116,6 +119,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("conv_igemm_glds_fwd", &conv_igemm_glds_fwd);
     m.def("conv_igemm_8p_fwd", &conv_igemm_8p_fwd,
           "8-phase deep-pipelined implicit-GEMM conv (Cout%256==0)");
+    m.def("conv_igemm_glds64_fwd", &conv_igemm_glds64_fwd);
     m.def("conv3x3_wgrad", &conv3x3_wgrad,
           "3x3 conv weight grad: transpose-staged MFMA split-K reduce");
     m.def("dense_lstm_fwd", &dense_lstm_fwd);

@@ -628,3 +628,167 @@ at::Tensor conv_igemm_glds_fwd(at::Tensor padded, at::Tensor w_ohwi,
     HIP_OK(hipGetLastError());
     return out;
 }
+
+// ---- BN=64 glds variant for Cout=64 3x3 layers (VGG conv1_2) ----
+// Same structure as conv_igemm_glds_kernel but a 128x64 tile with 4
+// waves stacked on M (per-wave 32x64 output): the register-staged
+// <4,1,2,4> kernel measured 350 TF on conv1_2 while the glds-staged
+// 128x128 class reaches 536-658 TF; Cout=64 layers could not use it.
+
+__global__ __launch_bounds__(256)
+void conv_igemm_glds64_kernel(const bf16* __restrict__ inp, // padded
+                              const bf16* __restrict__ w,   // [Cout,9Ci]
+                              const bf16* __restrict__ bias,
+                              bf16* __restrict__ out,
+                              int M, int Hh, int Ww, int Cin, int Cout,
+                              int relu) {
+    constexpr int BM = 128, BKc = 64;
+    __shared__ bf16 lds[(BM + 64) * BKc];   // A[128][64] + B[64][64]
+    bf16* As = lds;
+    bf16* Bs = lds + BM * BKc;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;              // 0..3, all stacked on M
+    const int bm = blockIdx.y * BM;
+    const int bn = blockIdx.x * 64;
+    const int Wp = Ww + 2;
+
+    // staging rows: this wave stages A rows wave*32 + i*8 + (lane>>3)
+    // (i = 0..3) and B rows wave*16 + j*8 + (lane>>3) (j = 0..1)
+    int aBase[4];
+    int rowA[4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+        int r = wave * 32 + i * 8 + (lane >> 3);
+        rowA[i] = r;
+        int m = bm + r;
+        if (m >= M) m = M - 1;
+        int b = m / (Hh * Ww);
+        int yx = m % (Hh * Ww);
+        int y = yx / Ww, x = yx % Ww;
+        aBase[i] = ((b * (Hh + 2) + y + 1) * Wp + x + 1) * Cin;
+    }
+    const bf16* wBase[2];
+    int rowB[2];
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+        int r = wave * 16 + j * 8 + (lane >> 3);
+        rowB[j] = r;
+        int cout = bn + r;
+        if (cout >= Cout) cout = Cout - 1;
+        wBase[j] = w + (int64_t)cout * 9 * Cin;
+    }
+    const int ci8 = (lane & 7) * 8;
+
+    floatx4 acc[2][4];
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+            acc[i][j] = floatx4{0.f, 0.f, 0.f, 0.f};
+
+    const int lrow = lane & 15;
+    const int kgrp = lane >> 4;
+    const int KT = 9 * (Cin / BKc);
+
+    for (int kt = 0; kt < KT; ++kt) {
+        const int dxy = kt / (Cin / BKc);
+        const int ci0 = (kt % (Cin / BKc)) * BKc;
+        const int dy = dxy / 3 - 1, dx = dxy % 3 - 1;
+        const int delta = (dy * Wp + dx) * Cin + ci0 + ci8;
+
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            int swz = (rowA[i] & 4) ? 16 : 0;
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) uint32_t*)
+                    (inp + aBase[i] + delta + (ci8 ^ swz) - ci8),
+                (__attribute__((address_space(3))) uint32_t*)
+                    (As + (wave * 32 + i * 8) * BKc),
+                16, 0, 0);
+        }
+#pragma unroll
+        for (int j = 0; j < 2; ++j) {
+            int swz = (rowB[j] & 4) ? 16 : 0;
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) uint32_t*)
+                    (wBase[j] + dxy * Cin + ci0 + (ci8 ^ swz)),
+                (__attribute__((address_space(3))) uint32_t*)
+                    (Bs + (wave * 16 + j * 8) * BKc),
+                16, 0, 0);
+        }
+        __syncthreads();
+
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk) {
+            bf16x8 a_frag[2], b_frag[4];
+            const int kof = kk * 32 + kgrp * 8;
+#pragma unroll
+            for (int mi = 0; mi < 2; ++mi) {
+                int r = wave * 32 + mi * 16 + lrow;
+                a_frag[mi] = *(const bf16x8*)(
+                    As + r * BKc + (kof ^ ((r & 4) ? 16 : 0)));
+            }
+#pragma unroll
+            for (int ni = 0; ni < 4; ++ni) {
+                int r = ni * 16 + lrow;
+                b_frag[ni] = *(const bf16x8*)(
+                    Bs + r * BKc + (kof ^ ((r & 4) ? 16 : 0)));
+            }
+#pragma unroll
+            for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+                for (int ni = 0; ni < 4; ++ni)
+                    acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a_frag[mi], b_frag[ni], acc[mi][ni], 0, 0, 0);
+        }
+        __syncthreads();
+    }
+
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+        int col = bn + ni * 16 + (lane & 15);
+        float bv = (bias != nullptr && col < Cout) ? bf2f(bias[col]) : 0.f;
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = bm + wave * 32 + mi * 16 + (lane >> 4) * 4 + r;
+                if (row < M && col < Cout) {
+                    float vv = acc[mi][ni][r] + bv;
+                    if (relu) vv = fmaxf(vv, 0.f);
+                    out[(int64_t)row * Cout + col] = f2bf(vv);
+                }
+            }
+        }
+    }
+}
+
+at::Tensor conv_igemm_glds64_fwd(at::Tensor padded, at::Tensor w_ohwi,
+                                 at::Tensor bias, int64_t Hh, int64_t Ww,
+                                 bool relu) {
+    CHECK_GPU(padded); CHECK_BF16(padded);
+    CHECK_GPU(w_ohwi); CHECK_CONTIG(w_ohwi); CHECK_BF16(w_ohwi);
+    TORCH_CHECK(padded.is_contiguous(at::MemoryFormat::ChannelsLast));
+    int B = padded.size(0), Cin = padded.size(1);
+    TORCH_CHECK(padded.size(2) == Hh + 2 && padded.size(3) == Ww + 2);
+    int Cout = w_ohwi.size(0);
+    TORCH_CHECK(Cin % 64 == 0 && Cout % 8 == 0);
+    auto out = at::empty({B, Cout, Hh, Ww},
+                         padded.options()
+                             .memory_format(at::MemoryFormat::ChannelsLast));
+    const bf16* bias_ptr = nullptr;
+    if (bias.defined() && bias.numel() > 0)
+        bias_ptr = (const bf16*)bias.contiguous().data_ptr();
+    int M = B * Hh * Ww;
+    dim3 grid(cdiv(Cout, 64), cdiv(M, 128));
+    hipStream_t s = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(conv_igemm_glds64_kernel, grid, dim3(256), 0, s,
+                       (const bf16*)padded.data_ptr(),
+                       (const bf16*)w_ohwi.data_ptr(), bias_ptr,
+                       (bf16*)out.data_ptr(), (int)M, (int)Hh, (int)Ww,
+                       (int)Cin, (int)Cout, relu ? 1 : 0);
+    HIP_OK(hipGetLastError());
+    return out;
+}

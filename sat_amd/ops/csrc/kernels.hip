@@ -453,7 +453,7 @@ std::vector<at::Tensor> attn_pool_bwd(at::Tensor ctx, at::Tensor alpha,
     // (attn_pool_bwd_fused_kernel above) measured SLOWER at batch 32 —
     // it cuts the p1 grid from B*4 to B blocks on a latency-bound
     // chain; the two-launch form stays (r02 A/B evidence).
-    hipLaunchKernelGGL(attn_pool_bwd_p1_kernel, dim3(B, 4), dim3(256), 0, s,
+    hipLaunchKernelGGL(attn_pool_bwd_p1_kernel, dim3(B, 8), dim3(256), 0, s,
                        (const bf16*)ctx.data_ptr(),
                        (const float*)alpha.data_ptr(),
                        (const bf16*)dpooled.data_ptr(),
@@ -550,7 +550,9 @@ std::vector<at::Tensor> attn_scores_bwd_acc(at::Tensor tdrop, at::Tensor v,
         dvf = dv_acc;   // caller-owned accumulator (atomicAdd accumulates)
     else
         dvf = at::zeros({A}, tdrop.options().dtype(at::kFloat));
-    int lchunk = ((int)L + 3) / 4;     // 4 chunks/image
+    int lchunk = ((int)L + 7) / 8;     // 8 chunks/image
+    // (4 -> 8 chunks: 256 blocks at batch 32 — the kernel is
+    // latency-bound at 12us, more blocks shorten the chain)
     int nchunk = ((int)L + lchunk - 1) / lchunk;
     hipStream_t s = at::cuda::getCurrentCUDAStream();
 #define LAUNCH_SB(NCH) \

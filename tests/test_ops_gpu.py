@@ -458,3 +458,20 @@ def test_conv3x3_train_function_grads():
         assert _rel_err(x.grad, xr.grad) < 3e-2, (Cin, Cout, relu)
         assert _rel_err(w.grad, wr.grad) < 3e-2, (Cin, Cout, relu)
         assert _rel_err(b.grad, br.grad) < 3e-2, (Cin, Cout, relu)
+
+
+def test_conv_igemm_glds64_matches_torch():
+    from sat_amd import _C
+    torch.manual_seed(15)
+    for Cin, Cout, B, H, W in [(64, 64, 2, 19, 17), (128, 64, 2, 14, 14),
+                               (64, 128, 1, 23, 23)]:
+        x = torch.randn(B, Cin, H, W).to(DEV, torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+        w = (torch.randn(Cout, Cin, 3, 3) * 0.05).to(DEV, torch.bfloat16)
+        b = torch.randn(Cout).to(DEV, torch.bfloat16)
+        w_ohwi = w.permute(0, 2, 3, 1).contiguous().reshape(Cout, -1)
+        xp = _C.pad1_nhwc(x)
+        y = _C.conv_igemm_glds64_fwd(xp, w_ohwi, b, H, W, True)
+        ref = torch.relu(torch.nn.functional.conv2d(
+            x.float(), w.float(), b.float(), padding=1))
+        assert _rel_err(y, ref) < 2e-2, (Cin, Cout, B, H, W)

@@ -95,3 +95,25 @@ for Cin, Cout, H in SHAPES:
     gf = 2.0 * B * H * H * Cout * Cin * 9 / 1e9
     print('Cin%4d Cout%4d H%4d  rel %.4f  8p %7.1fus (%5.0f TF) '
           '+pad %5.1fus' % (Cin, Cout, H, rel, t_8, gf / t_8 * 1e6, t_pad))
+
+print('== glds64 variant (padded, Cout=64 class) ==')
+for Cin, Cout, H in SHAPES:
+    if Cout > 128:
+        continue
+    x = torch.randn(B, Cin, H, H).to(DEV, torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    w = (torch.randn(Cout, Cin, 3, 3) * 0.02).to(DEV, torch.bfloat16)
+    bias = torch.randn(Cout).to(DEV, torch.bfloat16)
+    w_ohwi = w.permute(0, 2, 3, 1).contiguous().reshape(Cout, 9 * Cin)
+    xp = _C.pad1_nhwc(x)
+    y = _C.conv_igemm_glds64_fwd(xp, w_ohwi, bias, H, H, True)
+    ref = torch.relu(torch.nn.functional.conv2d(
+        x.float(), w.float(), bias.float(), padding=1))
+    rel = ((y.float() - ref).abs().max() /
+           ref.abs().max().clamp_min(1e-6)).item()
+    t_pad = timeit(lambda: _C.pad1_nhwc(x))
+    t_g = timeit(lambda: _C.conv_igemm_glds64_fwd(xp, w_ohwi, bias, H, H,
+                                                  True))
+    gf = 2.0 * B * H * H * Cout * Cin * 9 / 1e9
+    print('Cin%4d Cout%4d H%4d  rel %.4f  glds64 %7.1fus (%5.0f TF) '
+          '+pad %5.1fus' % (Cin, Cout, H, rel, t_g, gf / t_g * 1e6, t_pad))
