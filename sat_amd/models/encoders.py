@@ -48,6 +48,11 @@ class VGG16(tnn.Module):
                 layers.append(conv)
                 in_ch = out_ch
         self._layers = layers
+        # chained-pad fast path: conv1_1's direct kernel emits a 1-px
+        # zero-bordered output that conv1_2's glds igemm consumes
+        # without a pad pass (~70 us/step at 224^2 — models/nn.py)
+        self.conv1_1._emit_padded = True
+        self.conv1_2._accept_padded = True
 
     def forward(self, images):
         """images: [B,3,224,224] -> contexts [B,196,512]."""

@@ -158,6 +158,17 @@ class Conv2d(tnn.Module):
         w = self.weight.to(x.dtype)
         b = self.bias.to(x.dtype) if self.bias is not None else None
         k, st = self.kernel_size, self.stride
+        # chained-pad handshake: a producing conv may hand over a 1-px
+        # zero-bordered output (tagged `_sat_pad`); a 3x3/s1 consumer
+        # marked _accept_padded uses it directly, anyone else unpads
+        pad_hw = getattr(x, '_sat_pad', None)
+        if pad_hw is not None and not (
+                getattr(self, '_accept_padded', False)
+                and k == 3 and st == 1
+                and not torch.is_grad_enabled()):
+            x = x[:, :, 1:-1, 1:-1] \
+                .contiguous(memory_format=torch.channels_last)
+            pad_hw = None
         # --train_cnn: hand-written forward/dgrad/wgrad autograd triple
         # (reference model.py:505-511 backward surface on in-tree kernels)
         if (torch.is_grad_enabled()
@@ -182,11 +193,15 @@ class Conv2d(tnn.Module):
             from ..ops import hip
             if hip.available():
                 from sat_amd import _C
+                emit = bool(getattr(self, '_emit_padded', False)
+                            and self._glds_conv)
                 y = _C.conv3_fwd(
                     x, w.contiguous(),
                     b if b is not None
                     else torch.empty(0, dtype=x.dtype, device=x.device),
-                    self.activation == 'relu')
+                    self.activation == 'relu', emit)
+                if emit:
+                    y._sat_pad = (x.shape[2], x.shape[3])
                 return y
         # frozen GPU implicit-GEMM paths (per-shape winners,
         # profiles/r01_conv_shapes.txt + r02 8-phase kernel):
@@ -206,7 +221,9 @@ class Conv2d(tnn.Module):
             if hip.available():
                 from sat_amd import _C
                 Cout, Cin = w.shape[0], w.shape[1]
-                M = x.shape[0] * x.shape[2] * x.shape[3]
+                H = pad_hw[0] if pad_hw else x.shape[2]
+                W = pad_hw[1] if pad_hw else x.shape[3]
+                M = x.shape[0] * H * W
                 relu = self.activation == 'relu'
                 # 8p wins every Cout%256 shape with M>=12544; at
                 # conv5-batch-32 (M=6272) MIOpen gk still wins 77 vs
@@ -216,8 +233,14 @@ class Conv2d(tnn.Module):
                 use_glds = (self._glds_conv and not use_8p
                             and Cout >= 128 and Cout % 8 == 0
                             and Cin % 64 == 0 and Cin <= 128)
-                use_ig64 = (not use_8p and not use_glds and Cin == 64)
-                if use_8p or use_glds or use_ig64:
+                # Cout=64 class (conv1_2): glds64 525 TF vs the register
+                # igemm's 350 (profiles/r02_conv_shapes.txt)
+                use_g64 = (self._glds_conv and not use_8p
+                           and not use_glds and Cout < 128
+                           and Cin % 64 == 0)
+                use_ig64 = (not use_8p and not use_glds and not use_g64
+                            and Cin == 64 and pad_hw is None)
+                if use_8p or use_glds or use_g64 or use_ig64:
                     if getattr(self, '_w_ohwi', None) is None or \
                             self._w_ohwi_ver != self.weight._version:
                         self._w_ohwi_ver = self.weight._version
@@ -225,17 +248,22 @@ class Conv2d(tnn.Module):
                             .reshape(w.shape[0], -1)
                     eb = (b if b is not None else
                           torch.empty(0, dtype=x.dtype, device=x.device))
+                    if use_ig64:
+                        return _C.conv_igemm_fwd(x, self._w_ohwi, eb,
+                                                 relu)
+                    xp = x if pad_hw is not None else _C.pad1_nhwc(x)
                     if use_8p:
-                        xp = _C.pad1_nhwc(x)
                         return _C.conv_igemm_8p_fwd(
-                            xp, self._w_ohwi, eb, x.shape[2], x.shape[3],
-                            relu)
+                            xp, self._w_ohwi, eb, H, W, relu)
                     if use_glds:
-                        xp = _C.pad1_nhwc(x)
                         return _C.conv_igemm_glds_fwd(
-                            xp, self._w_ohwi, eb, x.shape[2], x.shape[3],
-                            relu)
-                    return _C.conv_igemm_fwd(x, self._w_ohwi, eb, relu)
+                            xp, self._w_ohwi, eb, H, W, relu)
+                    return _C.conv_igemm_glds64_fwd(
+                        xp, self._w_ohwi, eb, H, W, relu)
+        if pad_hw is not None:
+            # padded input fell through to a library path: unpad first
+            x = x[:, :, 1:-1, 1:-1] \
+                .contiguous(memory_format=torch.channels_last)
         # frozen GPU path: fused NHWC bias+ReLU kernel after the MIOpen
         # conv instead of two separate eager elementwise passes
         fuse_epi = (x.is_cuda and x.dtype == torch.bfloat16

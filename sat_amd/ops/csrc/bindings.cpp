@@ -6,7 +6,7 @@ at::Tensor dense_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, int64_t act);
 at::Tensor dense_fwd_drop(at::Tensor x, at::Tensor w, at::Tensor seed,
                           double p, int64_t salt);
 at::Tensor conv3_fwd(at::Tensor input, at::Tensor weight, at::Tensor bias,
-                     bool relu);
+                     bool relu, bool emit_pad);
 void bias_act_nhwc(at::Tensor y, at::Tensor bias, bool relu);
 void scale_bias_act_nhwc(at::Tensor y, at::Tensor scale, at::Tensor shift,
                          bool relu);

@@ -22,8 +22,10 @@ __global__ __launch_bounds__(256)
 void conv3_fwd_kernel(const bf16* __restrict__ in,   // [B,H,W,3]
                       const bf16* __restrict__ w,    // [Cout,3,3,3] OIHW
                       const bf16* __restrict__ bias, // [Cout]
-                      bf16* __restrict__ out,        // [B,H,W,Cout]
-                      int BHW, int Hh, int Ww, int Cout, int relu) {
+                      bf16* __restrict__ out,        // [B,H,W,Cout] or
+                                                     // padded variant
+                      int BHW, int Hh, int Ww, int Cout, int relu,
+                      int emit_pad) {
     __shared__ float ws[27 * MAX_COUT];   // ws[k*Cout + c], k = ci*9+ky*3+kx
     __shared__ float bs[MAX_COUT];
     for (int i = threadIdx.x; i < 27 * Cout; i += blockDim.x) {
@@ -57,7 +59,14 @@ void conv3_fwd_kernel(const bf16* __restrict__ in,   // [B,H,W,3]
         }
     }
 
-    bf16* op = out + (int64_t)pix * Cout;
+    // emit_pad: write into the interior of a 1-px zero-bordered buffer
+    // so the NEXT 3x3 conv consumes it without a pad pass (the thread
+    // already knows (b,y,x) — the padded store is free here, while a
+    // separate pad1_nhwc of a 224^2 x64 tensor costs ~70 us)
+    bf16* op = emit_pad
+        ? out + ((((int64_t)b * (Hh + 2) + y + 1) * (Ww + 2) + x + 1)
+                 * Cout)
+        : out + (int64_t)pix * Cout;
     for (int c0 = 0; c0 < Cout; c0 += 8) {
         bf16x8 o;
 #pragma unroll
@@ -74,8 +83,29 @@ void conv3_fwd_kernel(const bf16* __restrict__ in,   // [B,H,W,3]
     }
 }
 
+// zero only the 1-px border of a padded NHWC buffer (the interior is
+// fully overwritten by the producing conv)
+__global__ void zero_border_kernel(bf16* __restrict__ out,
+                                   int B, int Hp, int Wp, int C) {
+    int64_t border = (int64_t)B * 2 * (Hp + Wp - 2);  // border pixels
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t n = border * (C / 8);
+    if (i >= n) return;
+    int c8 = (int)(i % (C / 8)) * 8;
+    int64_t pb = i / (C / 8);
+    int b = (int)(pb / (2 * (Hp + Wp - 2)));
+    int q = (int)(pb % (2 * (Hp + Wp - 2)));
+    int y, x;
+    if (q < Wp) { y = 0; x = q; }
+    else if (q < 2 * Wp) { y = Hp - 1; x = q - Wp; }
+    else if (q < 2 * Wp + (Hp - 2)) { y = q - 2 * Wp + 1; x = 0; }
+    else { y = q - 2 * Wp - (Hp - 2) + 1; x = Wp - 1; }
+    bf16x8 z = {};
+    *(bf16x8*)(out + (((int64_t)b * Hp + y) * Wp + x) * C + c8) = z;
+}
+
 at::Tensor conv3_fwd(at::Tensor input, at::Tensor weight, at::Tensor bias,
-                     bool relu) {
+                     bool relu, bool emit_pad) {
     // input NHWC-contiguous [B,3,H,W] tensor in channels_last
     CHECK_GPU(input); CHECK_BF16(input);
     CHECK_GPU(weight); CHECK_BF16(weight);
@@ -88,20 +118,31 @@ at::Tensor conv3_fwd(at::Tensor input, at::Tensor weight, at::Tensor bias,
     int Cout = weight.size(0);
     TORCH_CHECK(Cout % 8 == 0 && Cout <= MAX_COUT);
     auto wc = weight.contiguous();  // OIHW
-    auto out = at::empty({B, Cout, Hh, Ww},
-                         input.options()
-                             .memory_format(at::MemoryFormat::ChannelsLast));
+    auto out = emit_pad
+        ? at::empty({B, Cout, Hh + 2, Ww + 2},
+                    input.options()
+                        .memory_format(at::MemoryFormat::ChannelsLast))
+        : at::empty({B, Cout, Hh, Ww},
+                    input.options()
+                        .memory_format(at::MemoryFormat::ChannelsLast));
     const bf16* bias_ptr = nullptr;
     if (bias.defined() && bias.numel() > 0)
         bias_ptr = (const bf16*)bias.contiguous().data_ptr();
     int BHW = B * Hh * Ww;
     hipStream_t s = at::cuda::getCurrentCUDAStream();
+    if (emit_pad) {
+        int64_t nb = (int64_t)B * 2 * (Hh + 2 + Ww + 2 - 2) * (Cout / 8);
+        hipLaunchKernelGGL(zero_border_kernel, dim3(cdiv(nb, 256)),
+                           dim3(256), 0, s,
+                           (bf16*)out.data_ptr(), B, Hh + 2, Ww + 2,
+                           Cout);
+    }
     hipLaunchKernelGGL(conv3_fwd_kernel, dim3(cdiv(BHW, 256)), dim3(256),
                        0, s,
                        (const bf16*)input.data_ptr(),
                        (const bf16*)wc.data_ptr(), bias_ptr,
                        (bf16*)out.data_ptr(), BHW, Hh, Ww, Cout,
-                       relu ? 1 : 0);
+                       relu ? 1 : 0, emit_pad ? 1 : 0);
     HIP_OK(hipGetLastError());
     return out;
 }

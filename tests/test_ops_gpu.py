@@ -281,7 +281,7 @@ def test_conv3_fwd_matches_torch():
         .contiguous(memory_format=torch.channels_last)
     w = (torch.randn(C, 3, 3, 3) * 0.2).to(DEV, torch.bfloat16)
     b = torch.randn(C).to(DEV, torch.bfloat16)
-    y = _C.conv3_fwd(x, w, b, True)
+    y = _C.conv3_fwd(x, w, b, True, False)
     ref = torch.relu(torch.nn.functional.conv2d(
         x.float(), w.float(), b.float(), padding=1))
     assert y.shape == ref.shape
@@ -294,7 +294,7 @@ def test_conv3_fwd_no_bias_no_relu():
         .contiguous(memory_format=torch.channels_last)
     w = (torch.randn(8, 3, 3, 3) * 0.2).to(DEV, torch.bfloat16)
     y = _C.conv3_fwd(x, w, torch.empty(0, dtype=torch.bfloat16,
-                                       device=DEV), False)
+                                       device=DEV), False, False)
     ref = torch.nn.functional.conv2d(x.float(), w.float(), padding=1)
     assert _rel_err(y, ref) < 2e-2
 
@@ -475,3 +475,32 @@ def test_conv_igemm_glds64_matches_torch():
         ref = torch.relu(torch.nn.functional.conv2d(
             x.float(), w.float(), b.float(), padding=1))
         assert _rel_err(y, ref) < 2e-2, (Cin, Cout, B, H, W)
+
+
+def test_conv3_fwd_emit_pad_and_glds64_chain():
+    """conv1_1 emitting a zero-bordered padded output that conv1_2
+    (glds64) consumes directly — the chained-pad fast path."""
+    from sat_amd import _C
+    torch.manual_seed(16)
+    B, H, W = 2, 20, 24
+    x = torch.randn(B, 3, H, W).to(DEV, torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    w1 = (torch.randn(64, 3, 3, 3) * 0.2).to(DEV, torch.bfloat16)
+    b1 = torch.randn(64).to(DEV, torch.bfloat16)
+    w2 = (torch.randn(64, 64, 3, 3) * 0.05).to(DEV, torch.bfloat16)
+    b2 = torch.randn(64).to(DEV, torch.bfloat16)
+
+    yp = _C.conv3_fwd(x, w1, b1, True, True)
+    assert yp.shape == (B, 64, H + 2, W + 2)
+    # border must be exactly zero
+    assert float(yp[:, :, 0, :].float().abs().max()) == 0.0
+    assert float(yp[:, :, :, -1].float().abs().max()) == 0.0
+    # interior equals the unpadded conv
+    y0 = _C.conv3_fwd(x, w1, b1, True, False)
+    assert torch.equal(yp[:, :, 1:H + 1, 1:W + 1], y0)
+
+    w2_ohwi = w2.permute(0, 2, 3, 1).contiguous().reshape(64, -1)
+    y2 = _C.conv_igemm_glds64_fwd(yp, w2_ohwi, b2, H, W, True)
+    ref = torch.relu(torch.nn.functional.conv2d(
+        y0.float(), w2.float(), b2.float(), padding=1))
+    assert _rel_err(y2, ref) < 2e-2
