@@ -260,6 +260,30 @@ class Conv2d(tnn.Module):
                             xp, self._w_ohwi, eb, H, W, relu)
                     return _C.conv_igemm_glds64_fwd(
                         xp, self._w_ohwi, eb, H, W, relu)
+        # 1x1/s1 conv IS a GEMM over the NHWC row view: route through
+        # the in-tree tiled MFMA GEMM instead of MIOpen (ResNet50's
+        # bottleneck convs; bias handled by the GEMM epilogue)
+        if (x.is_cuda and x.dtype == torch.bfloat16
+                and not torch.is_grad_enabled()
+                and k == 1 and st == 1 and pad_hw is None
+                and self._glds_conv
+                and w.shape[0] % 8 == 0 and w.shape[1] % 8 == 0
+                and x.is_contiguous(memory_format=torch.channels_last)):
+            from ..ops import hip
+            if hip.available():
+                from sat_amd import _C
+                B_, Cin, H, W = x.shape
+                Cout = w.shape[0]
+                rows = x.permute(0, 2, 3, 1).reshape(B_ * H * W, Cin)
+                if getattr(self, '_w_1x1', None) is None or \
+                        self._w_1x1_ver != self.weight._version:
+                    self._w_1x1_ver = self.weight._version
+                    self._w_1x1 = w.reshape(Cout, Cin).contiguous()
+                eb = (b if b is not None else
+                      torch.empty(0, dtype=x.dtype, device=x.device))
+                act = 2 if self.activation == 'relu' else 0
+                y = _C.dense_fwd(rows, self._w_1x1, eb, act)
+                return y.reshape(B_, H, W, Cout).permute(0, 3, 1, 2)
         if pad_hw is not None:
             # padded input fell through to a library path: unpad first
             x = x[:, :, 1:-1, 1:-1] \

@@ -504,3 +504,27 @@ def test_conv3_fwd_emit_pad_and_glds64_chain():
     ref = torch.relu(torch.nn.functional.conv2d(
         y0.float(), w2.float(), b2.float(), padding=1))
     assert _rel_err(y2, ref) < 2e-2
+
+
+def test_conv1x1_as_gemm_route():
+    """ResNet bottleneck 1x1 convs route through the tiled MFMA GEMM."""
+    import torch.nn as tnn
+    from config import Config
+    from sat_amd.models.nn import NN, Conv2d
+    cfg = Config()
+    cfg.phase = 'eval'
+    pol = NN(cfg)
+    torch.manual_seed(17)
+    for Cin, Cout, relu in [(64, 256, True), (256, 64, False)]:
+        conv = Conv2d(pol, Cin, Cout, 1, 1,
+                      'relu' if relu else None, use_bias=False)
+        conv = conv.to(DEV)
+        x = torch.randn(2, Cin, 14, 14).to(DEV, torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+        with torch.no_grad():
+            y = conv(x)
+            ref = torch.nn.functional.conv2d(
+                x.float(), conv.weight.float())
+            if relu:
+                ref = torch.relu(ref)
+        assert _rel_err(y, ref) < 2e-2, (Cin, Cout)
