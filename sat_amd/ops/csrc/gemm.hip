@@ -20,6 +20,9 @@
 // guarded edges so any K%8==0 shape works.
 
 #include "common.h"
+
+#include <mutex>
+#include <unordered_map>
 #include <map>
 #include <mutex>
 
@@ -290,6 +293,30 @@ __global__ void skinny_epilogue_kernel(const float* __restrict__ Yf,
     Y[idx] = f2bf(apply_act(v, act));
 }
 
+// persistent per-grid ticket counters for the in-launch split-K
+// combine (monotonic, modulo-read — never reset, so graph replays and
+// repeated launches stay correct; zero-initialized once per size)
+static at::Tensor _skinny_cnt(int nblocks, const at::TensorOptions& o) {
+    static std::unordered_map<int, at::Tensor> cache;
+    static std::mutex mu;
+    std::lock_guard<std::mutex> g(mu);
+    auto it = cache.find(nblocks);
+    if (it == cache.end())
+        it = cache.emplace(nblocks,
+                           at::zeros({nblocks},
+                                     o.dtype(at::kInt))).first;
+    return it->second;
+}
+
+static bool _skinny_fused() {
+    static int v = -1;
+    if (v < 0) {
+        const char* e = getenv("SAT_SKINNY_FUSED");
+        v = (e != nullptr && e[0] == '0') ? 0 : 1;
+    }
+    return v == 1;
+}
+
 at::Tensor dense_fwd_out(at::Tensor x, at::Tensor w, at::Tensor bias,
                          int64_t act, at::Tensor out) {
     CHECK_GPU(x); CHECK_CONTIG(x); CHECK_BF16(x);
@@ -337,6 +364,18 @@ at::Tensor dense_fwd_out(at::Tensor x, at::Tensor w, at::Tensor bias,
             else
                 LAUNCH_SKINNY(8, dim3(nblocks, 1), bias_ptr,
                               (bf16*)y.data_ptr(), nullptr, 1);
+        } else if (_skinny_fused()) {
+            // in-launch combine: each 64-col tile's last-arriving block
+            // reduces the slabs (agent-scope release/acquire + ticket,
+            // §6 G16 recipe) — no epilogue launch
+            auto yf = at::empty({splitk, M, N},
+                                x.options().dtype(at::kFloat));
+            auto cnt = _skinny_cnt(nblocks, x.options());
+#define LAUNCH_SKF(RF)             hipLaunchKernelGGL((skinny_gemm_kernel<RF>),                                dim3(nblocks, splitk), dim3(256), 0,                                stream, (const bf16*)x.data_ptr(),                                (const bf16*)w.data_ptr(), bias_ptr,                                (bf16*)y.data_ptr(),                                (float*)yf.data_ptr(),                                (uint32_t*)cnt.data_ptr(), (int)M,                                (int)N, (int)K, (int)act, splitk)
+            if (M <= 32) LAUNCH_SKF(2);
+            else if (M <= 64) LAUNCH_SKF(4);
+            else LAUNCH_SKF(8);
+#undef LAUNCH_SKF
         } else {
             auto yf = at::empty({splitk, M, N},
                                 x.options().dtype(at::kFloat));
