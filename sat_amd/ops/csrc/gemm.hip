@@ -309,10 +309,15 @@ static at::Tensor _skinny_cnt(int nblocks, const at::TensorOptions& o) {
 }
 
 static bool _skinny_fused() {
+    // DEFAULT OFF: the in-launch combine measured 6.87 vs 6.19 ms/step
+    // on the flagship (r02 same-box A/B) — at these B=32 shapes the
+    // last-arriving block's serial slab read blocks the dependent next
+    // kernel, while the separate 256-block epilogue parallelizes it.
+    // Kept behind SAT_SKINNY_FUSED=1 for larger-M experiments.
     static int v = -1;
     if (v < 0) {
         const char* e = getenv("SAT_SKINNY_FUSED");
-        v = (e != nullptr && e[0] == '0') ? 0 : 1;
+        v = (e != nullptr && e[0] == '1') ? 1 : 0;
     }
     return v == 1;
 }
