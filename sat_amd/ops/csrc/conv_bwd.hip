@@ -208,7 +208,11 @@ at::Tensor conv3x3_wgrad(at::Tensor xpad, at::Tensor dy_rows,
 
     int nchunks = cdiv(M, 64);
     int tiles = cdiv(9 * Cin, 128) * cdiv(Cout, 128);
-    int splitk = std::max(1, std::min(32, 512 / tiles));
+    // target ~2 blocks/CU; the old cap of 32 left small-tile layers at
+    // 1 block/CU where the 2-barrier chunk loop is latency-bound
+    // (conv1_2 wgrad: 775 us capped vs 386 at full split; big-tile
+    // layers prefer 512/tiles — per-shape sweep in r02 logs)
+    int splitk = std::max(1, std::min(128, 512 / tiles));
     int chunks_per = cdiv(nchunks, splitk);
     splitk = cdiv(nchunks, chunks_per);
     auto dw = (splitk > 1)
