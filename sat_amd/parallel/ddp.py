@@ -55,6 +55,11 @@ class DataParallelGrads(object):
         self.world = dist.get_world_size(process_group)
         self.buckets = []
         self._param_bucket = {}
+        # defer_comm: hooks fill/scale the flat buffers but do NOT launch
+        # collectives — launch_deferred_comm() issues them later.  Used
+        # by the engine's split-capture mode: the fwd+bwd graph captures
+        # the bucket copies while the RCCL calls stay outside capture.
+        self.defer_comm = False
 
         named = [(n, p) for n, p in model.named_parameters()
                  if p.requires_grad]
@@ -102,8 +107,45 @@ class DataParallelGrads(object):
         b.pending -= 1
         if b.pending == 0:
             b.flat.div_(self.world)
+            if not self.defer_comm:
+                b.work = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
+                                         group=self.group, async_op=True)
+
+    # ---- deferred-comm protocol (engine split-capture mode) ----
+    # The fwd+bwd hipGraph captures the bucket fill/scale kernels; the
+    # collectives run EAGERLY between graph replay and the optimizer.
+    # Capture-static graphs fill the same buckets every replay, so the
+    # capture-time bookkeeping (which buckets completed) holds for
+    # every replay.
+
+    def snapshot_capture_state(self):
+        for b in self.buckets:
+            b_completed = (b.pending == 0)
+            setattr(b, 'completed_in_capture', b_completed)
+
+    def launch_deferred_comm(self):
+        for b in self.buckets:
+            if b.flat is None:
+                b.flat = torch.zeros(
+                    b.numel, dtype=torch.float32,
+                    device=next(iter(b.offsets)).device)
+            if not getattr(b, 'completed_in_capture', False):
+                # hook-side div_ never ran for this bucket
+                b.flat.div_(self.world)
             b.work = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
                                      group=self.group, async_op=True)
+
+    def wait_deferred_comm(self):
+        for b in self.buckets:
+            if b.work is not None:
+                b.work.wait()
+                b.work = None
+            for p in b.params:
+                if p.grad is None:
+                    continue
+                off = b.offsets[p]
+                g = b.flat[off:off + p.numel()].reshape(p.shape)
+                p.grad.detach().copy_(g.to(p.dtype))
 
     def finish_backward(self):
         """Wait for in-flight buckets and write averaged grads back."""

@@ -228,3 +228,45 @@ def test_bucket_phase_segregation():
             assert len(kinds) == 1, [names[p] for p in b.params]
     finally:
         dist.destroy_process_group()
+
+
+def _worker_deferred(rank, world, tmpfile, q):
+    dist.init_process_group(
+        'gloo', init_method='file://%s' % tmpfile,
+        rank=rank, world_size=world)
+    try:
+        torch.manual_seed(9)
+        model = torch.nn.Sequential(torch.nn.Linear(8, 8),
+                                    torch.nn.Linear(8, 4))
+        from sat_amd.parallel.ddp import DataParallelGrads
+        ddp = DataParallelGrads(model, bucket_mb=1)
+        ddp.defer_comm = True
+
+        torch.manual_seed(70 + rank)
+        x = torch.randn(4, 8)
+        model(x).pow(2).mean().backward()
+        ddp.snapshot_capture_state()
+        ddp.launch_deferred_comm()
+        ddp.wait_deferred_comm()
+
+        g = model[0].weight.grad.clone()
+        gathered = [torch.zeros_like(g) for _ in range(world)]
+        dist.all_gather(gathered, g)
+        same = all(torch.allclose(gathered[0], gi, atol=1e-6)
+                   for gi in gathered)
+        q.put(('ok', rank, bool(same)))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_deferred_comm_protocol_averages_grads(tmp_path):
+    """The split-capture path (hooks fill buckets, collectives launched
+    separately) must produce the same averaged gradients as the inline
+    hook-launched path."""
+    world = 2
+    q = mp.get_context('spawn').Queue()
+    f = str(tmp_path / 'init4')
+    mp.spawn(_worker_deferred, args=(world, f, q), nprocs=world,
+             join=True)
+    results = [q.get() for _ in range(world)]
+    assert all(r[0] == 'ok' and r[2] for r in results)
