@@ -37,7 +37,7 @@ def _phase_key(name):
 
 class _Bucket:
     __slots__ = ('params', 'numel', 'flat', 'offsets', 'pending', 'work',
-                 'unfilled')
+                 'unfilled', 'completed_in_capture')
 
     def __init__(self):
         self.params = []
@@ -47,6 +47,7 @@ class _Bucket:
         self.pending = 0
         self.work = None
         self.unfilled = set()
+        self.completed_in_capture = False
 
 
 class DataParallelGrads(object):
@@ -120,8 +121,7 @@ class DataParallelGrads(object):
 
     def snapshot_capture_state(self):
         for b in self.buckets:
-            b_completed = (b.pending == 0)
-            setattr(b, 'completed_in_capture', b_completed)
+            b.completed_in_capture = (b.pending == 0)
 
     def launch_deferred_comm(self):
         for b in self.buckets:
@@ -129,7 +129,7 @@ class DataParallelGrads(object):
                 b.flat = torch.zeros(
                     b.numel, dtype=torch.float32,
                     device=next(iter(b.offsets)).device)
-            if not getattr(b, 'completed_in_capture', False):
+            if not b.completed_in_capture:
                 # hook-side div_ never ran for this bucket
                 b.flat.div_(self.world)
             b.work = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
