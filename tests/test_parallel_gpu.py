@@ -68,3 +68,74 @@ def test_two_rank_gpu_train_step_params_identical(tmp_path):
     for r in results:
         assert r[0] == 'ok', r
         assert r[2], 'rank %d saw divergent params' % r[1]
+
+
+def _worker_split(rank, world, tmpfile, q):
+    dist.init_process_group(
+        'gloo', init_method='file://%s' % tmpfile,
+        rank=rank, world_size=world)
+    try:
+        from config import Config
+        from sat_amd.models.base_model import BaseModel
+        from sat_amd import engine as eng
+
+        # force the full-capture attempt to fail so the SPLIT mode
+        # (fwd+bwd graph + deferred collectives + eager Adam) runs
+        def boom(self, *a):
+            raise RuntimeError('forced full-capture failure (test)')
+        orig = eng.GraphedTrainStep._capture
+        eng.GraphedTrainStep._capture = boom
+        try:
+            cfg = Config()
+            cfg.phase = 'train'
+            cfg.train_cnn = False
+            cfg.synthetic_data = True
+            cfg.device = 'cuda'
+            cfg.batch_size = 4
+            cfg.use_hip_graph = True
+            torch.manual_seed(cfg.seed)
+            torch.cuda.manual_seed_all(cfg.seed)
+            m = BaseModel(cfg)
+            for step in range(3):
+                torch.manual_seed(3000 * (step + 1) + rank)
+                images = torch.randn(4, 3, 224, 224,
+                                     device='cuda') * 40.0
+                sentences = torch.randint(
+                    1, cfg.vocabulary_size,
+                    (4, cfg.max_caption_length), device='cuda')
+                masks = torch.ones(4, cfg.max_caption_length,
+                                   device='cuda')
+                out = m.train_step(images, sentences, masks)
+                assert torch.isfinite(out['total_loss']).item()
+            mode = m._engine.mode
+            vec = torch.cat([p.detach().float().reshape(-1)
+                             for p in m.model.decoder.parameters()]) \
+                .cpu()
+            gathered = [torch.zeros_like(vec) for _ in range(world)]
+            dist.all_gather(gathered, vec)
+            same = all(bool(torch.equal(gathered[0], g))
+                       for g in gathered)
+            q.put(('ok', rank, same, mode))
+        finally:
+            eng.GraphedTrainStep._capture = orig
+    except Exception:
+        import traceback
+        q.put(('err', rank, traceback.format_exc(), None))
+        raise
+    finally:
+        dist.destroy_process_group()
+
+
+def test_split_capture_mode_two_ranks(tmp_path):
+    """Full capture forced to fail -> the engine must land in SPLIT mode
+    (graph without collectives) and keep ranks bit-identical."""
+    world = 2
+    os.environ.setdefault('HSA_ENABLE_IPC_MODE_LEGACY', '0')
+    q = mp.get_context('spawn').Queue()
+    f = str(tmp_path / 'init_split')
+    mp.spawn(_worker_split, args=(world, f, q), nprocs=world, join=True)
+    results = [q.get() for _ in range(world)]
+    for r in results:
+        assert r[0] == 'ok', r
+        assert r[2], 'rank %d diverged' % r[1]
+        assert r[3] == 'split', r
