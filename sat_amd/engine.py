@@ -78,6 +78,8 @@ class GraphedTrainStep(object):
         opt.step_count = snap['step_count']
         if getattr(opt, 'step_dev', None) is not None:
             opt.step_dev.fill_(float(opt.step_count))
+        if hasattr(opt, 'sync_shadows'):
+            opt.sync_shadows()  # restored params -> refresh bf16 shadows
         rng = self._rng_tensor()
         if rng is not None and 'rng' in snap:
             rng.copy_(snap['rng'])

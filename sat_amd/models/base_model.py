@@ -67,6 +67,18 @@ class BaseModel(object):
         self.optimizer = Optimizer(
             config, self.model.parameters()) if self.is_train else None
 
+        # bf16 shadow weights: the fused Adam refreshes persistent bf16
+        # copies of the BPTT weights in its update pass, removing the
+        # per-forward weight casts and cast-node backwards from the
+        # training step (sat_amd/optim.py, models/decoder.py)
+        if (self.optimizer is not None and self.device.type == 'cuda'
+                and getattr(config, 'use_hip_kernels', True)
+                and getattr(config, 'use_bptt', True)):
+            smap = self.model.decoder.ensure_shadows()
+            if smap:
+                self.optimizer.register_shadows(smap)
+                self.model.decoder._shadows_active = True
+
         # distributed (one process per GPU over RCCL; gloo on CPU)
         self.ddp = None
         if torch.distributed.is_available() \

@@ -53,12 +53,26 @@ class DecoderCoreBPTT(torch.autograd.Function):
     @staticmethod
     def forward(ctx_ag, contexts, init_memory, init_output, sentences,
                 masks, emb, w1a, b1a, w1b, b1b, v, wl, bl, seed,
-                p_fc, p_lstm, train_cnn):
+                p_fc, p_lstm, train_cnn, shadows=None):
+        # shadow mode: the weight INPUTS are the fp32 leaves (grads
+        # return straight to AccumulateGrad) while the COMPUTE tensors
+        # are the Adam-refreshed bf16 shadows — no per-forward casts
+        # and no cast-node backwards.
+        if shadows is not None:
+            emb_c = shadows['emb']
+            w1a_c, b1a_c = shadows['w1a'], shadows['b1a']
+            w1b_c, b1b_c = shadows['w1b'], shadows['b1b']
+            v_c = shadows['v'].reshape(-1)
+            wl_c, bl_c = shadows['wl'], shadows['bl']
+        else:
+            emb_c, w1a_c, b1a_c = emb, w1a, b1a
+            w1b_c, b1b_c, v_c = w1b, b1b, v
+            wl_c, bl_c = wl, bl
         B, L, D = contexts.shape
         T = sentences.shape[1]
-        A = w1a.shape[0]
+        A = w1a_c.shape[0]
         H = init_memory.shape[1]
-        E = emb.shape[1]
+        E = emb_c.shape[1]
         I = D + E
         dev = contexts.device
 
@@ -70,10 +84,11 @@ class DecoderCoreBPTT(torch.autograd.Function):
         # instead of 20 per-step launches.
         if p_fc > 0.0:
             CDROP = _C.hash_dropout_steps(ctx_flat, seed, p_fc, 0, 16, T)
-            T1 = _C.dense_fwd(CDROP.reshape(T * B * L, D), w1a, b1a, 1)
+            T1 = _C.dense_fwd(CDROP.reshape(T * B * L, D), w1a_c,
+                              b1a_c, 1)
         else:
             CDROP = None
-            T1 = _C.dense_fwd(ctx_flat, w1a, b1a, 1)  # shared by all t
+            T1 = _C.dense_fwd(ctx_flat, w1a_c, b1a_c, 1)  # shared
 
         # forward-side batched buffers (consumed by backward's batched dW)
         XH = torch.empty(T * B, I + H, dtype=torch.bfloat16, device=dev)
@@ -103,24 +118,24 @@ class DecoderCoreBPTT(torch.autograd.Function):
 
             t1 = T1.reshape(-1, A)[t * B * L:(t + 1) * B * L] \
                 if CDROP is not None else T1
-            t2 = _C.dense_fwd(ODROP[sl], w1b, b1b, ACT_TANH)
+            t2 = _C.dense_fwd(ODROP[sl], w1b_c, b1b_c, ACT_TANH)
             tdrop, att_logits = _C.attn_scores_fused(
-                t1, t2, v, seed, p_fc, s + 2, L)
+                t1, t2, v_c, seed, p_fc, s + 2, L)
             alpha, pooled = _C.attn_pool_fwd(contexts, att_logits)
 
-            _C.lstm_in_fuse(pooled, emb, last_word, state_h, seed,
+            _C.lstm_in_fuse(pooled, emb_c, last_word, state_h, seed,
                             p_lstm, s + 3, XH[sl])
             if fuse_small:
                 gates, h_raw, c_new = _C.dense_lstm_fwd(
-                    XH[sl], wl, bl, memory, 1.0)
+                    XH[sl], wl_c, bl_c, memory, 1.0)
             else:
-                gates = _C.dense_fwd(XH[sl], wl, bl, ACT_NONE)
+                gates = _C.dense_fwd(XH[sl], wl_c, bl_c, ACT_NONE)
                 h_raw, c_new = _C.lstm_pointwise_fwd(gates, memory, 1.0)
             od_next = ODROP[(t + 1) * B:(t + 2) * B] if t + 1 < T \
                 else empty_b
             out_t, sth_t = _C.expand_fuse(
-                h_raw, pooled, emb, last_word, seed, EXPD[sl], od_next,
-                p_lstm, p_fc, s)
+                h_raw, pooled, emb_c, last_word, seed, EXPD[sl],
+                od_next, p_lstm, p_fc, s)
 
             t1s.append(t1)
             t2s.append(t2)
@@ -138,8 +153,9 @@ class DecoderCoreBPTT(torch.autograd.Function):
                     * masks.t().reshape(T, B, 1)).sum(dim=0)
 
         ctx_ag.save_for_backward(
-            contexts, emb, w1a, b1a, w1b, b1b, v, wl, bl, seed, XH,
-            ODROP, labels_cat, masks)
+            contexts, emb_c, w1a_c, b1a_c, w1b_c, b1b_c, v_c, wl_c,
+            bl_c, seed, XH, ODROP, labels_cat, masks)
+        ctx_ag.leaf_mode = shadows is not None
         ctx_ag.cdrop = CDROP
         ctx_ag.saved_lists = (t1s, t2s, tdrops, alphas, gates_l, cprev_l)
         ctx_ag.dims = (B, L, D, T, A, H, E, I)
@@ -243,11 +259,19 @@ class DecoderCoreBPTT(torch.autograd.Function):
                          + d_sth_carry.float()).to(torch.bfloat16)
         d_init_memory = dc_carry
 
+        if ctx_ag.leaf_mode:
+            # fp32 grads straight to the leaves' AccumulateGrad (the
+            # fp32 sums are already fp32 — only the bf16 GEMM outputs
+            # need one cast each); dv keeps the leaf's [1, A] shape
+            return (dctx_acc, d_init_memory, d_init_output, None, None,
+                    demb_table.float(), dW1a.float(), db1a,
+                    dW1b.float(), db1b, dv_acc.reshape(1, -1),
+                    dWl.float(), dbl, None, None, None, None, None)
         bf = torch.bfloat16
         return (dctx_acc, d_init_memory, d_init_output, None, None,
                 demb_table.to(bf), dW1a.to(bf), db1a.to(bf),
                 dW1b.to(bf), db1b.to(bf), dv_acc.to(bf),
-                dWl.to(bf), dbl.to(bf), None, None, None, None)
+                dWl.to(bf), dbl.to(bf), None, None, None, None, None)
 
 
 class DecodeHeadBPTT(torch.autograd.Function):
@@ -263,15 +287,19 @@ class DecodeHeadBPTT(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx_ag, expd, wd1, bd1, wd2, bd2, sentences, masks,
-                seed, p_fc):
+                seed, p_fc, shadows=None):
         B, T = sentences.shape
         V = wd2.shape[0]
         bf = torch.bfloat16
 
-        wd1c = wd1.to(bf)
-        bd1c = bd1.to(bf)
-        wd2c = wd2.to(bf)
-        bd2c = bd2.to(bf)
+        if shadows is not None:
+            wd1c, bd1c = shadows['wd1'], shadows['bd1']
+            wd2c, bd2c = shadows['wd2'], shadows['bd2']
+        else:
+            wd1c = wd1.to(bf)
+            bd1c = bd1.to(bf)
+            wd2c = wd2.to(bf)
+            bd2c = bd2.to(bf)
 
         HID = _C.dense_fwd(expd, wd1c, bd1c, ACT_TANH)   # [T·B, Dd]
         HD = _C.hash_dropout_slabs(HID, seed, p_fc, 7, 16, T) \
@@ -316,7 +344,8 @@ class DecodeHeadBPTT(torch.autograd.Function):
         dbd1 = DP1.float().sum(0)
         d_expd = DP1.matmul(wd1c)                 # [T·B, H+D+E]
 
-        return (d_expd, dWd1, dbd1, dWd2, dbd2, None, None, None, None)
+        return (d_expd, dWd1, dbd1, dWd2, dbd2, None, None, None, None,
+                None)
 
 
 _EMPTY = {}
@@ -334,15 +363,32 @@ def run_decoder_bptt(decoder, contexts, init_memory, init_output,
     """Run the fused BPTT decoder (core + head Functions). Returns
     (ce [B,T], attentions [B,L], predictions [B,T])."""
     d = decoder
+    sh = getattr(d, '_shadows', None) \
+        if getattr(d, '_shadows_active', False) else None
+    if sh:
+        # shadow mode: fp32 leaves as autograd inputs, Adam-refreshed
+        # bf16 shadows as compute tensors
+        expd, attn_acc = DecoderCoreBPTT.apply(
+            contexts, init_memory, init_output, sentences, masks,
+            d.embedding, d.att_fc_1a.weight, d.att_fc_1a.bias,
+            d.att_fc_1b.weight, d.att_fc_1b.bias, d.att_fc_2.weight,
+            d.lstm_w, d.lstm_b,
+            d._rng, d.nn.fc_drop_rate, d.nn.lstm_drop_rate,
+            d.nn.train_cnn, sh)
+        ce, predictions = DecodeHeadBPTT.apply(
+            expd, d.dec_fc_1.weight, d.dec_fc_1.bias,
+            d.dec_fc_2.weight, d.dec_fc_2.bias,
+            sentences, masks, d._rng, d.nn.fc_drop_rate, sh)
+        return ce, attn_acc, predictions
     expd, attn_acc = DecoderCoreBPTT.apply(
         contexts, init_memory, init_output, sentences, masks,
         d._emb_c, d.att_fc_1a._wc, d.att_fc_1a._bc,
         d.att_fc_1b._wc, d.att_fc_1b._bc, d._att_vc,
         d._lstm_wc, d._lstm_bc,
         d._rng, d.nn.fc_drop_rate, d.nn.lstm_drop_rate,
-        d.nn.train_cnn)
+        d.nn.train_cnn, None)
     ce, predictions = DecodeHeadBPTT.apply(
         expd, d.dec_fc_1.weight, d.dec_fc_1.bias,
         d.dec_fc_2.weight, d.dec_fc_2.bias,
-        sentences, masks, d._rng, d.nn.fc_drop_rate)
+        sentences, masks, d._rng, d.nn.fc_drop_rate, None)
     return ce, attn_acc, predictions

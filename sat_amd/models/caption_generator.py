@@ -113,14 +113,15 @@ class CaptionGenerator(tnn.Module):
 
         contexts = self.compute_contexts(images)
         contexts_flat = contexts.reshape(-1, self.dim_ctx)
-        self.decoder.precast(contexts.dtype)
+        use_bptt = self._use_bptt(contexts)
+        self.decoder.precast(contexts.dtype, skip_bptt=use_bptt)
 
         context_mean = contexts.float().mean(dim=1).to(contexts.dtype)
         initial_memory, initial_output = self.decoder.initialize(
             context_mean)
 
         mask_sum = masks.sum()
-        if self._use_bptt(contexts):
+        if use_bptt:
             # fused hand-written BPTT over all T steps (sat_amd.models.bptt)
             from .bptt import run_decoder_bptt
             ce, attentions, predictions = run_decoder_bptt(

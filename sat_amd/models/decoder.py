@@ -90,20 +90,64 @@ class AttentionDecoder(tnn.Module):
 
     # ---- per-forward weight cast cache ----
 
-    def precast(self, dtype):
+    def ensure_shadows(self):
+        """Persistent bf16 shadow tensors for the fused-BPTT weights.
+        Registered with the fused Adam (sat_amd/optim.py), which
+        refreshes them in its own update pass — the BPTT path then reads
+        shadows instead of re-casting weights every forward.  Returns a
+        {param: shadow} map (empty for non-default architectures)."""
+        if getattr(self, '_shadows', None) is not None:
+            return self._shadow_map
+        cfg = self.config
+        if cfg.num_attend_layers != 2 or cfg.num_decode_layers != 2:
+            self._shadows = {}
+            self._shadow_map = {}
+            return {}
+        pairs = [
+            ('emb', self.embedding), ('wl', self.lstm_w),
+            ('bl', self.lstm_b),
+            ('w1a', self.att_fc_1a.weight), ('b1a', self.att_fc_1a.bias),
+            ('w1b', self.att_fc_1b.weight), ('b1b', self.att_fc_1b.bias),
+            ('v', self.att_fc_2.weight),
+            ('wd1', self.dec_fc_1.weight), ('bd1', self.dec_fc_1.bias),
+            ('wd2', self.dec_fc_2.weight), ('bd2', self.dec_fc_2.bias),
+        ]
+        self._shadows = {k: torch.empty_like(p, dtype=torch.bfloat16)
+                         for k, p in pairs}
+        self._shadow_map = {p: self._shadows[k] for k, p in pairs}
+        return self._shadow_map
+
+    def precast(self, dtype, skip_bptt=False):
         """Cast every decoder weight to the compute dtype ONCE per model
         forward (instead of once per dense call x T steps): removes ~450
         bf16-cast kernel launches per training step while keeping the casts
-        inside the autograd graph."""
+        inside the autograd graph.  With skip_bptt (shadow-weight mode)
+        the BPTT-covered weights are not cast at all — the fused BPTT
+        reads the Adam-refreshed bf16 shadows instead."""
         from .nn import Dense
+        skip = set()
+        use_shadows = skip_bptt and getattr(self, '_shadows_active',
+                                            False)
+        if use_shadows:
+            skip = {self.att_fc_1a, self.att_fc_1b, self.att_fc_2,
+                    self.dec_fc_1, self.dec_fc_2}
         for m in self.modules():
             if isinstance(m, Dense):
-                m.precast(dtype)
-        self._emb_c = self.embedding.to(dtype)
-        self._lstm_wc = self.lstm_w.to(dtype)
-        self._lstm_bc = self.lstm_b.to(dtype)
-        if self.config.num_attend_layers != 1:
-            self._att_vc = self.att_fc_2._wc.reshape(-1)
+                if m in skip:
+                    m.clear_cast()
+                else:
+                    m.precast(dtype)
+        if use_shadows:
+            self._emb_c = None
+            self._lstm_wc = None
+            self._lstm_bc = None
+            self._att_vc = None
+        else:
+            self._emb_c = self.embedding.to(dtype)
+            self._lstm_wc = self.lstm_w.to(dtype)
+            self._lstm_bc = self.lstm_b.to(dtype)
+            if self.config.num_attend_layers != 1:
+                self._att_vc = self.att_fc_2._wc.reshape(-1)
         dev = self.embedding.device
         if dev.type == 'cuda':
             if self._rng is None or self._rng.device != dev:

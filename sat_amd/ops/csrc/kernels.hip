@@ -1017,7 +1017,7 @@ __global__ void sq_norm_mt_kernel(const int64_t* __restrict__ desc,
             int mid = (lo + hi) >> 1;
             if (i >= scum[mid]) lo = mid; else hi = mid;
         }
-        const float* g = (const float*)desc[lo * 4 + 1];
+        const float* g = (const float*)desc[lo * 5 + 1];
         float gv = g[i - scum[lo]];
         acc += gv * gv;
     }
@@ -1077,16 +1077,23 @@ __global__ void adam_mt_kernel(const int64_t* __restrict__ desc,
             if (i >= scum[mid]) lo = mid; else hi = mid;
         }
         int64_t off = i - scum[lo];
-        float* p = (float*)desc[lo * 4 + 0];
-        float* g = (float*)desc[lo * 4 + 1];
-        float* m = (float*)desc[lo * 4 + 2];
-        float* v = (float*)desc[lo * 4 + 3];
+        float* p = (float*)desc[lo * 5 + 0];
+        float* g = (float*)desc[lo * 5 + 1];
+        float* m = (float*)desc[lo * 5 + 2];
+        float* v = (float*)desc[lo * 5 + 3];
+        bf16* sh = (bf16*)desc[lo * 5 + 4];
         float gi = g[off] * scale;
         float mi = b1 * m[off] + (1.f - b1) * gi;
         float vi = b2 * v[off] + (1.f - b2) * gi * gi;
         m[off] = mi;
         v[off] = vi;
-        p[off] -= lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+        float pn = p[off] - lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+        p[off] = pn;
+        // write-through bf16 shadow: the compute path reads persistent
+        // bf16 copies instead of re-casting every forward (~29 cast
+        // kernels/step removed); the shadow refresh is free here — the
+        // updated value is already in a register
+        if (sh != nullptr) sh[off] = f2bf(pn);
         // fold next step's grad zeroing into the update pass: removes
         // one fill launch per parameter tensor per step (autograd then
         // accumulates into the zeroed buffer at fixed addresses, which

@@ -52,6 +52,24 @@ class Optimizer(object):
         for p in self.params:
             p.grad = None
 
+    # ---- bf16 shadow weights ----
+    # The fused Adam refreshes persistent bf16 copies of registered
+    # params in its own update pass (free — the new value is already in
+    # a register), so the compute path reads shadows instead of casting
+    # weights every forward.  Non-fused/eager steps fall back to an
+    # explicit sync so shadows can never go stale.
+
+    def register_shadows(self, mapping):
+        """mapping: {param: bf16 tensor of the same shape}."""
+        self._shadows = dict(mapping)
+        self._mt_ptrs = None  # force desc rebuild
+        self.sync_shadows()
+
+    def sync_shadows(self):
+        for p, sh in getattr(self, '_shadows', {}).items():
+            with torch.no_grad():
+                sh.copy_(p.detach().to(sh.dtype))
+
     @torch.no_grad()
     def step(self):
         cfg = self.config
@@ -77,10 +95,12 @@ class Optimizer(object):
                     device=self.params[0].device)
                 self.step_dev.fill_(float(self.step_count - 1))
             self.step_dev.add_(1.0)
+            shadows = getattr(self, '_shadows', {})
             ptrs = tuple(
                 (p.data.data_ptr(), g.data_ptr(),
                  self.state[p]['m'].data_ptr(),
-                 self.state[p]['v'].data_ptr())
+                 self.state[p]['v'].data_ptr(),
+                 shadows[p].data_ptr() if p in shadows else 0)
                 for p, g in zip(self.params, grads))
             if getattr(self, '_mt_ptrs', None) != ptrs:
                 dev = self.params[0].device
@@ -159,6 +179,7 @@ class Optimizer(object):
             else:
                 raise ValueError('unknown optimizer %r' % (self.kind,))
             p.data.add_(upd.to(p.dtype), alpha=-lr)
+        self.sync_shadows()
 
     # ---- checkpoint support: optimizer slots are saved like TF's ----
 
@@ -183,3 +204,4 @@ class Optimizer(object):
             self.step_count = int(float(arrays['optimizer/step_count']))
             if getattr(self, 'step_dev', None) is not None:
                 self.step_dev.fill_(float(self.step_count))
+        self.sync_shadows()

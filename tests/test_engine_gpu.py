@@ -81,3 +81,43 @@ def test_graphed_step_counter_advances(tiny_config):
     assert m.optimizer.step_count == 5
     if getattr(m.optimizer, 'step_dev', None) is not None:
         assert int(m.optimizer.step_dev.item()) == 5
+
+
+def test_shadow_weights_stay_in_sync():
+    """bf16 shadow weights (refreshed inside the fused Adam pass) must
+    bit-match a fresh cast of the fp32 leaves after real optimizer
+    steps, in both eager and captured modes."""
+    import torch
+    from config import Config
+    from sat_amd.models.base_model import BaseModel
+
+    for use_graph in (False, True):
+        cfg = Config()
+        cfg.phase = 'train'
+        cfg.train_cnn = False
+        cfg.synthetic_data = True
+        cfg.device = 'cuda'
+        cfg.batch_size = 4
+        cfg.use_hip_graph = use_graph
+        torch.manual_seed(cfg.seed)
+        m = BaseModel(cfg)
+        d = m.model.decoder
+        assert getattr(d, '_shadows_active', False)
+
+        for step in range(3):
+            torch.manual_seed(50 + step)
+            images = torch.randn(4, 3, 224, 224, device='cuda') * 40.0
+            sentences = torch.randint(
+                1, cfg.vocabulary_size, (4, cfg.max_caption_length),
+                device='cuda')
+            masks = torch.ones(4, cfg.max_caption_length, device='cuda')
+            out = m.train_step(images, sentences, masks)
+            assert torch.isfinite(out['total_loss']).item()
+
+        torch.cuda.synchronize()
+        smap = d._shadow_map
+        for p, sh in smap.items():
+            ref = p.detach().to(torch.bfloat16)
+            assert torch.equal(sh, ref), \
+                'stale shadow for %s (graph=%s)' % (tuple(p.shape),
+                                                    use_graph)
