@@ -13,6 +13,8 @@ import json
 import os
 import time
 
+import torch
+
 from .tb_events import TBEventWriter
 
 
@@ -36,7 +38,8 @@ class SummaryWriter(object):
         self._tb.add_scalars(scalars, step)
 
     def variable_summary(self, name, tensor, step):
-        """mean/stddev/max/min per variable (reference model.py:534-542)."""
+        """mean/stddev/max/min + histogram per variable (reference
+        model.py:534-542 logs exactly this set)."""
         t = tensor.detach().float()
         self.add_scalars({
             '%s/mean' % name: t.mean().item(),
@@ -44,6 +47,14 @@ class SummaryWriter(object):
             '%s/max' % name: t.max().item(),
             '%s/min' % name: t.min().item(),
         }, step)
+        # histogram on a bounded sample (full tensors are ~20 MB)
+        flat = t.reshape(-1)
+        if flat.numel() > 4096:
+            idx = torch.linspace(0, flat.numel() - 1, 4096,
+                                 dtype=torch.int64)
+            flat = flat.cpu()[idx]
+        self._tb.add_histogram('%s/hist' % name,
+                               flat.cpu().tolist(), step)
 
     def flush(self):
         self._f.flush()

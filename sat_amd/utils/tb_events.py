@@ -12,8 +12,8 @@ it, so the Event protobuf and TFRecord framing are hand-encoded here:
     5: Summary { repeated 1: Value { 1: string tag,
     2: float simple_value } }.
 
-Only scalar summaries are emitted — the scalar set matches what the
-reference logs; histograms stay in the JSONL sidecar.
+Scalar and histogram summaries are emitted — the same per-variable
+set the reference logs (model.py:534-542).
 """
 
 import os
@@ -86,6 +86,50 @@ def _scalar_value(tag, value):
     return _field_bytes(1, v)
 
 
+def _histo_value(tag, values):
+    """Summary.Value with a HistogramProto (field 5): fields
+    1 min, 2 max, 3 num, 4 sum, 5 sum_squares, 6 bucket_limit,
+    7 bucket (both packed repeated double)."""
+    import math
+    n = len(values)
+    vmin = min(values) if n else 0.0
+    vmax = max(values) if n else 0.0
+    vsum = float(sum(values))
+    vsq = float(sum(x * x for x in values))
+    # exponential bucket edges covering [vmin, vmax], TB-style
+    limits = []
+    edge = 1e-12
+    while edge < max(abs(vmin), abs(vmax), 1e-12) * 1.1:
+        edge *= 1.1
+        limits.append(edge)
+    edges = sorted({-e for e in limits} | set(limits) | {0.0})
+    counts = [0] * (len(edges) + 1)
+    for x in values:
+        lo, hi = 0, len(edges)
+        while lo < hi:
+            mid = (lo + hi) // 2
+            if x <= edges[mid]:
+                hi = mid
+            else:
+                lo = mid + 1
+        counts[lo] += 1
+    # trim empty tail buckets (keeps files small)
+    last = max((i for i, c in enumerate(counts) if c), default=0)
+    edges = edges[:last + 1]
+    counts = counts[:last + 1]
+
+    h = _varint((1 << 3) | 1) + struct.pack('<d', vmin)
+    h += _varint((2 << 3) | 1) + struct.pack('<d', vmax)
+    h += _varint((3 << 3) | 1) + struct.pack('<d', float(n))
+    h += _varint((4 << 3) | 1) + struct.pack('<d', vsum)
+    h += _varint((5 << 3) | 1) + struct.pack('<d', vsq)
+    h += _field_bytes(6, b''.join(struct.pack('<d', e) for e in edges))
+    h += _field_bytes(7, b''.join(struct.pack('<d', float(c))
+                                  for c in counts))
+    v = _field_bytes(1, tag.encode()) + _field_bytes(5, h)
+    return _field_bytes(1, v)
+
+
 class TBEventWriter(object):
     """Writes events.out.tfevents.* files TensorBoard can load."""
 
@@ -113,6 +157,10 @@ class TBEventWriter(object):
                            for t, v in scalars.items())
         self._record(_event(time.time(), step=int(step),
                             summary=summary))
+
+    def add_histogram(self, tag, values, step):
+        self._record(_event(time.time(), step=int(step),
+                            summary=_histo_value(tag, list(values))))
 
     def flush(self):
         self._f.flush()

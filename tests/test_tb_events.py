@@ -115,3 +115,27 @@ def test_summary_writer_emits_both_formats(tmp_path):
     w.close()
     assert (tmp_path / 'events.jsonl').exists()
     assert glob.glob(str(tmp_path / 'events.out.tfevents.*'))
+
+
+def test_tb_histogram_round_trips(tmp_path):
+    w = TBEventWriter(str(tmp_path))
+    w.add_histogram('wts/hist', [-1.0, -0.5, 0.0, 0.25, 1.0, 1.0], 3)
+    w.close()
+    files = glob.glob(str(tmp_path / 'events.out.tfevents.*'))
+    recs = _read_records(files[0])
+    f = _parse_fields(recs[1])
+    assert f[2][0] == 3
+    summ = _parse_fields(f[5][0])
+    val = _parse_fields(summ[1][0])
+    assert val[1][0] == b'wts/hist'
+    histo = _parse_fields(val[5][0])
+    assert histo[1][0] == -1.0          # min
+    assert histo[2][0] == 1.0           # max
+    assert histo[3][0] == 6.0           # num
+    assert abs(histo[4][0] - 0.75) < 1e-9   # sum
+    # packed repeated doubles: counts sum to num
+    counts = histo[7][0]
+    import struct as st
+    cs = [st.unpack('<d', counts[i:i + 8])[0]
+          for i in range(0, len(counts), 8)]
+    assert sum(cs) == 6.0
