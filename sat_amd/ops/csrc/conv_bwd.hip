@@ -25,32 +25,41 @@ constexpr int PAD = 8;                 // LDS row pad (bf16 elements)
 constexpr int LROW = 64 + PAD;         // LDS row stride (m-extent 64)
 }
 
+// TCO = co-tile rows (128, or 64 for Cout=64 layers where a 128-row
+// tile wastes half its loads and MFMA on clamped duplicates)
+template <int TCO>
 __global__ __launch_bounds__(256)
 void conv3x3_wgrad_kernel(const bf16* __restrict__ xpad, // [B,H+2,W+2,Ci]
                           const bf16* __restrict__ dy,   // [M, Cout] NHWC
                           float* __restrict__ dw,        // [Cout, 9*Cin]
                           int M, int Hh, int Ww, int Cin, int Cout,
                           int chunks_per, int nchunks, int use_atomic) {
-    __shared__ bf16 lds[2 * 128 * LROW];                 // 36 KiB
-    bf16* A = lds;                      // [128 co][LROW m]
-    bf16* Bx = lds + 128 * LROW;        // [128 n ][LROW m]
+    __shared__ bf16 lds[(128 + TCO) * LROW];
+    bf16* A = lds;                      // [TCO co][LROW m]
+    bf16* Bx = lds + TCO * LROW;        // [128 n ][LROW m]
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
     const int wave = tid >> 6;
-    const int wr = wave >> 1, wc = wave & 1;
+    // wave grid: TCO=128 -> 2(co) x 2(n); TCO=64 -> 1 x 4
+    const int wr = (TCO == 128) ? (wave >> 1) : 0;
+    const int wc = (TCO == 128) ? (wave & 1) : (wave >> 1);
+    const int wn32 = (TCO == 128) ? 0 : (wave & 1);  // 32-col sub-split
     const int n0 = blockIdx.x * 128;    // within 9*Cin
-    const int co0 = blockIdx.y * 128;
+    const int co0 = blockIdx.y * TCO;
     const int Wp = Ww + 2;
     const int HW = Hh * Ww;
     const int N9 = 9 * Cin;
 
-    // ---- staging role of this thread (constant): waves 0-1 stage dy,
-    // waves 2-3 stage x; each stages 8 m-rows x 8 channels per chunk ----
-    const int st = tid & 127;           // 0..127 within the role group
-    const int sc8 = (st & 15) * 8;      // channel slice [sc8, sc8+8)
-    const int sm8 = (st >> 4) * 8;      // m slice [sm8, sm8+8) of chunk
-    const bool stage_dy = wave < 2;
+    // ---- staging role of this thread (constant): the first TCO
+    // threads stage dy (8 m-rows x 8 channels each), the next 128
+    // stage x ----
+    const bool stage_dy = (TCO == 128) ? (wave < 2) : (tid < 64);
+    const int st = (TCO == 128) ? (tid & 127)
+                                : (stage_dy ? tid : (tid - 64) & 127);
+    const int snch = stage_dy ? (TCO / 8) : 16;   // 8-ch slices per row
+    const int sc8 = (st % snch) * 8;    // channel slice [sc8, sc8+8)
+    const int sm8 = (st / snch) * 8;    // m slice [sm8, sm8+8) of chunk
     // clamped global channel base (epilogue guards discard duplicates)
     int dyco = co0 + sc8;
     if (dyco > Cout - 8) dyco = Cout - 8;
@@ -137,7 +146,7 @@ void conv3x3_wgrad_kernel(const bf16* __restrict__ xpad, // [B,H+2,W+2,Ci]
             }
         }
         __syncthreads();                // previous MFMA done before write
-        {
+        if (TCO == 128 || tid < 64 + 128) {
             bf16* dst = (stage_dy ? A : Bx) + (int64_t)sc8 * LROW + sm8;
             // repack: row e of LDS gets channel e across the 8 m's
 #pragma unroll
@@ -154,18 +163,23 @@ void conv3x3_wgrad_kernel(const bf16* __restrict__ xpad, // [B,H+2,W+2,Ci]
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk) {
             const int kof = kk * 32 + kgrp * 8;
-            bf16x8 af[4], bf[4];
+            // per-wave output: TCO=128 -> 64co x 64n (4x4 frags);
+            // TCO=64 -> 64co x 32n (4x2 frags, n split by wn32)
+            constexpr int NJ = (TCO == 128) ? 4 : 2;
+            bf16x8 af[4], bf[NJ];
 #pragma unroll
-            for (int i = 0; i < 4; ++i) {
+            for (int i = 0; i < 4; ++i)
                 af[i] = *(const bf16x8*)(
                     A + (wr * 64 + i * 16 + lrow) * LROW + kof);
-                bf[i] = *(const bf16x8*)(
-                    Bx + (wc * 64 + i * 16 + lrow) * LROW + kof);
-            }
+#pragma unroll
+            for (int j = 0; j < NJ; ++j)
+                bf[j] = *(const bf16x8*)(
+                    Bx + (wc * 64 + wn32 * 32 + j * 16 + lrow) * LROW
+                    + kof);
 #pragma unroll
             for (int i = 0; i < 4; ++i)
 #pragma unroll
-                for (int j = 0; j < 4; ++j)
+                for (int j = 0; j < NJ; ++j)
                     acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         af[i], bf[j], acc[i][j], 0, 0, 0);
         }
@@ -173,8 +187,8 @@ void conv3x3_wgrad_kernel(const bf16* __restrict__ xpad, // [B,H+2,W+2,Ci]
 
     // epilogue: atomics only when K was split (dw pre-zeroed by host)
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-        const int col = n0 + wc * 64 + j * 16 + (lane & 15);
+    for (int j = 0; j < ((TCO == 128) ? 4 : 2); ++j) {
+        const int col = n0 + wc * 64 + wn32 * 32 + j * 16 + (lane & 15);
         if (col >= N9) continue;
 #pragma unroll
         for (int i = 0; i < 4; ++i) {
@@ -207,7 +221,9 @@ at::Tensor conv3x3_wgrad(at::Tensor xpad, at::Tensor dy_rows,
     int M = (int)M64;
 
     int nchunks = cdiv(M, 64);
-    int tiles = cdiv(9 * Cin, 128) * cdiv(Cout, 128);
+    int tco = (Cout % 128 == 0) ? 128 : 64;
+    TORCH_CHECK(Cout % tco == 0, "wgrad: Cout % 64");
+    int tiles = cdiv(9 * Cin, 128) * (Cout / tco);
     // target ~2 blocks/CU; the old cap of 32 left small-tile layers at
     // 1 block/CU where the 2-barrier chunk loop is latency-bound
     // (conv1_2 wgrad: 775 us capped vs 386 at full split; big-tile
@@ -218,14 +234,24 @@ at::Tensor conv3x3_wgrad(at::Tensor xpad, at::Tensor dy_rows,
     auto dw = (splitk > 1)
         ? at::zeros({Cout, 9 * Cin}, xpad.options().dtype(at::kFloat))
         : at::empty({Cout, 9 * Cin}, xpad.options().dtype(at::kFloat));
-    dim3 grid(cdiv(9 * Cin, 128), cdiv(Cout, 128), splitk);
+    dim3 grid(cdiv(9 * Cin, 128), Cout / tco, splitk);
     hipStream_t s = at::cuda::getCurrentCUDAStream();
-    hipLaunchKernelGGL(conv3x3_wgrad_kernel, grid, dim3(256), 0, s,
-                       (const bf16*)xpad.data_ptr(),
-                       (const bf16*)dy_rows.data_ptr(),
-                       (float*)dw.data_ptr(), M, (int)Hh, (int)Ww,
-                       (int)Cin, (int)Cout, chunks_per, nchunks,
-                       splitk > 1 ? 1 : 0);
+    if (tco == 128)
+        hipLaunchKernelGGL((conv3x3_wgrad_kernel<128>), grid, dim3(256),
+                           0, s,
+                           (const bf16*)xpad.data_ptr(),
+                           (const bf16*)dy_rows.data_ptr(),
+                           (float*)dw.data_ptr(), M, (int)Hh, (int)Ww,
+                           (int)Cin, (int)Cout, chunks_per, nchunks,
+                           splitk > 1 ? 1 : 0);
+    else
+        hipLaunchKernelGGL((conv3x3_wgrad_kernel<64>), grid, dim3(256),
+                           0, s,
+                           (const bf16*)xpad.data_ptr(),
+                           (const bf16*)dy_rows.data_ptr(),
+                           (float*)dw.data_ptr(), M, (int)Hh, (int)Ww,
+                           (int)Cin, (int)Cout, chunks_per, nchunks,
+                           splitk > 1 ? 1 : 0);
     HIP_OK(hipGetLastError());
     return dw;
 }
