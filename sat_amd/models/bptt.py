@@ -245,11 +245,11 @@ class DecoderCoreBPTT(torch.autograd.Function):
             # p_fc == 0: every step saw the same ctx_flat
             dW1a = DPRE1A.reshape(T, B * L, A).sum(0).t() \
                 .matmul(ctx_flat)
-        db1a = DPRE1A.float().sum(0)
+        db1a = DPRE1A.sum(0, dtype=torch.float32)
         dWl = DG.t().matmul(XH)
-        dbl = DG.float().sum(0)
+        dbl = DG.sum(0, dtype=torch.float32)
         dW1b = DPRE1B.t().matmul(ODROP)
-        db1b = DPRE1B.float().sum(0)
+        db1b = DPRE1B.sum(0, dtype=torch.float32)
         demb_table = _C.embedding_bwd(
             torch.cat([torch.zeros(B, dtype=torch.int64, device=dev),
                        labels_cat[:-B]]),
@@ -339,9 +339,9 @@ class DecodeHeadBPTT(torch.autograd.Function):
         # weight grads first: their AccumulateGrad + DDP hooks are what
         # the core's backward overlaps with
         dWd2 = DL.t().matmul(HD).float()
-        dbd2 = DL.float().sum(0)
+        dbd2 = DL.sum(0, dtype=torch.float32)
         dWd1 = DP1.t().matmul(expd).float()
-        dbd1 = DP1.float().sum(0)
+        dbd1 = DP1.sum(0, dtype=torch.float32)
         d_expd = DP1.matmul(wd1c)                 # [T·B, H+D+E]
 
         return (d_expd, dWd1, dbd1, dWd2, dbd2, None, None, None, None,

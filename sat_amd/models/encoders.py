@@ -9,10 +9,12 @@ ResNet50 (model.py:62-188): conv1 7x7-s2 + BN + ReLU + maxpool 3x3-s2, then
 bottleneck stages (3,4,6,3) with 1-3-1 convs, BN everywhere, projection
 shortcuts at stage entry; output is [B,7,7,2048] -> [B,49,2048].
 
-Conv compute is `sat_amd.models.nn.Conv2d` (MIOpen through PyTorch-ROCm —
-plain library convolutions, the decoder's fused hot path is what gets
-hand-written CDNA4 kernels).  Frozen-CNN policy: parameters are created
-non-trainable unless config.train_cnn (reference nn.py:66).
+Conv compute is `sat_amd.models.nn.Conv2d`, which routes per shape to
+the in-tree CDNA4 kernels (direct 3-ch conv, glds/glds64/8-phase
+implicit-GEMM, 1x1-as-GEMM; hand-written dgrad/wgrad under
+--train_cnn) with MIOpen only where it still measures faster
+(profiles/r02_conv_shapes.txt).  Frozen-CNN policy: parameters are
+created non-trainable unless config.train_cnn (reference nn.py:66).
 """
 
 import torch
