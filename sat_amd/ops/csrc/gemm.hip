@@ -49,7 +49,11 @@ void tiled_gemm_kernel(const bf16* __restrict__ A,    // [M,K]
                        const bf16* __restrict__ W,    // [N,K]
                        const bf16* __restrict__ bias, // [N] or null
                        bf16* __restrict__ Y,          // [M,N]
-                       int M, int N, int K, int act) {
+                       int M, int N, int K, int act,
+                       // split-K: when Yf != null, block z covers
+                       // K-range [kq*kchunk, ...) and stores an fp32
+                       // slab; skinny_epilogue_kernel combines
+                       float* __restrict__ Yf, int kchunk) {
     constexpr int BM = WR * FM * 16;
     constexpr int BN = WC * FN * 16;
     __shared__ bf16 As[BM * LDS_STRIDE];
@@ -77,7 +81,12 @@ void tiled_gemm_kernel(const bf16* __restrict__ A,    // [M,K]
     constexpr int A_CHUNKS = BM * BK / 8 / 256;  // bf16x8 chunks per thread
     constexpr int B_CHUNKS = BN * BK / 8 / 256;
 
-    for (int k0 = 0; k0 < K; k0 += BK) {
+    int kbeg = 0, kend = K;
+    if (Yf != nullptr) {
+        kbeg = blockIdx.z * kchunk;
+        kend = min(K, kbeg + kchunk);
+    }
+    for (int k0 = kbeg; k0 < kend; k0 += BK) {
 #pragma unroll
         for (int i = 0; i < A_CHUNKS; ++i) {
             int q = tid + 256 * i;
@@ -135,6 +144,8 @@ void tiled_gemm_kernel(const bf16* __restrict__ A,    // [M,K]
     }
 
     // epilogue (C/D map: col = lane&15, row = (lane>>4)*4 + reg)
+    float* slab = (Yf != nullptr)
+        ? Yf + (int64_t)blockIdx.z * M * N : nullptr;
 #pragma unroll
     for (int ni = 0; ni < FN; ++ni) {
         int col = bn + wc * FN * 16 + ni * 16 + (lane & 15);
@@ -144,9 +155,13 @@ void tiled_gemm_kernel(const bf16* __restrict__ A,    // [M,K]
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
                 int row = bm + wr * FM * 16 + mi * 16 + (lane >> 4) * 4 + r;
-                if (row < M && col < N)
-                    Y[(int64_t)row * N + col] =
-                        f2bf(apply_act(acc[mi][ni][r] + bv, act));
+                if (row < M && col < N) {
+                    if (slab != nullptr)
+                        slab[(int64_t)row * N + col] = acc[mi][ni][r];
+                    else
+                        Y[(int64_t)row * N + col] =
+                            f2bf(apply_act(acc[mi][ni][r] + bv, act));
+                }
             }
         }
     }
@@ -401,22 +416,49 @@ at::Tensor dense_fwd_out(at::Tensor x, at::Tensor w, at::Tensor bias,
                                splitk);
         }
 #undef LAUNCH_SKINNY
-    } else if (N <= 512) {
-        dim3 grid(cdiv(N, 64), cdiv(M, 128));
-        hipLaunchKernelGGL((tiled_gemm_kernel<4, 1, 2, 4>), grid, dim3(256),
-                           0, stream,
-                           (const bf16*)x.data_ptr(),
-                           (const bf16*)w.data_ptr(), bias_ptr,
-                           (bf16*)y.data_ptr(), (int)M, (int)N, (int)K,
-                           (int)act);
     } else {
-        dim3 grid(cdiv(N, 128), cdiv(M, 128));
-        hipLaunchKernelGGL((tiled_gemm_kernel<2, 2, 4, 4>), grid, dim3(256),
-                           0, stream,
-                           (const bf16*)x.data_ptr(),
-                           (const bf16*)w.data_ptr(), bias_ptr,
-                           (bf16*)y.data_ptr(), (int)M, (int)N, (int)K,
-                           (int)act);
+        // mid-M shapes (the decode-head GEMMs: M=640, N=1024..1536)
+        // leave the 2-barrier tile grid at 40-60 blocks on 256 CUs;
+        // split K until ~2/3 of the chip has a block (measured 28-42 TF
+        // -> the slab+epilogue pair more than doubles it)
+        int bn_t = (N <= 512) ? 64 : 128;
+        int tiles = cdiv(N, bn_t) * cdiv(M, 128);
+        int splitk = 1;
+        while (tiles * splitk < 160 && splitk < 8
+               && K >= 128 * 2 * splitk)
+            splitk *= 2;
+        int kchunk = cdiv(cdiv(K, splitk), 64) * 64;
+        splitk = cdiv(K, kchunk);
+        float* yf_ptr = nullptr;
+        at::Tensor yf;
+        if (splitk > 1) {
+            yf = at::empty({splitk, M, N},
+                           x.options().dtype(at::kFloat));
+            yf_ptr = (float*)yf.data_ptr();
+        }
+        dim3 grid(cdiv(N, bn_t), cdiv(M, 128), splitk);
+        if (N <= 512)
+            hipLaunchKernelGGL((tiled_gemm_kernel<4, 1, 2, 4>), grid,
+                               dim3(256), 0, stream,
+                               (const bf16*)x.data_ptr(),
+                               (const bf16*)w.data_ptr(), bias_ptr,
+                               (bf16*)y.data_ptr(), (int)M, (int)N,
+                               (int)K, (int)act, yf_ptr, kchunk);
+        else
+            hipLaunchKernelGGL((tiled_gemm_kernel<2, 2, 4, 4>), grid,
+                               dim3(256), 0, stream,
+                               (const bf16*)x.data_ptr(),
+                               (const bf16*)w.data_ptr(), bias_ptr,
+                               (bf16*)y.data_ptr(), (int)M, (int)N,
+                               (int)K, (int)act, yf_ptr, kchunk);
+        if (splitk > 1) {
+            int64_t n = M * N;
+            hipLaunchKernelGGL(skinny_epilogue_kernel,
+                               dim3(cdiv(n, 256)), dim3(256), 0, stream,
+                               (const float*)yf.data_ptr(), bias_ptr,
+                               (bf16*)y.data_ptr(), n, (int)N, (int)act,
+                               splitk);
+        }
     }
     HIP_OK(hipGetLastError());
     return y;
