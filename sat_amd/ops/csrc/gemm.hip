@@ -308,6 +308,12 @@ __global__ void skinny_epilogue_kernel(const float* __restrict__ Yf,
     Y[idx] = f2bf(apply_act(v, act));
 }
 
+__global__ void dense_glds_kernel(const bf16* __restrict__ A,
+                                  const bf16* __restrict__ W,
+                                  const bf16* __restrict__ bias,
+                                  bf16* __restrict__ Y,
+                                  int M, int N, int K, int act);
+
 // persistent per-grid ticket counters for the in-launch split-K
 // combine (monotonic, modulo-read — never reset, so graph replays and
 // repeated launches stay correct; zero-initialized once per size)
@@ -416,6 +422,17 @@ at::Tensor dense_fwd_out(at::Tensor x, at::Tensor w, at::Tensor bias,
                                splitk);
         }
 #undef LAUNCH_SKINNY
+    } else if (M >= 16384 && N % 128 == 0 && K % 64 == 0) {
+        // big-M dense shapes: glds-staged 128^2 tile (T1 attention
+        // projection measured 148 TF on the register-staged tile; the
+        // glds structure runs the same geometry at 536-658 TF on conv)
+        dim3 grid(cdiv(N, 128), cdiv(M, 128));
+        hipLaunchKernelGGL(dense_glds_kernel, grid, dim3(256), 0,
+                           stream,
+                           (const bf16*)x.data_ptr(),
+                           (const bf16*)w.data_ptr(), bias_ptr,
+                           (bf16*)y.data_ptr(), (int)M, (int)N, (int)K,
+                           (int)act);
     } else {
         // mid-M shapes (the decode-head GEMMs: M=640, N=1024..1536)
         // leave the 2-barrier tile grid at 40-60 blocks on 256 CUs;
@@ -677,4 +694,120 @@ void dense_drop_fwd(at::Tensor x, at::Tensor w, at::Tensor b, int64_t act,
                        (bf16*)y.data_ptr(), (bf16*)ydrop.data_ptr(),
                        n, (int)N, (int)act, splitk, (float)p, (int)salt);
     HIP_OK(hipGetLastError());
+}
+
+// ---------------------------------------------------------------------
+// glds-staged 128x128 dense GEMM for big-M shapes (the attention
+// projection T1: [T*B*L, 512] x [512, 512] ran at 148 TF on the
+// register-staged 2-barrier tile; the glds + XOR-swizzle structure
+// measured 536-658 TF on the conv igemm with identical geometry).
+// Same discipline as conv_igemm_glds_kernel: global_load_lds width 16,
+// lane-linear LDS, st_16x32 swizzle pre-applied to the per-lane GLOBAL
+// source address and XOR'd on the fragment reads.  K % 64 == 0.
+// ---------------------------------------------------------------------
+
+__global__ __launch_bounds__(256)
+void dense_glds_kernel(const bf16* __restrict__ A,   // [M,K]
+                       const bf16* __restrict__ W,   // [N,K]
+                       const bf16* __restrict__ bias,
+                       bf16* __restrict__ Y,
+                       int M, int N, int K, int act) {
+    constexpr int BM = 128, BN = 128, BKc = 64;
+    __shared__ bf16 lds[2 * BM * BKc];
+    bf16* As = lds;
+    bf16* Bs = lds + BM * BKc;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wave = tid >> 6;
+    const int wr = wave >> 1, wc = wave & 1;
+    const int bm = blockIdx.y * BM;
+    const int bn = blockIdx.x * BN;
+
+    // per-lane staged rows (4 per wave for each operand) + source swz
+    const bf16* aSrc[4];
+    const bf16* wSrc[4];
+    const int ci8 = (lane & 7) * 8;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+        int r = wave * 32 + i * 8 + (lane >> 3);
+        int swz = (r & 4) ? 16 : 0;
+        int ga = bm + r;
+        if (ga >= M) ga = M - 1;
+        aSrc[i] = A + (int64_t)ga * K + (ci8 ^ swz);
+        int gb = bn + r;
+        if (gb >= N) gb = N - 1;
+        wSrc[i] = W + (int64_t)gb * K + (ci8 ^ swz);
+    }
+
+    floatx4 acc[4][4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+            acc[i][j] = floatx4{0.f, 0.f, 0.f, 0.f};
+
+    const int lrow = lane & 15;
+    const int kgrp = lane >> 4;
+
+    for (int k0 = 0; k0 < K; k0 += BKc) {
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) uint32_t*)
+                    (aSrc[i] + k0),
+                (__attribute__((address_space(3))) uint32_t*)
+                    (As + (wave * 32 + i * 8) * BKc),
+                16, 0, 0);
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) uint32_t*)
+                    (wSrc[i] + k0),
+                (__attribute__((address_space(3))) uint32_t*)
+                    (Bs + (wave * 32 + i * 8) * BKc),
+                16, 0, 0);
+        }
+        __syncthreads();
+
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk) {
+            bf16x8 a_frag[4], b_frag[4];
+            const int kof = kk * 32 + kgrp * 8;
+#pragma unroll
+            for (int mi = 0; mi < 4; ++mi) {
+                int r = wr * 64 + mi * 16 + lrow;
+                a_frag[mi] = *(const bf16x8*)(
+                    As + r * BKc + (kof ^ ((r & 4) ? 16 : 0)));
+            }
+#pragma unroll
+            for (int ni = 0; ni < 4; ++ni) {
+                int r = wc * 64 + ni * 16 + lrow;
+                b_frag[ni] = *(const bf16x8*)(
+                    Bs + r * BKc + (kof ^ ((r & 4) ? 16 : 0)));
+            }
+#pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+                for (int ni = 0; ni < 4; ++ni)
+                    acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a_frag[mi], b_frag[ni], acc[mi][ni], 0, 0, 0);
+        }
+        __syncthreads();
+    }
+
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+        int col = bn + wc * 64 + ni * 16 + (lane & 15);
+        float bv = (bias != nullptr && col < N) ? bf2f(bias[col]) : 0.f;
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+                int row = bm + wr * 64 + mi * 16 + (lane >> 4) * 4 + r;
+                if (row < M && col < N) {
+                    float vv = acc[mi][ni][r] + bv;
+                    Y[(int64_t)row * N + col] = f2bf(apply_act(vv, act));
+                }
+            }
+        }
+    }
 }
