@@ -52,7 +52,10 @@ class Conv3x3Train(torch.autograd.Function):
         eb = (bias if bias is not None else
               torch.empty(0, dtype=x.dtype, device=x.device))
         y = _route_fwd(x, w_ohwi, eb, relu)
-        ctx.save_for_backward(x, w, y)
+        if relu:
+            ctx.save_for_backward(x, w, y)  # y needed for the ReLU peel
+        else:
+            ctx.save_for_backward(x, w)
         ctx.relu = relu
         ctx.has_bias = bias is not None
         return y
@@ -60,7 +63,11 @@ class Conv3x3Train(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         from sat_amd import _C
-        x, w, y = ctx.saved_tensors
+        if ctx.relu:
+            x, w, y = ctx.saved_tensors
+        else:
+            x, w = ctx.saved_tensors
+            y = None
         Cout, Cin = w.shape[0], w.shape[1]
         B, _, H, W = x.shape
         M = B * H * W

@@ -289,3 +289,27 @@ def test_device_beam_matches_host_beam():
         for (s1, p1), (s2, p2) in zip(hs, ds):
             assert s1 == s2, (k, s1, s2)
             assert abs(p1 - p2) < 1e-6 * max(abs(p1), 1e-30), (p1, p2)
+
+
+def test_chained_pad_tag_unpads_on_fallback():
+    """A tensor tagged `_sat_pad` by a padded-emit conv must be unpadded
+    before any consumer that cannot use it (library conv, non-accepting
+    layer) — the handshake degrades safely."""
+    import torch
+    from config import Config
+    from sat_amd.models.nn import NN, Conv2d
+
+    cfg = Config()
+    cfg.phase = 'eval'
+    pol = NN(cfg)
+    conv = Conv2d(pol, 8, 8, 3, 1, 'relu')
+    torch.manual_seed(21)
+    x = torch.randn(1, 8, 6, 6)
+    ref = conv(x)
+
+    xp = torch.zeros(1, 8, 8, 8)
+    xp[:, :, 1:7, 1:7] = x
+    xp._sat_pad = (6, 6)
+    got = conv(xp)  # CPU conv cannot consume padded input -> must unpad
+    assert got.shape == ref.shape
+    assert torch.allclose(got, ref, atol=1e-5)
