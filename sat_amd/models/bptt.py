@@ -206,10 +206,17 @@ class DecoderCoreBPTT(torch.autograd.Function):
                 s, D, E)
             dgates, dc_prev = _C.lstm_pointwise_bwd_out(
                 gates_l[t], cprev_l[t], dh_raw, dc_carry, 1.0, DG[sl])
-            dxh = _C.dense_fwd(dgates, wl_t, _EMPTY_B(dev), ACT_NONE)
-            dpooled, d_sth_carry = _C.dx_fuse(
-                dxh, dpool_dec, demb_dec, seed, DEMB[sl], p_lstm,
-                s + 3, H)
+            if B <= 128:
+                # dxh GEMM + the dx scatter in one epilogue
+                dpooled, d_sth_carry = _C.dense_dx_fuse(
+                    dgates, wl_t, dpool_dec, demb_dec, seed, DEMB[sl],
+                    p_lstm, s + 3, D, E, H)
+            else:
+                dxh = _C.dense_fwd(dgates, wl_t, _EMPTY_B(dev),
+                                   ACT_NONE)
+                dpooled, d_sth_carry = _C.dx_fuse(
+                    dxh, dpool_dec, demb_dec, seed, DEMB[sl], p_lstm,
+                    s + 3, H)
             dc_carry = dc_prev
 
             dalpha_t = d_attn * masks[:, t].unsqueeze(1)
@@ -218,8 +225,13 @@ class DecoderCoreBPTT(torch.autograd.Function):
                 dpooled, need_dctx)
             if need_dctx:
                 dctx_acc += dctx_t
-            dt1, dt2f, _dv = _C.attn_scores_bwd_acc(
-                tdrops[t], v, dlog_att, seed, p_fc, s + 2, L, dv_acc)
+            sl_a = slice(t * B * L, (t + 1) * B * L)
+            # dt1 tanh-backward fused into the scores kernel, written
+            # straight into the DPRE1A slab (one launch + no dt1
+            # round-trip instead of scores_bwd + act_bwd_out)
+            _dt1, dt2f, _dv = _C.attn_scores_bwd_tanh(
+                tdrops[t], v, dlog_att, seed, p_fc, s + 2, L, dv_acc,
+                t1s[t], DPRE1A[sl_a])
 
             _C.act_bwd_f32_out(dt2f, t2s[t], ACT_TANH, DPRE1B[sl])
             # dodrop GEMM with the ODROP mask (salt s+1) regenerated in
@@ -227,8 +239,6 @@ class DecoderCoreBPTT(torch.autograd.Function):
             d_out_carry = _C.dense_fwd_drop(DPRE1B[sl], w1b_t, seed,
                                             p_fc, s + 1)
 
-            sl_a = slice(t * B * L, (t + 1) * B * L)
-            _C.act_bwd_out(dt1, t1s[t], ACT_TANH, DPRE1A[sl_a])
             if need_dctx:
                 dcd = DPRE1A[sl_a].matmul(w1a)
                 dctx_acc += _drop(dcd, seed, p_fc, s + 0) \

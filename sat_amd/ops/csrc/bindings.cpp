@@ -5,6 +5,13 @@
 at::Tensor dense_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, int64_t act);
 at::Tensor dense_fwd_drop(at::Tensor x, at::Tensor w, at::Tensor seed,
                           double p, int64_t salt);
+std::vector<at::Tensor> dense_dx_fuse(at::Tensor dgates, at::Tensor wl_t,
+                                      at::Tensor dpool_dec,
+                                      at::Tensor demb_dec,
+                                      at::Tensor seed,
+                                      at::Tensor demb_out,
+                                      double p, int64_t salt,
+                                      int64_t D, int64_t E, int64_t H);
 at::Tensor conv3_fwd(at::Tensor input, at::Tensor weight, at::Tensor bias,
                      bool relu, bool emit_pad);
 void bias_act_nhwc(at::Tensor y, at::Tensor bias, bool relu);
@@ -87,6 +94,10 @@ std::vector<at::Tensor> attn_scores_bwd_acc(at::Tensor tdrop, at::Tensor v,
                                             at::Tensor seed, double p,
                                             int64_t salt, int64_t L,
                                             at::Tensor dv_acc);
+std::vector<at::Tensor> attn_scores_bwd_tanh(
+        at::Tensor tdrop, at::Tensor v, at::Tensor dlogits,
+        at::Tensor seed, double p, int64_t salt, int64_t L,
+        at::Tensor dv_acc, at::Tensor t1y, at::Tensor dt1_out);
 at::Tensor embedding_fwd(at::Tensor ids, at::Tensor table);
 at::Tensor embedding_bwd(at::Tensor ids, at::Tensor dy, int64_t rows);
 std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor labels,
@@ -111,6 +122,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("dense_fwd", &dense_fwd, "MFMA GEMM + bias/act (bf16)");
     m.def("dense_fwd_drop", &dense_fwd_drop,
           "skinny GEMM with hash-dropout fused into the split-K epilogue");
+    m.def("dense_dx_fuse", &dense_dx_fuse,
+          "dxh GEMM with the dx scatter fused into the epilogue");
     m.def("bias_act_nhwc", &bias_act_nhwc);
     m.def("scale_bias_act_nhwc", &scale_bias_act_nhwc);
     m.def("maxpool2x2_nhwc", &maxpool2x2_nhwc);
@@ -146,6 +159,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("attn_pool_bwd", &attn_pool_bwd);
     m.def("attn_scores_bwd", &attn_scores_bwd);
     m.def("attn_scores_bwd_acc", &attn_scores_bwd_acc);
+    m.def("attn_scores_bwd_tanh", &attn_scores_bwd_tanh,
+          "scores backward with fused tanh-bwd dt1 into a given slab");
     m.def("embedding_fwd", &embedding_fwd);
     m.def("embedding_bwd", &embedding_bwd);
     m.def("ce_fwd", &ce_fwd);

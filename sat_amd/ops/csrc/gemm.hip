@@ -811,3 +811,93 @@ void dense_glds_kernel(const bf16* __restrict__ A,   // [M,K]
         }
     }
 }
+
+// ---------------------------------------------------------------------
+// dxh GEMM with the dx_fuse scatter folded into the split-K epilogue:
+// dxh = dgates @ wl^T, then per element: input-dropout mask regen and
+// the split into dpooled (+dpool_dec), DEMB slab (+demb_dec), dsth.
+// Removes one launch + the dxh round trip per decoder reverse step.
+// ---------------------------------------------------------------------
+
+__global__ void skinny_epi_dx_kernel(const float* __restrict__ Yf,
+                                     const bf16* __restrict__ dpool_dec,
+                                     const bf16* __restrict__ demb_dec,
+                                     const int64_t* __restrict__ seed_p,
+                                     bf16* __restrict__ dpooled,
+                                     bf16* __restrict__ demb_out,
+                                     bf16* __restrict__ dsth,
+                                     int64_t n, int splitk,
+                                     int B, int D, int E, int H,
+                                     float p, int salt) {
+    const uint32_t seed = (uint32_t)(*seed_p);
+    const int I = D + E;
+    const int W = I + H;
+    int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= n) return;
+    float g = 0.f;
+    for (int k = 0; k < splitk; ++k) g += Yf[k * n + idx];
+    int b = (int)(idx / W), j = (int)(idx % W);
+    if (j < I) {
+        g *= dscaleg(seed, salt, (uint32_t)(b * I + j), p);
+        if (j < D)
+            dpooled[(int64_t)b * D + j] =
+                f2bf(g + bf2f(dpool_dec[(int64_t)b * D + j]));
+        else
+            demb_out[(int64_t)b * E + (j - D)] =
+                f2bf(g + bf2f(demb_dec[(int64_t)b * E + (j - D)]));
+    } else {
+        dsth[(int64_t)b * H + (j - I)] = f2bf(g);
+    }
+}
+
+std::vector<at::Tensor> dense_dx_fuse(at::Tensor dgates, at::Tensor wl_t,
+                                      at::Tensor dpool_dec,
+                                      at::Tensor demb_dec,
+                                      at::Tensor seed,
+                                      at::Tensor demb_out,
+                                      double p, int64_t salt,
+                                      int64_t D, int64_t E, int64_t H) {
+    CHECK_GPU(dgates); CHECK_CONTIG(dgates); CHECK_BF16(dgates);
+    CHECK_GPU(wl_t); CHECK_CONTIG(wl_t); CHECK_BF16(wl_t);
+    int64_t M = dgates.size(0), K = dgates.size(1);
+    int64_t N = wl_t.size(0);
+    TORCH_CHECK(N == D + E + H);
+    TORCH_CHECK(M <= 128 && K % 32 == 0,
+                "dense_dx_fuse: skinny shapes only");
+    auto dpooled = at::empty({M, D}, dgates.options());
+    auto dsth = at::empty({M, H}, dgates.options());
+    hipStream_t stream = at::cuda::getCurrentCUDAStream();
+
+    int nblocks = cdiv(N, 64);
+    int splitk = 1;
+    while (nblocks * splitk < 192 && splitk < 8 &&
+           (int)(K / 32) >= 2 * splitk)
+        splitk *= 2;
+    auto yf = at::empty({splitk, M, N}, dgates.options()
+                                            .dtype(at::kFloat));
+#define LAUNCH_SKX(RF) \
+    hipLaunchKernelGGL((skinny_gemm_kernel<RF>), dim3(nblocks, splitk), \
+                       dim3(256), 0, stream, \
+                       (const bf16*)dgates.data_ptr(), \
+                       (const bf16*)wl_t.data_ptr(), nullptr, nullptr, \
+                       (float*)yf.data_ptr(), nullptr, (int)M, (int)N, \
+                       (int)K, 0, splitk)
+    if (M <= 32) LAUNCH_SKX(2);
+    else if (M <= 64) LAUNCH_SKX(4);
+    else LAUNCH_SKX(8);
+#undef LAUNCH_SKX
+    int64_t n = M * N;
+    hipLaunchKernelGGL(skinny_epi_dx_kernel, dim3(cdiv(n, 256)),
+                       dim3(256), 0, stream,
+                       (const float*)yf.data_ptr(),
+                       (const bf16*)dpool_dec.data_ptr(),
+                       (const bf16*)demb_dec.data_ptr(),
+                       (const int64_t*)seed.data_ptr(),
+                       (bf16*)dpooled.data_ptr(),
+                       (bf16*)demb_out.data_ptr(),
+                       (bf16*)dsth.data_ptr(),
+                       n, splitk, (int)M, (int)D, (int)E, (int)H,
+                       (float)p, (int)salt);
+    HIP_OK(hipGetLastError());
+    return {dpooled, dsth};
+}

@@ -477,6 +477,7 @@ __global__ void attn_scores_bwd_kernel(
         const int64_t* __restrict__ seed_p,
         bf16* __restrict__ dt1, float* __restrict__ dt2,
         float* __restrict__ dvf,
+        const bf16* __restrict__ t1y,  // when set: dt1 <- dt * (1-y^2)
         int B, int L, int A, int lchunk, float p, int salt) {
     // wave-per-row bf16x8 streaming; per-wave column partials are combined
     // through LDS and flushed with ONE atomic per thread-column — a lane
@@ -505,12 +506,20 @@ __global__ void attn_scores_bwd_kernel(
             bf16x8 o;
             float sc[8];
             drop_scale8(seed, salt, (uint32_t)((row * A + a0) >> 3), p, sc);
+            bf16x8 yv = {};
+            if (t1y != nullptr)
+                yv = *(const bf16x8*)(t1y + row * A + a0);
 #pragma unroll
             for (int e = 0; e < 8; ++e) {
                 dv_acc[ch][e] += bf2f(td[e]) * dl;
                 float dt = dl * bf2f(vv[e]) * sc[e];
-                o[e] = f2bf(dt);
                 dt2_acc[ch][e] += dt;
+                if (t1y != nullptr) {
+                    // fused tanh backward: dt1 is dpre of att fc_1a
+                    float y = bf2f(yv[e]);
+                    dt *= (1.f - y * y);
+                }
+                o[e] = f2bf(dt);
             }
             *(bf16x8*)(dt1 + row * A + a0) = o;
         }
@@ -533,17 +542,41 @@ __global__ void attn_scores_bwd_kernel(
     }
 }
 
+std::vector<at::Tensor> attn_scores_bwd_tanh(
+        at::Tensor tdrop, at::Tensor v, at::Tensor dlogits,
+        at::Tensor seed, double p, int64_t salt, int64_t L,
+        at::Tensor dv_acc, at::Tensor t1y, at::Tensor dt1_out);
+
 std::vector<at::Tensor> attn_scores_bwd_acc(at::Tensor tdrop, at::Tensor v,
                                             at::Tensor dlogits,
                                             at::Tensor seed, double p,
                                             int64_t salt, int64_t L,
                                             at::Tensor dv_acc) {
+    return attn_scores_bwd_tanh(tdrop, v, dlogits, seed, p, salt, L,
+                                dv_acc, at::Tensor(), at::Tensor());
+}
+
+std::vector<at::Tensor> attn_scores_bwd_tanh(
+        at::Tensor tdrop, at::Tensor v, at::Tensor dlogits,
+        at::Tensor seed, double p, int64_t salt, int64_t L,
+        at::Tensor dv_acc, at::Tensor t1y, at::Tensor dt1_out) {
     CHECK_GPU(tdrop); CHECK_CONTIG(tdrop); CHECK_BF16(tdrop);
     int rows = tdrop.size(0), A = tdrop.size(1);
     TORCH_CHECK(A % 512 == 0 && A <= 2048,
                 "attn_scores_bwd requires A % 512 == 0, A <= 2048");
     int B = rows / (int)L;
-    auto dt1 = at::empty_like(tdrop);
+    at::Tensor dt1;
+    if (dt1_out.defined() && dt1_out.numel() > 0) {
+        TORCH_CHECK(dt1_out.numel() == tdrop.numel());
+        dt1 = dt1_out;
+    } else {
+        dt1 = at::empty_like(tdrop);
+    }
+    const bf16* t1y_ptr = nullptr;
+    if (t1y.defined() && t1y.numel() > 0) {
+        TORCH_CHECK(t1y.numel() == tdrop.numel());
+        t1y_ptr = (const bf16*)t1y.data_ptr();
+    }
     auto dt2 = at::zeros({B, A}, tdrop.options().dtype(at::kFloat));
     at::Tensor dvf;
     if (dv_acc.defined() && dv_acc.numel() > 0)
@@ -563,7 +596,7 @@ std::vector<at::Tensor> attn_scores_bwd_acc(at::Tensor tdrop, at::Tensor v,
                        (const float*)dlogits.data_ptr(), \
                        (const int64_t*)seed.data_ptr(), \
                        (bf16*)dt1.data_ptr(), (float*)dt2.data_ptr(), \
-                       (float*)dvf.data_ptr(), \
+                       (float*)dvf.data_ptr(), t1y_ptr, \
                        B, (int)L, A, lchunk, (float)p, (int)salt)
     switch (A / 512) {
         case 1: LAUNCH_SB(1); break;
