@@ -201,11 +201,13 @@ class DecoderCoreBPTT(torch.autograd.Function):
         for t in range(T - 1, -1, -1):
             s = t * 16
             sl = slice(t * B, (t + 1) * B)
-            dh_raw, dpool_dec, demb_dec = _C.dexp_fuse(
-                d_expd[sl], d_out_carry, d_sth_carry, seed, p_fc, p_lstm,
-                s, D, E)
-            dgates, dc_prev = _C.lstm_pointwise_bwd_out(
-                gates_l[t], cprev_l[t], dh_raw, dc_carry, 1.0, DG[sl])
+            # dexp scatter + LSTM pointwise backward fused: dh_raw
+            # stays in registers inside the one kernel
+            dc_prev, dpool_dec, demb_dec = _C.dexp_lstm_bwd(
+                d_expd[sl], d_out_carry, d_sth_carry, seed,
+                gates_l[t], cprev_l[t], dc_carry, DG[sl],
+                p_fc, p_lstm, s, D, E, 1.0)
+            dgates = DG[sl]
             if B <= 128:
                 # dxh GEMM + the dx scatter in one epilogue
                 dpooled, d_sth_carry = _C.dense_dx_fuse(

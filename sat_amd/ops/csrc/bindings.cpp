@@ -5,6 +5,15 @@
 at::Tensor dense_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, int64_t act);
 at::Tensor dense_fwd_drop(at::Tensor x, at::Tensor w, at::Tensor seed,
                           double p, int64_t salt);
+std::vector<at::Tensor> dexp_lstm_bwd(at::Tensor dexpd,
+                                      at::Tensor d_out_carry,
+                                      at::Tensor d_sth_carry,
+                                      at::Tensor seed, at::Tensor gates,
+                                      at::Tensor c_prev, at::Tensor dc,
+                                      at::Tensor dgates_out,
+                                      double p_fc, double p_lstm,
+                                      int64_t s, int64_t D, int64_t E,
+                                      double fb);
 std::vector<at::Tensor> dense_dx_fuse(at::Tensor dgates, at::Tensor wl_t,
                                       at::Tensor dpool_dec,
                                       at::Tensor demb_dec,
@@ -122,6 +131,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("dense_fwd", &dense_fwd, "MFMA GEMM + bias/act (bf16)");
     m.def("dense_fwd_drop", &dense_fwd_drop,
           "skinny GEMM with hash-dropout fused into the split-K epilogue");
+    m.def("dexp_lstm_bwd", &dexp_lstm_bwd,
+          "dexp scatter + LSTM pointwise backward in one launch");
     m.def("dense_dx_fuse", &dense_dx_fuse,
           "dxh GEMM with the dx scatter fused into the epilogue");
     m.def("bias_act_nhwc", &bias_act_nhwc);

@@ -412,3 +412,97 @@ at::Tensor hash_dropout_slabs(at::Tensor x, at::Tensor seed, double p,
     HIP_OK(hipGetLastError());
     return y;
 }
+
+// ---- dexp_fuse + LSTM pointwise backward in ONE launch: the H-range
+// threads continue straight into the gate math (dh_raw never touches
+// HBM); D/E-range threads do the dexp scatter as before ----
+
+__global__ void dexp_lstm_bwd_kernel(const bf16* __restrict__ dexpd,
+                                     const bf16* __restrict__ d_out_carry,
+                                     const bf16* __restrict__ d_sth_carry,
+                                     const int64_t* __restrict__ seed_p,
+                                     const bf16* __restrict__ gates,
+                                     const bf16* __restrict__ c,
+                                     const bf16* __restrict__ dc,
+                                     bf16* __restrict__ dgates,
+                                     bf16* __restrict__ dc_prev,
+                                     bf16* __restrict__ dpool_dec,
+                                     bf16* __restrict__ demb_dec,
+                                     int B, int H, int D, int E,
+                                     float p_fc, float p_lstm, int s,
+                                     float fb) {
+    const uint32_t seed = (uint32_t)(*seed_p);
+    const int W = H + D + E;
+    int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= (int64_t)B * W) return;
+    int b = idx / W, j = idx % W;
+    float dexp = bf2f(dexpd[idx])
+        * dscale(seed, s + 6, (uint32_t)(b * W + j), p_fc);
+    if (j < H) {
+        uint32_t hidx = (uint32_t)(b * H + j);
+        float dout = dexp + bf2f(d_out_carry[(int64_t)b * H + j]);
+        float dhv = dout * dscale(seed, s + 4, hidx, p_lstm)
+            + bf2f(d_sth_carry[(int64_t)b * H + j])
+            * dscale(seed, s + 5, hidx, p_lstm);
+        // LSTM pointwise backward (same math as lstm_pw_bwd_kernel)
+        const bf16* g = gates + (int64_t)b * 4 * H;
+        float gi = 1.f / (1.f + __expf(-bf2f(g[j])));
+        float gj = tanhf(bf2f(g[H + j]));
+        float gf = 1.f / (1.f + __expf(-(bf2f(g[2 * H + j]) + fb)));
+        float go = 1.f / (1.f + __expf(-bf2f(g[3 * H + j])));
+        float cp = bf2f(c[(int64_t)b * H + j]);
+        float cn = cp * gf + gi * gj;
+        float tc = tanhf(cn);
+        float dcv = (dc != nullptr) ? bf2f(dc[(int64_t)b * H + j]) : 0.f;
+        float dct = dcv + dhv * go * (1.f - tc * tc);
+        bf16* dg = dgates + (int64_t)b * 4 * H;
+        dg[j]         = f2bf(dct * gj * gi * (1.f - gi));
+        dg[H + j]     = f2bf(dct * gi * (1.f - gj * gj));
+        dg[2 * H + j] = f2bf(dct * cp * gf * (1.f - gf));
+        dg[3 * H + j] = f2bf(dhv * tc * go * (1.f - go));
+        dc_prev[(int64_t)b * H + j] = f2bf(dct * gf);
+    } else if (j < H + D) {
+        dpool_dec[(int64_t)b * D + (j - H)] = f2bf(dexp);
+    } else {
+        demb_dec[(int64_t)b * E + (j - H - D)] = f2bf(dexp);
+    }
+}
+
+std::vector<at::Tensor> dexp_lstm_bwd(at::Tensor dexpd,
+                                      at::Tensor d_out_carry,
+                                      at::Tensor d_sth_carry,
+                                      at::Tensor seed, at::Tensor gates,
+                                      at::Tensor c_prev, at::Tensor dc,
+                                      at::Tensor dgates_out,
+                                      double p_fc, double p_lstm,
+                                      int64_t s, int64_t D, int64_t E,
+                                      double fb) {
+    CHECK_GPU(dexpd); CHECK_BF16(dexpd);
+    int B = d_out_carry.size(0), H = d_out_carry.size(1);
+    TORCH_CHECK(dexpd.numel() == (int64_t)B * (H + D + E));
+    TORCH_CHECK(dgates_out.numel() == (int64_t)B * 4 * H);
+    auto dc_prev = at::empty({B, H}, dexpd.options());
+    auto dpool_dec = at::empty({B, D}, dexpd.options());
+    auto demb_dec = at::empty({B, E}, dexpd.options());
+    const bf16* dc_ptr = nullptr;
+    if (dc.defined() && dc.numel() > 0)
+        dc_ptr = (const bf16*)dc.data_ptr();
+    int64_t n = (int64_t)B * (H + D + E);
+    hipStream_t st = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(dexp_lstm_bwd_kernel, dim3(cdiv(n, 256)),
+                       dim3(256), 0, st,
+                       (const bf16*)dexpd.data_ptr(),
+                       (const bf16*)d_out_carry.data_ptr(),
+                       (const bf16*)d_sth_carry.data_ptr(),
+                       (const int64_t*)seed.data_ptr(),
+                       (const bf16*)gates.data_ptr(),
+                       (const bf16*)c_prev.data_ptr(), dc_ptr,
+                       (bf16*)dgates_out.data_ptr(),
+                       (bf16*)dc_prev.data_ptr(),
+                       (bf16*)dpool_dec.data_ptr(),
+                       (bf16*)demb_dec.data_ptr(),
+                       B, H, (int)D, (int)E,
+                       (float)p_fc, (float)p_lstm, (int)s, (float)fb);
+    HIP_OK(hipGetLastError());
+    return {dc_prev, dpool_dec, demb_dec};
+}
