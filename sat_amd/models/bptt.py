@@ -125,17 +125,21 @@ class DecoderCoreBPTT(torch.autograd.Function):
 
             _C.lstm_in_fuse(pooled, emb_c, last_word, state_h, seed,
                             p_lstm, s + 3, XH[sl])
+            od_next = ODROP[(t + 1) * B:(t + 2) * B] if t + 1 < T \
+                else empty_b
             if fuse_small:
-                gates, h_raw, c_new = _C.dense_lstm_fwd(
-                    XH[sl], wl_c, bl_c, memory, 1.0)
+                # gates GEMM -> gate math + expand scatter in ONE
+                # epilogue (h_raw / out_t never touch HBM)
+                gates, c_new, sth_t = _C.dense_lstm_expand_fwd(
+                    XH[sl], wl_c, bl_c, memory, pooled, emb_c,
+                    last_word, seed, EXPD[sl], od_next,
+                    1.0, p_lstm, p_fc, s)
             else:
                 gates = _C.dense_fwd(XH[sl], wl_c, bl_c, ACT_NONE)
                 h_raw, c_new = _C.lstm_pointwise_fwd(gates, memory, 1.0)
-            od_next = ODROP[(t + 1) * B:(t + 2) * B] if t + 1 < T \
-                else empty_b
-            out_t, sth_t = _C.expand_fuse(
-                h_raw, pooled, emb_c, last_word, seed, EXPD[sl],
-                od_next, p_lstm, p_fc, s)
+                out_t, sth_t = _C.expand_fuse(
+                    h_raw, pooled, emb_c, last_word, seed, EXPD[sl],
+                    od_next, p_lstm, p_fc, s)
 
             t1s.append(t1)
             t2s.append(t2)

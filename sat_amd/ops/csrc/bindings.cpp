@@ -5,6 +5,11 @@
 at::Tensor dense_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, int64_t act);
 at::Tensor dense_fwd_drop(at::Tensor x, at::Tensor w, at::Tensor seed,
                           double p, int64_t salt);
+std::vector<at::Tensor> dense_lstm_expand_fwd(
+        at::Tensor xh, at::Tensor wl, at::Tensor bl, at::Tensor c_prev,
+        at::Tensor pooled, at::Tensor table, at::Tensor ids,
+        at::Tensor seed, at::Tensor expdrop, at::Tensor od_next,
+        double fb, double p_lstm, double p_fc, int64_t s);
 std::vector<at::Tensor> dexp_lstm_bwd(at::Tensor dexpd,
                                       at::Tensor d_out_carry,
                                       at::Tensor d_sth_carry,
@@ -131,6 +136,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("dense_fwd", &dense_fwd, "MFMA GEMM + bias/act (bf16)");
     m.def("dense_fwd_drop", &dense_fwd_drop,
           "skinny GEMM with hash-dropout fused into the split-K epilogue");
+    m.def("dense_lstm_expand_fwd", &dense_lstm_expand_fwd,
+          "gates GEMM + LSTM gate math + expand scatter in 2 launches");
     m.def("dexp_lstm_bwd", &dexp_lstm_bwd,
           "dexp scatter + LSTM pointwise backward in one launch");
     m.def("dense_dx_fuse", &dense_dx_fuse,

@@ -901,3 +901,123 @@ std::vector<at::Tensor> dense_dx_fuse(at::Tensor dgates, at::Tensor wl_t,
     HIP_OK(hipGetLastError());
     return {dpooled, dsth};
 }
+
+// ---------------------------------------------------------------------
+// LSTM gates epilogue + the expand scatter in ONE launch: the H-range
+// threads finish the gate math and write sth / EXPD-h / od_next
+// directly (h_raw and out_t never touch HBM); D/E threads do the
+// pooled / embedding-gather parts of the expand row.  Replaces
+// skinny_epi_lstm + expand_fuse.  Dropout salts identical to the pair
+// it replaces (s+4 out, s+5 state, s+6 expand, s+17 next attend).
+// ---------------------------------------------------------------------
+
+__global__ void skinny_epi_lstm_expand_kernel(
+        const float* __restrict__ Yf, const bf16* __restrict__ bias,
+        const bf16* __restrict__ c_prev,
+        const bf16* __restrict__ pooled,
+        const bf16* __restrict__ table,
+        const int64_t* __restrict__ ids,
+        const int64_t* __restrict__ seed_p,
+        bf16* __restrict__ gates, bf16* __restrict__ c_out,
+        bf16* __restrict__ sth_t, bf16* __restrict__ expdrop,
+        bf16* __restrict__ od_next,
+        int B, int H, int D, int E, int splitk,
+        float fb, float p_lstm, float p_fc, int s) {
+    const uint32_t seed = (uint32_t)(*seed_p);
+    const int W = H + D + E;
+    int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (idx >= (int64_t)B * W) return;
+    int b = idx / W, j = idx % W;
+    float val;
+    if (j < H) {
+        int64_t n = (int64_t)B * 4 * H;
+        float g[4];
+#pragma unroll
+        for (int q = 0; q < 4; ++q) {
+            int64_t off = (int64_t)b * 4 * H + q * H + j;
+            float acc = 0.f;
+            for (int k = 0; k < splitk; ++k) acc += Yf[k * n + off];
+            if (bias != nullptr) acc += bf2f(bias[q * H + j]);
+            gates[off] = f2bf(acc);
+            g[q] = acc;
+        }
+        float gi = sigmoidf(g[0]);
+        float gj = tanhf(g[1]);
+        float gf = sigmoidf(g[2] + fb);
+        float go = sigmoidf(g[3]);
+        float cn = bf2f(c_prev[(int64_t)b * H + j]) * gf + gi * gj;
+        float h = tanhf(cn) * go;
+        c_out[(int64_t)b * H + j] = f2bf(cn);
+        uint32_t hidx = (uint32_t)(b * H + j);
+        float ot = h * dscaleg(seed, s + 4, hidx, p_lstm);
+        sth_t[(int64_t)b * H + j] =
+            f2bf(h * dscaleg(seed, s + 5, hidx, p_lstm));
+        if (od_next != nullptr)
+            od_next[(int64_t)b * H + j] =
+                f2bf(ot * dscaleg(seed, s + 16 + 1, hidx, p_fc));
+        val = ot;
+    } else if (j < H + D) {
+        val = bf2f(pooled[(int64_t)b * D + (j - H)]);
+    } else {
+        val = bf2f(table[ids[b] * E + (j - H - D)]);
+    }
+    expdrop[idx] = f2bf(val * dscaleg(seed, s + 6,
+                                      (uint32_t)(b * W + j), p_fc));
+}
+
+std::vector<at::Tensor> dense_lstm_expand_fwd(
+        at::Tensor xh, at::Tensor wl, at::Tensor bl, at::Tensor c_prev,
+        at::Tensor pooled, at::Tensor table, at::Tensor ids,
+        at::Tensor seed, at::Tensor expdrop, at::Tensor od_next,
+        double fb, double p_lstm, double p_fc, int64_t s) {
+    int64_t M = xh.size(0), K = xh.size(1), N = wl.size(0);
+    int B = c_prev.size(0), H = c_prev.size(1);
+    int D = pooled.size(1), E = table.size(1);
+    TORCH_CHECK(M == B && N == 4 * H && M <= 128 && K % 32 == 0);
+    auto gates = at::empty({M, N}, xh.options());
+    auto c_out = at::empty({B, H}, xh.options());
+    auto sth_t = at::empty({B, H}, xh.options());
+    hipStream_t stream = at::cuda::getCurrentCUDAStream();
+
+    int nblocks = cdiv(N, 64);
+    int splitk = 1;
+    while (nblocks * splitk < 192 && splitk < 8 &&
+           (int)(K / 32) >= 2 * splitk)
+        splitk *= 2;
+    auto yf = at::empty({splitk, M, N},
+                        xh.options().dtype(at::kFloat));
+#define LAUNCH_SKL(RF) \
+    hipLaunchKernelGGL((skinny_gemm_kernel<RF>), dim3(nblocks, splitk), \
+                       dim3(256), 0, stream, \
+                       (const bf16*)xh.data_ptr(), \
+                       (const bf16*)wl.data_ptr(), nullptr, nullptr, \
+                       (float*)yf.data_ptr(), nullptr, (int)M, (int)N, \
+                       (int)K, 0, splitk)
+    if (M <= 32) LAUNCH_SKL(2);
+    else if (M <= 64) LAUNCH_SKL(4);
+    else LAUNCH_SKL(8);
+#undef LAUNCH_SKL
+    const bf16* bias_ptr = nullptr;
+    if (bl.defined() && bl.numel() > 0)
+        bias_ptr = (const bf16*)bl.data_ptr();
+    bf16* od_ptr = nullptr;
+    if (od_next.defined() && od_next.numel() > 0)
+        od_ptr = (bf16*)od_next.data_ptr();
+    int64_t n = (int64_t)B * (H + D + E);
+    hipLaunchKernelGGL(skinny_epi_lstm_expand_kernel,
+                       dim3(cdiv(n, 256)), dim3(256), 0, stream,
+                       (const float*)yf.data_ptr(), bias_ptr,
+                       (const bf16*)c_prev.data_ptr(),
+                       (const bf16*)pooled.data_ptr(),
+                       (const bf16*)table.data_ptr(),
+                       (const int64_t*)ids.data_ptr(),
+                       (const int64_t*)seed.data_ptr(),
+                       (bf16*)gates.data_ptr(),
+                       (bf16*)c_out.data_ptr(),
+                       (bf16*)sth_t.data_ptr(),
+                       (bf16*)expdrop.data_ptr(), od_ptr,
+                       B, H, D, E, splitk,
+                       (float)fb, (float)p_lstm, (float)p_fc, (int)s);
+    HIP_OK(hipGetLastError());
+    return {gates, c_out, sth_t};
+}
