@@ -365,8 +365,15 @@ class DecodeHeadBPTT(torch.autograd.Function):
         DP1 = _C.act_bwd(DHID, HID, ACT_TANH)     # dpre of dec fc_1
 
         # weight grads first: their AccumulateGrad + DDP hooks are what
-        # the core's backward overlaps with
-        dWd2 = DL.t().matmul(HD).float()
+        # the core's backward overlaps with.  dWd2 through the in-tree
+        # tiled kernel: hipBLASLt ran this [V,TB]x[TB,Dd] shape at
+        # 43 TF / 152 us (r02e profile); two small transposes + the
+        # MFMA tile land at ~25 us.
+        if fast_dx:
+            dWd2 = _C.dense_fwd(DL.t().contiguous(),
+                                HD.t().contiguous(), eb, 0).float()
+        else:
+            dWd2 = DL.t().matmul(HD).float()
         dbd2 = DL.sum(0, dtype=torch.float32)
         dWd1 = DP1.t().matmul(expd).float()
         dbd1 = DP1.sum(0, dtype=torch.float32)
