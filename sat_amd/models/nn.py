@@ -73,11 +73,19 @@ class NN(object):
 
     def reg_loss(self):
         """Σ scale · l2_loss(w) (l2_loss = sum(w²)/2, TF semantics) plus
-        any activity-regularizer terms recorded this forward."""
+        any activity-regularizer terms recorded this forward.
+
+        On GPU the kernel-L2 term runs as ONE multi-tensor foreach pass
+        (a per-param eager loop cost ~55 launches/step)."""
         total = torch.zeros(())
         if self._regularized:
-            total = sum(s * 0.5 * (p.float() ** 2).sum()
-                        for p, s in self._regularized)
+            if self._regularized[0][0].is_cuda:
+                params = tuple(p for p, _ in self._regularized)
+                scales = tuple(s for _, s in self._regularized)
+                total = _RegL2.apply(scales, *params)
+            else:
+                total = sum(s * 0.5 * (p.float() ** 2).sum()
+                            for p, s in self._regularized)
         if self._act_losses:
             act = sum(self._act_losses)
             self._act_losses = []
@@ -86,6 +94,37 @@ class NN(object):
 
     def dropout(self, x):
         return ops.dropout(x, self.fc_drop_rate, self.is_train)
+
+
+class _RegL2(torch.autograd.Function):
+    """Multi-tensor Σ scale·0.5·||p||² with closed-form backward
+    (d/dp = d·scale·p): ~6 launches forward + one foreach backward
+    instead of per-param pow/reduce/mul/add chains."""
+
+    _scale_cache = {}
+
+    @staticmethod
+    def forward(ctx, scales, *params):
+        ctx.save_for_backward(*params)
+        ctx.scales = scales
+        norms = torch._foreach_norm(params, 2)
+        sv = torch.stack([n.float() for n in norms])
+        # cached device constant: creating it per call would be a host
+        # copy inside the captured graph
+        key = (scales, sv.device)
+        sc = _RegL2._scale_cache.get(key)
+        if sc is None:
+            sc = torch.tensor(scales, dtype=torch.float32,
+                              device=sv.device)
+            _RegL2._scale_cache[key] = sc
+        return (sv * sv * sc * 0.5).sum()
+
+    @staticmethod
+    def backward(ctx, d):
+        params = ctx.saved_tensors
+        gs = torch._foreach_mul(params, list(ctx.scales))
+        gs = torch._foreach_mul(gs, d)
+        return (None, *gs)
 
 
 class Dense(tnn.Module):
