@@ -528,3 +528,102 @@ def test_conv1x1_as_gemm_route():
             if relu:
                 ref = torch.relu(ref)
         assert _rel_err(y, ref) < 2e-2, (Cin, Cout)
+
+
+def test_fused_kernels_match_their_unfused_pairs_bitwise():
+    """Every r02 fusion must reproduce the kernel pair it replaced
+    BITWISE — the backward regenerates dropout masks from the same
+    counter hash, so even a 1-ulp divergence would corrupt training."""
+    from sat_amd import _C
+    torch.manual_seed(33)
+    B, H, D, E, A = 32, 512, 512, 512, 512
+    V_ = 1000
+    I = D + E
+    W = H + D + E
+    seed = torch.tensor(777, dtype=torch.int64, device=DEV)
+    p_fc, p_lstm, s = 0.5, 0.3, 48
+
+    # dense_fwd_drop == dense_fwd + hash_dropout
+    x = _bf(torch.randn(B, A))
+    w = _bf(torch.randn(H, A) * 0.05)
+    eb = torch.empty(0, dtype=torch.bfloat16, device=DEV)
+    y1 = _C.dense_fwd_drop(x, w, seed, p_fc, s + 1)
+    y2 = _C.hash_dropout(_C.dense_fwd(x, w, eb, 0), seed, p_fc, s + 1)
+    assert torch.equal(y1, y2)
+
+    # dense_dx_fuse == dense_fwd + dx_fuse
+    dgates = _bf(torch.randn(B, 4 * H) * 0.1)
+    wl_t = _bf(torch.randn(I + H, 4 * H) * 0.05)
+    dpool_dec = _bf(torch.randn(B, D))
+    demb_dec = _bf(torch.randn(B, E))
+    demb1 = torch.empty(B, E, dtype=torch.bfloat16, device=DEV)
+    demb2 = torch.empty(B, E, dtype=torch.bfloat16, device=DEV)
+    dp1, ds1 = _C.dense_dx_fuse(dgates, wl_t, dpool_dec, demb_dec,
+                                seed, demb1, p_lstm, s + 3, D, E, H)
+    dxh = _C.dense_fwd(dgates, wl_t, eb, 0)
+    dp2, ds2 = _C.dx_fuse(dxh, dpool_dec, demb_dec, seed, demb2,
+                          p_lstm, s + 3, H)
+    assert torch.equal(dp1, dp2)
+    assert torch.equal(ds1, ds2)
+    assert torch.equal(demb1, demb2)
+
+    # dense_lstm_expand_fwd == dense_lstm_fwd + expand_fuse
+    xh = _bf(torch.randn(B, I + H) * 0.1)
+    wl = _bf(torch.randn(4 * H, I + H) * 0.05)
+    bl = _bf(torch.zeros(4 * H))
+    cprev = _bf(torch.randn(B, H))
+    pooled = _bf(torch.randn(B, D))
+    table = _bf(torch.randn(V_, E))
+    ids = torch.randint(0, V_, (B,), device=DEV)
+    e1 = torch.empty(B, W, dtype=torch.bfloat16, device=DEV)
+    e2 = torch.empty(B, W, dtype=torch.bfloat16, device=DEV)
+    od1 = torch.empty(B, H, dtype=torch.bfloat16, device=DEV)
+    od2 = torch.empty(B, H, dtype=torch.bfloat16, device=DEV)
+    g1, c1, sth1 = _C.dense_lstm_expand_fwd(
+        xh, wl, bl, cprev, pooled, table, ids, seed, e1, od1,
+        1.0, p_lstm, p_fc, s)
+    g2, h2, c2 = _C.dense_lstm_fwd(xh, wl, bl, cprev, 1.0)
+    _out2, sth2 = _C.expand_fuse(h2, pooled, table, ids, seed, e2, od2,
+                                 p_lstm, p_fc, s)
+    assert torch.equal(g1, g2)
+    assert torch.equal(c1, c2)
+    assert torch.equal(sth1, sth2)
+    assert torch.equal(e1, e2)
+    assert torch.equal(od1, od2)
+
+    # dexp_lstm_bwd == dexp_fuse + lstm_pointwise_bwd_out
+    dexpd = _bf(torch.randn(B, W) * 0.1)
+    doc = _bf(torch.randn(B, H) * 0.1)
+    dsc = _bf(torch.randn(B, H) * 0.1)
+    dcc = _bf(torch.randn(B, H) * 0.1)
+    DG1 = torch.empty(B, 4 * H, dtype=torch.bfloat16, device=DEV)
+    DG2 = torch.empty(B, 4 * H, dtype=torch.bfloat16, device=DEV)
+    dcp1, dpd1, ded1 = _C.dexp_lstm_bwd(
+        dexpd, doc, dsc, seed, g2, cprev, dcc, DG1,
+        p_fc, p_lstm, s, D, E, 1.0)
+    dh2, dpd2, ded2 = _C.dexp_fuse(dexpd, doc, dsc, seed, p_fc, p_lstm,
+                                   s, D, E)
+    dg2, dcp2 = _C.lstm_pointwise_bwd_out(g2, cprev, dh2, dcc, 1.0, DG2)
+    assert torch.equal(DG1, DG2)
+    assert torch.equal(dcp1, dcp2)
+    assert torch.equal(dpd1, dpd2)
+    assert torch.equal(ded1, ded2)
+
+    # attn_scores_bwd_tanh == attn_scores_bwd_acc + act_bwd_out
+    L = 196
+    t1y = _bf(torch.tanh(torch.randn(B * L, A)))
+    tdrop = _bf(torch.randn(B * L, A))
+    dlog = torch.randn(B, L, device=DEV)
+    vvec = _bf(torch.randn(A) * 0.05)
+    dv1 = torch.zeros(A, dtype=torch.float32, device=DEV)
+    dv2 = torch.zeros(A, dtype=torch.float32, device=DEV)
+    out1 = torch.empty(B * L, A, dtype=torch.bfloat16, device=DEV)
+    _d1, dt2a, _ = _C.attn_scores_bwd_tanh(
+        tdrop, vvec, dlog, seed, p_fc, s + 2, L, dv1, t1y, out1)
+    dt1b, dt2b, _ = _C.attn_scores_bwd_acc(
+        tdrop, vvec, dlog, seed, p_fc, s + 2, L, dv2)
+    out2 = torch.empty_like(out1)
+    _C.act_bwd_out(dt1b, t1y, 1, out2)
+    assert torch.equal(out1, out2)
+    assert torch.equal(dt2a, dt2b)
+    assert torch.equal(dv1, dv2)
