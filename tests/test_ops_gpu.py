@@ -637,5 +637,9 @@ def test_fused_kernels_match_their_unfused_pairs_bitwise():
     out2 = torch.empty_like(out1)
     _C.act_bwd_out(dt1b, t1y, 1, out2)
     _one_ulp(out1, out2)         # pair rounds dt1 to bf16 first
-    assert torch.equal(dt2a, dt2b)
-    assert torch.equal(dv1, dv2)
+    # dt2/dv are fp32 atomicAdd accumulations across l-chunk blocks:
+    # summation order is nondeterministic between runs, so two
+    # IDENTICAL kernels differ in the last ulps — compare to fp32
+    # accumulation-order tolerance
+    assert torch.allclose(dt2a, dt2b, rtol=1e-4, atol=1e-4)
+    assert torch.allclose(dv1, dv2, rtol=1e-4, atol=1e-4)
