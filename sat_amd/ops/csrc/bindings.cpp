@@ -41,6 +41,8 @@ at::Tensor conv_igemm_8p_fwd(at::Tensor padded, at::Tensor w_ohwi,
 at::Tensor conv_igemm_glds64_fwd(at::Tensor padded, at::Tensor w_ohwi,
                                  at::Tensor bias, int64_t Hh, int64_t Ww,
                                  bool relu);
+at::Tensor dense_8p_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
+                        int64_t act);
 at::Tensor conv3x3_wgrad(at::Tensor xpad, at::Tensor dy_rows,
                          int64_t Hh, int64_t Ww);
 at::Tensor conv_igemm_glds_fwd(at::Tensor padded, at::Tensor w_ohwi,
@@ -151,6 +153,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("conv_igemm_8p_fwd", &conv_igemm_8p_fwd,
           "8-phase deep-pipelined implicit-GEMM conv (Cout%256==0)");
     m.def("conv_igemm_glds64_fwd", &conv_igemm_glds64_fwd);
+    m.def("dense_8p_fwd", &dense_8p_fwd,
+          "8-phase deep-pipelined dense GEMM (big-M shapes)");
     m.def("conv3x3_wgrad", &conv3x3_wgrad,
           "3x3 conv weight grad: transpose-staged MFMA split-K reduce");
     m.def("dense_lstm_fwd", &dense_lstm_fwd);

@@ -422,10 +422,17 @@ at::Tensor dense_fwd_out(at::Tensor x, at::Tensor w, at::Tensor bias,
                                splitk);
         }
 #undef LAUNCH_SKINNY
+    } else if (M >= 16384 && N >= 256 && N % 8 == 0 && K % 64 == 0
+               && K >= 128 && !(out.defined() && out.numel() > 0)) {
+        // big-M dense shapes: the 8-phase deep-pipelined 256^2 tile
+        // (gemm8p.hip; the 2-barrier glds tile measured only ~190 TF
+        // here — short K leaves it prologue-bound)
+        extern at::Tensor dense_8p_fwd(at::Tensor, at::Tensor,
+                                       at::Tensor, int64_t);
+        at::Tensor bb = (bias_ptr != nullptr) ? bias : at::Tensor();
+        return dense_8p_fwd(x, w, bb, act);
     } else if (M >= 16384 && N % 128 == 0 && K % 64 == 0) {
-        // big-M dense shapes: glds-staged 128^2 tile (T1 attention
-        // projection measured 148 TF on the register-staged tile; the
-        // glds structure runs the same geometry at 536-658 TF on conv)
+        // (retained) glds-staged 128^2 tile
         dim3 grid(cdiv(N, 128), cdiv(M, 128));
         hipLaunchKernelGGL(dense_glds_kernel, grid, dim3(256), 0,
                            stream,

@@ -33,6 +33,7 @@ setup(
                 os.path.join(CSRC, 'conv3.hip'),
                 os.path.join(CSRC, 'conv8p.hip'),
                 os.path.join(CSRC, 'conv_bwd.hip'),
+                os.path.join(CSRC, 'gemm8p.hip'),
             ],
             extra_compile_args={
                 'cxx': ['-O3', '-std=c++17'],

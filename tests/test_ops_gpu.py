@@ -643,3 +643,23 @@ def test_fused_kernels_match_their_unfused_pairs_bitwise():
     # accumulation-order tolerance
     assert torch.allclose(dt2a, dt2b, rtol=1e-4, atol=1e-4)
     assert torch.allclose(dv1, dv2, rtol=1e-4, atol=1e-4)
+
+
+def test_dense_8p_matches_torch_and_is_race_stable():
+    """8-phase dense GEMM (gemm8p.hip): numerics vs fp32 torch on edge
+    shapes + repeated-launch race screen (new-template discipline)."""
+    from sat_amd import _C
+    torch.manual_seed(44)
+    for M, N, K, act in [(300, 512, 512, 0), (257, 256, 128, 1),
+                         (1024, 520, 192, 2)]:
+        x = _bf(torch.randn(M, K))
+        w = _bf(torch.randn(N, K) * 0.05)
+        b = _bf(torch.randn(N))
+        ref = x.float() @ w.float().t() + b.float()
+        if act == 1:
+            ref = torch.tanh(ref)
+        elif act == 2:
+            ref = torch.relu(ref)
+        for run in range(5):
+            y = _C.dense_8p_fwd(x, w, b, act)
+            assert _rel_err(y, ref) < 2e-2, (M, N, K, act, run)
