@@ -127,7 +127,7 @@ void adam_step_mt(at::Tensor desc, at::Tensor cum, int64_t n_tensors,
                   int64_t total, at::Tensor step_dev, double lr0,
                   double decay_factor, double steps_per_decay, double b1,
                   double b2, double eps, double clip, at::Tensor gsq,
-                  bool zero_grads);
+                  bool zero_grads, bool vec4);
 void adam_step(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
                std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
                at::Tensor step_dev, double lr0, double decay_factor,

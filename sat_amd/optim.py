@@ -115,6 +115,8 @@ class Optimizer(object):
                 self._mt_cum = torch.tensor(
                     cum, dtype=torch.int64).to(dev)
                 self._mt_total = cum[-1]
+                # float4 path needs every tensor boundary 16B-aligned
+                self._mt_vec4 = all(c % 4 == 0 for c in cum)
                 for g, p0 in zip(grads, self.params):
                     # dense storage in any layout (channels_last conv
                     # grads included) is fine: p/g/m/v share the layout
@@ -138,7 +140,8 @@ class Optimizer(object):
                 cfg.initial_learning_rate,
                 cfg.learning_rate_decay_factor, cfg.num_steps_per_decay,
                 cfg.beta1, cfg.beta2, cfg.epsilon,
-                cfg.clip_gradients, gsq, True)
+                cfg.clip_gradients, gsq, True,
+                getattr(self, '_mt_vec4', False))
             return
 
         # ---- eager path (CPU, or non-Adam optimizers) ----
